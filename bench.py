@@ -1,0 +1,149 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: RCCL all-reduce bus bandwidth via the gang path.
+
+This measures BASELINE.json's headline metric — "RCCL all-reduce bus
+GB/s (multi-instance task) 1/2/4/8 GPU; submit->launch p50" — the
+MI355X-native analogue of the reference's mpiBench multi-instance-task
+recipe (reference recipes/mpiBench-OpenMPI/config/docker/jobs.yaml).
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W`.
+For N>1 the driver launches this under torch.distributed.run with one
+rank per GPU; ranks read RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* from env.
+W untimed warmup steps, then EXACTLY K timed steps bracketed by a
+barrier + torch.cuda.synchronize() on both sides; time is the MAX over
+ranks; rank 0 prints ONE JSON line.
+
+One "step" = one in-place all-reduce of a fixed per-rank payload
+(default 256 MiB bf16) through the framework's GangComm — the same code
+path a multi-instance gang task uses.  Value = bus GB/s
+(nccl-tests convention, busbw = algbw * 2*(N-1)/N).  At N=1 an
+all-reduce is a no-op, so the degenerate gang-of-1 step is a local
+in-place reduction y += x of the same payload and the reported value is
+2*S/t (bytes of the two operands over time); this N=1 semantics is
+recorded in config.n1_semantics.
+
+Also reports submit->launch p50 latency (ms) through the local executor
+(single process task path), mirroring BASELINE.md's second metric.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import statistics
+import sys
+import time
+
+
+def measure_submit_launch_p50(samples: int = 10):
+    """Submit tiny process tasks through the executor; p50 of
+    submit->launch latency in ms.  Returns None if the executor is not
+    available (early bootstrap)."""
+    try:
+        from shipyard_amd.executor.latency import measure_submit_launch
+    except Exception:
+        return None
+    try:
+        return measure_submit_launch(samples=samples)
+    except Exception:
+        return None
+
+
+def main() -> int:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=32)
+    p.add_argument("--warmup", type=int, default=8)
+    p.add_argument("--payload-mb", type=int, default=256,
+                   help="per-rank all-reduce payload in MiB")
+    p.add_argument("--latency-samples", type=int, default=10)
+    args = p.parse_args()
+
+    import torch
+
+    from shipyard_amd.comm import GangComm, bus_bandwidth_gbps
+
+    comm = GangComm()
+    world = comm.world
+    dev = comm.device
+    on_gpu = dev.type == "cuda"
+
+    payload_mb = args.payload_mb
+    if not on_gpu:
+        payload_mb = min(payload_mb, 8)  # CPU smoke only
+    numel = payload_mb * (1 << 20) // 2  # bf16 = 2 bytes
+    payload_bytes = numel * 2
+
+    x = torch.randn(numel, dtype=torch.float32, device=dev).to(torch.bfloat16)
+    y = None
+    if world == 1:
+        y = torch.randn(numel, dtype=torch.float32, device=dev).to(
+            torch.bfloat16)
+
+    def step() -> None:
+        if world > 1:
+            comm.all_reduce_(x)
+        else:
+            y.add_(x)
+
+    # submit->launch p50 (rank 0 only, before the timed region)
+    p50_ms = None
+    if comm.rank == 0:
+        p50_ms = measure_submit_launch_p50(args.latency_samples)
+
+    for _ in range(args.warmup):
+        step()
+
+    comm.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t1 = time.perf_counter()
+    comm.barrier()
+
+    elapsed = comm.max_scalar(t1 - t0)
+    ms_per_step = elapsed / args.steps * 1e3
+
+    if world > 1:
+        value = bus_bandwidth_gbps(payload_bytes, elapsed / args.steps, world)
+    else:
+        value = 2.0 * payload_bytes / (elapsed / args.steps) / 1e9
+
+    if comm.rank == 0:
+        out = {
+            "metric": "rccl_allreduce_bus_GBps",
+            "value": round(value, 3),
+            "unit": "GB/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 4),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": "mpibench-rccl-allreduce",
+                "global_batch": world,
+                "seq_len": numel,
+                "parallelism": f"gang{world}",
+                "payload_mb_per_rank": payload_mb,
+                "n1_semantics": "local in-place add (2S bytes moved); "
+                                "busbw undefined at N=1",
+                "submit_launch_p50_ms": p50_ms,
+                "device": dev.type,
+            },
+        }
+        print(json.dumps(out), flush=True)
+
+    comm.shutdown()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,10 @@
+"""RCCL-over-xGMI gang communication layer.
+
+MI355X-native replacement for the reference's MPI runtime synthesis
+(reference convoy/batch.py:4362-4486 `_construct_mpi_command`) — ranks
+rendezvous through a TCP store instead of ``$AZ_BATCH_HOST_LIST`` +
+mpirun, and collectives run over RCCL (torch.distributed backend
+"nccl" on ROCm) bounded by 7 xGMI p2p links x ~153 GB/s per GPU.
+"""
+
+from .collectives import GangComm, bus_bandwidth_gbps  # noqa: F401

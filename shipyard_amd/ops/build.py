@@ -1,0 +1,68 @@
+"""Build the shipyard_amd HIP ops library for gfx950 (MI355X).
+
+The library is pure HIP with a C API (no torch ABI): it is driven through
+ctypes with raw device pointers + the current torch HIP stream, so it
+builds in seconds, travels in-tree to GPU boxes, and is immune to torch
+C++ ABI drift.  ``python -m shipyard_amd.ops.build`` or ``build()``.
+"""
+from __future__ import annotations
+
+import os
+import subprocess
+import sys
+from pathlib import Path
+
+HERE = Path(__file__).resolve().parent
+CSRC = HERE / "csrc"
+LIB = HERE / "libshipyardops.so"
+ARCH = os.environ.get("SHIPYARD_GPU_ARCH", "gfx950")
+
+SOURCES = ["crc32c.hip", "lz4_decode.hip", "sha256.hip"]
+
+
+def hipcc() -> str:
+    for cand in (os.environ.get("HIPCC"), "/opt/rocm/bin/hipcc", "hipcc"):
+        if not cand:
+            continue
+        import shutil
+
+        found = shutil.which(cand) or (cand if os.path.exists(cand) else None)
+        if found:
+            return found
+    raise RuntimeError("hipcc not found; is ROCm installed?")
+
+
+def needs_build() -> bool:
+    if not LIB.exists():
+        return True
+    lib_mtime = LIB.stat().st_mtime
+    for src in SOURCES + ["common.h"]:
+        if (CSRC / src).stat().st_mtime > lib_mtime:
+            return True
+    return False
+
+
+def build(force: bool = False, verbose: bool = True) -> Path:
+    if not force and not needs_build():
+        return LIB
+    cmd = [
+        hipcc(),
+        f"--offload-arch={ARCH}",
+        "-O3",
+        "-std=c++17",
+        "-fPIC",
+        "-shared",
+        "-fvisibility=hidden",
+        *(str(CSRC / s) for s in SOURCES),
+        "-o",
+        str(LIB),
+    ]
+    if verbose:
+        print("[shipyard_amd.ops.build]", " ".join(cmd), file=sys.stderr)
+    subprocess.run(cmd, check=True)
+    return LIB
+
+
+if __name__ == "__main__":
+    build(force="--force" in sys.argv)
+    print(LIB)

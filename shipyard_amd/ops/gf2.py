@@ -1,0 +1,123 @@
+"""GF(2) operator algebra for CRC32C combination.
+
+The CRC register after appending L zero bytes is a linear function of the
+register: ``crc' = M_L . crc`` over GF(2).  These helpers build the 32x32
+bit matrices (stored as 32 uint32 *columns*: ``matvec(M, v) = XOR of M[i]
+for set bits i of v``) used by
+
+  * the crc32c.hip combine tree (8 per-level matrices, shift by
+    ``seg * 2^k`` bytes), and
+  * the host-side finish/combine of raw chunk CRCs (same math as zlib's
+    crc32_combine, re-derived for the Castagnoli polynomial).
+
+Pure Python on ints: matrices are built once per (chunk_size) and cached.
+"""
+from __future__ import annotations
+
+import functools
+from typing import List
+
+CRC32C_POLY_REFLECTED = 0x82F63B78
+
+Matrix = List[int]  # 32 uint32 columns
+
+
+def matvec(m: Matrix, v: int) -> int:
+    r = 0
+    i = 0
+    while v:
+        if v & 1:
+            r ^= m[i]
+        v >>= 1
+        i += 1
+    return r
+
+
+def matmul(a: Matrix, b: Matrix) -> Matrix:
+    return [matvec(a, b[i]) for i in range(32)]
+
+
+def identity() -> Matrix:
+    return [1 << i for i in range(32)]
+
+
+@functools.lru_cache(maxsize=None)
+def _one_bit_operator() -> tuple:
+    # Shift the (reflected) CRC register by one zero bit.
+    m = [0] * 32
+    m[0] = CRC32C_POLY_REFLECTED
+    row = 1
+    for n in range(1, 32):
+        m[n] = row
+        row <<= 1
+    return tuple(m)
+
+
+@functools.lru_cache(maxsize=4096)
+def zero_shift_operator(n_bytes: int) -> tuple:
+    """Operator for appending ``n_bytes`` zero bytes to the message."""
+    if n_bytes < 0:
+        raise ValueError("n_bytes must be >= 0")
+    nbits = n_bytes * 8
+    acc = identity()
+    sq = list(_one_bit_operator())
+    while nbits:
+        if nbits & 1:
+            acc = matmul(sq, acc)
+        nbits >>= 1
+        if nbits:
+            sq = matmul(sq, sq)
+    return tuple(acc)
+
+
+def combine_raw(raw_a: int, raw_b: int, len_b: int) -> int:
+    """raw CRC (init=0, no final xor) of A||B from raw CRCs of A and B."""
+    return matvec(list(zero_shift_operator(len_b)), raw_a) ^ raw_b
+
+
+def finish(raw: int, msg_len: int) -> int:
+    """Standard CRC32C (init 0xFFFFFFFF, final xor) from a raw CRC."""
+    init_term = matvec(list(zero_shift_operator(msg_len)), 0xFFFFFFFF)
+    return (raw ^ init_term) ^ 0xFFFFFFFF
+
+
+def level_matrices(chunk_size: int, threads: int = 256) -> List[int]:
+    """Flattened 8x32 words for crc32c.hip's combine tree."""
+    if chunk_size % (threads * 16):
+        raise ValueError("chunk_size must be a multiple of threads*16")
+    seg = chunk_size // threads
+    out: List[int] = []
+    levels = threads.bit_length() - 1  # 8 for 256
+    for k in range(levels):
+        out.extend(zero_shift_operator(seg * (1 << k)))
+    return out
+
+
+# --- pure-python reference CRC32C (for CPU tests and tails) ---
+@functools.lru_cache(maxsize=1)
+def _table() -> tuple:
+    tab = []
+    for i in range(256):
+        c = i
+        for _ in range(8):
+            c = (c >> 1) ^ (CRC32C_POLY_REFLECTED if c & 1 else 0)
+        tab.append(c)
+    return tuple(tab)
+
+
+def crc32c(data: bytes, crc: int = 0) -> int:
+    """Standard CRC32C of ``data`` (init/final-xor included)."""
+    tab = _table()
+    c = crc ^ 0xFFFFFFFF
+    for b in data:
+        c = (c >> 8) ^ tab[(c ^ b) & 0xFF]
+    return c ^ 0xFFFFFFFF
+
+
+def crc32c_raw(data: bytes) -> int:
+    """Raw CRC register (init 0, no final xor) — matches the kernel."""
+    tab = _table()
+    c = 0
+    for b in data:
+        c = (c >> 8) ^ tab[(c ^ b) & 0xFF]
+    return c
