@@ -1,0 +1,499 @@
+"""Typed settings accessors — the framework's real config API.
+
+Analogue of the reference's convoy/settings.py (~400 accessors returning
+namedtuples; reference convoy/settings.py:154-527).  Raw validated YAML
+dicts go in; frozen dataclasses come out.  Includes the task->runtime
+option compiler (the analogue of reference settings.py:3727
+``task_settings`` which emits docker/singularity run options — here it
+emits the local runtime spec with ROCm GPU binding via
+HIP_VISIBLE_DEVICES instead of ``--gpus``/``--nv``).
+"""
+from __future__ import annotations
+
+import dataclasses
+import datetime
+from typing import Any, Dict, List, Optional, Tuple
+
+from shipyard_amd import utils
+
+MAX_GPUS_PER_NODE = 8
+
+
+def _get(d: Optional[dict], *path, default=None):
+    cur: Any = d
+    for p in path:
+        if not isinstance(cur, dict) or p not in cur:
+            return default
+        cur = cur[p]
+    if cur is None:
+        return default
+    return cur
+
+
+# --------------------------------------------------------------------
+# pool
+# --------------------------------------------------------------------
+@dataclasses.dataclass(frozen=True)
+class AutoscaleScenarioSettings:
+    name: str
+    maximum_gpu_count_dedicated: int
+    maximum_gpu_count_low_priority: int
+    maximum_increment_dedicated: Optional[int]
+    maximum_increment_low_priority: Optional[int]
+    node_deallocation_option: str
+    sample_lookback: datetime.timedelta
+    required_sample_percentage: int
+    bias_last_sample: bool
+    bias_node_type: str
+    rebalance_preemption_percentage: Optional[int]
+    weekdays: Tuple[int, int]
+    work_hours: Tuple[int, int]
+
+
+@dataclasses.dataclass(frozen=True)
+class AutoscaleSettings:
+    enabled: bool
+    evaluation_interval: datetime.timedelta
+    formula: Optional[str]
+    scenario: Optional[AutoscaleScenarioSettings]
+
+
+@dataclasses.dataclass(frozen=True)
+class PoolSettings:
+    id: str
+    gpus_dedicated: int
+    gpus_low_priority: int
+    device_ids: Optional[List[int]]
+    max_tasks_per_gpu: int
+    cpu_slots: int
+    node_fill_type: str
+    resize_timeout: datetime.timedelta
+    inter_node_communication_enabled: bool
+    per_job_auto_scratch: bool
+    restart_slot_on_start_task_failed: bool
+    attempt_recovery_on_unusable: bool
+    block_until_all_global_resources_loaded: bool
+    rocm_arch: str
+    rocm_verify: bool
+    container_runtimes_install: List[str]
+    container_runtime_default: str
+    autoscale: AutoscaleSettings
+    environment_variables: Dict[str, str]
+    start_task_pre: List[str]
+    start_task_post: List[str]
+    prometheus_rocm_exporter: bool
+    prometheus_rocm_port: int
+    prometheus_rocm_interval: float
+
+
+def pool_settings(conf: Dict[str, Any]) -> PoolSettings:
+    """reference convoy/settings.py:1277 `pool_settings`"""
+    p = conf["pool_specification"]
+    scen = _get(p, "autoscale", "scenario")
+    scenario = None
+    if scen:
+        scenario = AutoscaleScenarioSettings(
+            name=scen["name"],
+            maximum_gpu_count_dedicated=_get(
+                scen, "maximum_gpu_count", "dedicated",
+                default=MAX_GPUS_PER_NODE),
+            maximum_gpu_count_low_priority=_get(
+                scen, "maximum_gpu_count", "low_priority", default=0),
+            maximum_increment_dedicated=_get(
+                scen, "maximum_increment_per_evaluation", "dedicated"),
+            maximum_increment_low_priority=_get(
+                scen, "maximum_increment_per_evaluation", "low_priority"),
+            node_deallocation_option=scen.get("node_deallocation_option",
+                                              "requeue"),
+            sample_lookback=utils.parse_timedelta(
+                scen.get("sample_lookback_interval", "00:10:00")),
+            required_sample_percentage=scen.get("required_sample_percentage",
+                                                70),
+            bias_last_sample=scen.get("bias_last_sample", True),
+            bias_node_type=scen.get("bias_node_type", "auto"),
+            rebalance_preemption_percentage=scen.get(
+                "rebalance_preemption_percentage"),
+            weekdays=(_get(scen, "time_ranges", "weekdays", "start",
+                           default=1),
+                      _get(scen, "time_ranges", "weekdays", "end", default=5)),
+            work_hours=(_get(scen, "time_ranges", "work_hours", "start",
+                             default=8),
+                        _get(scen, "time_ranges", "work_hours", "end",
+                             default=17)),
+        )
+    autoscale = AutoscaleSettings(
+        enabled=bool(_get(p, "autoscale", "scenario") or
+                     _get(p, "autoscale", "formula")),
+        evaluation_interval=utils.parse_timedelta(
+            _get(p, "autoscale", "evaluation_interval", default="00:15:00")),
+        formula=_get(p, "autoscale", "formula"),
+        scenario=scenario,
+    )
+    runtimes = _get(p, "node_configuration", "container_runtimes", "install",
+                    default=["process"])
+    if "process" not in runtimes:
+        runtimes = ["process"] + list(runtimes)
+    return PoolSettings(
+        id=p["id"],
+        gpus_dedicated=_get(p, "gpus", "dedicated", default=0),
+        gpus_low_priority=_get(p, "gpus", "low_priority", default=0),
+        device_ids=_get(p, "gpus", "device_ids"),
+        max_tasks_per_gpu=p.get("max_tasks_per_gpu", 1),
+        cpu_slots=p.get("cpu_slots", 0),
+        node_fill_type=p.get("node_fill_type", "pack"),
+        resize_timeout=utils.parse_timedelta(
+            p.get("resize_timeout", "00:20:00")),
+        inter_node_communication_enabled=p.get(
+            "inter_node_communication_enabled", False),
+        per_job_auto_scratch=p.get("per_job_auto_scratch", False),
+        restart_slot_on_start_task_failed=p.get(
+            "restart_slot_on_start_task_failed", False),
+        attempt_recovery_on_unusable=p.get("attempt_recovery_on_unusable",
+                                           False),
+        block_until_all_global_resources_loaded=p.get(
+            "block_until_all_global_resources_loaded", True),
+        rocm_arch=_get(p, "node_configuration", "rocm", "arch",
+                       default="gfx950"),
+        rocm_verify=_get(p, "node_configuration", "rocm", "verify",
+                         default=True),
+        container_runtimes_install=runtimes,
+        container_runtime_default=_get(
+            p, "node_configuration", "container_runtimes", "default",
+            default="process"),
+        autoscale=autoscale,
+        environment_variables=dict(p.get("environment_variables") or {}),
+        start_task_pre=list(_get(p, "start_task", "commands", "pre",
+                                 default=[])),
+        start_task_post=list(_get(p, "start_task", "commands", "post",
+                                  default=[])),
+        prometheus_rocm_exporter=_get(p, "prometheus", "rocm_exporter",
+                                      "enabled", default=False),
+        prometheus_rocm_port=_get(p, "prometheus", "rocm_exporter", "port",
+                                  default=9400),
+        prometheus_rocm_interval=_get(p, "prometheus", "rocm_exporter",
+                                      "interval_seconds", default=1.0),
+    )
+
+
+# --------------------------------------------------------------------
+# jobs / tasks
+# --------------------------------------------------------------------
+@dataclasses.dataclass(frozen=True)
+class ExitOptions:
+    job_action: str = "none"
+    dependency_action: str = "block"
+
+
+@dataclasses.dataclass(frozen=True)
+class GangSettings:
+    backend: str
+    gpus_per_rank: int
+    master_port: Optional[int]
+    rendezvous_timeout: datetime.timedelta
+
+
+@dataclasses.dataclass(frozen=True)
+class MultiInstanceSettings:
+    num_instances: Any  # int | 'pool_current_dedicated' | ...
+    coordination_command: Optional[str]
+    pre_execution_command: Optional[str]
+    resource_files: List[dict]
+    gang: GangSettings
+
+
+@dataclasses.dataclass(frozen=True)
+class DataSpec:
+    kind: str  # local_batch | local_storage
+    spec: dict
+
+
+@dataclasses.dataclass(frozen=True)
+class TaskSettings:
+    id: Optional[str]
+    image: Optional[str]
+    runtime: str  # process | docker | singularity
+    command: Optional[str]
+    entrypoint: Optional[str]
+    environment_variables: Dict[str, str]
+    gpus: int  # resolved count (0 == none)
+    exclusive_gpus: bool
+    shm_size: Optional[int]
+    data_volumes: List[str]
+    shared_data_volumes: List[str]
+    resource_files: List[dict]
+    input_data: List[DataSpec]
+    output_data: List[DataSpec]
+    depends_on: List[str]
+    depends_on_range: Optional[Tuple[int, int]]
+    max_task_retries: int
+    max_wall_time: Optional[datetime.timedelta]
+    retention_time: Optional[datetime.timedelta]
+    exit_options: ExitOptions
+    multi_instance: Optional[MultiInstanceSettings]
+    remove_container_after_exit: bool
+    additional_docker_run_options: List[str]
+    default_working_dir: str
+    task_factory: Optional[dict]
+    labels: List[str]
+
+
+@dataclasses.dataclass(frozen=True)
+class RecurrenceSettings:
+    interval: datetime.timedelta
+    do_not_run_until: Optional[str]
+    do_not_run_after: Optional[str]
+    start_window: Optional[datetime.timedelta]
+    monitor_task_completion: bool
+
+
+@dataclasses.dataclass(frozen=True)
+class JobSettings:
+    id: str
+    auto_complete: bool
+    environment_variables: Dict[str, str]
+    max_task_retries: int
+    max_wall_time: Optional[datetime.timedelta]
+    retention_time: Optional[datetime.timedelta]
+    priority: int
+    gpus_default: Optional[str]
+    exit_options: ExitOptions
+    job_preparation_command: Optional[str]
+    job_release_command: Optional[str]
+    input_data: List[DataSpec]
+    default_working_dir: str
+    autogen_task_id_prefix: str
+    autogen_task_id_zfill: int
+    force_enable_task_dependencies: bool
+    recurrence: Optional[RecurrenceSettings]
+    federation_constraints: Optional[dict]
+    allow_run_on_missing_image: bool
+    tasks: List[dict]  # raw task dicts (expanded by task_factory later)
+    merge_task: Optional[dict]
+
+
+def _exit_options(d: Optional[dict]) -> ExitOptions:
+    eo = _get(d, "default", "exit_options", default={})
+    return ExitOptions(job_action=eo.get("job_action", "none"),
+                       dependency_action=eo.get("dependency_action", "block"))
+
+
+def _data_specs(d: Optional[dict]) -> List[DataSpec]:
+    out: List[DataSpec] = []
+    if not d:
+        return out
+    for kind in ("local_batch", "local_storage"):
+        for spec in d.get(kind) or []:
+            out.append(DataSpec(kind=kind, spec=spec))
+    return out
+
+
+def job_settings(jobspec: Dict[str, Any]) -> JobSettings:
+    """reference convoy/settings.py job_* accessors (3420-3727)"""
+    rec = None
+    if jobspec.get("recurrence"):
+        r = jobspec["recurrence"]
+        rec = RecurrenceSettings(
+            interval=utils.parse_timedelta(
+                _get(r, "schedule", "recurrence_interval",
+                     default="00:05:00")),
+            do_not_run_until=_get(r, "schedule", "do_not_run_until"),
+            do_not_run_after=_get(r, "schedule", "do_not_run_after"),
+            start_window=utils.parse_timedelta(
+                _get(r, "schedule", "start_window")),
+            monitor_task_completion=_get(r, "job_manager",
+                                         "monitor_task_completion",
+                                         default=False),
+        )
+    return JobSettings(
+        id=jobspec["id"],
+        auto_complete=jobspec.get("auto_complete", False),
+        environment_variables=dict(jobspec.get("environment_variables")
+                                   or {}),
+        max_task_retries=jobspec.get("max_task_retries", 0),
+        max_wall_time=utils.parse_timedelta(jobspec.get("max_wall_time")),
+        retention_time=utils.parse_timedelta(jobspec.get("retention_time")),
+        priority=jobspec.get("priority", 0),
+        gpus_default=(str(jobspec["gpus"])
+                      if jobspec.get("gpus") is not None else None),
+        exit_options=_exit_options(jobspec.get("exit_conditions")),
+        job_preparation_command=_get(jobspec, "job_preparation", "command"),
+        job_release_command=_get(jobspec, "job_release", "command"),
+        input_data=_data_specs(jobspec.get("input_data")),
+        default_working_dir=jobspec.get("default_working_dir", "batch"),
+        autogen_task_id_prefix=_get(jobspec, "autogenerated_task_id",
+                                    "prefix", default="task-"),
+        autogen_task_id_zfill=_get(jobspec, "autogenerated_task_id",
+                                   "zfill_width", default=5),
+        force_enable_task_dependencies=jobspec.get(
+            "force_enable_task_dependencies", False),
+        recurrence=rec,
+        federation_constraints=jobspec.get("federation_constraints"),
+        allow_run_on_missing_image=jobspec.get("allow_run_on_missing_image",
+                                               False),
+        tasks=list(jobspec.get("tasks") or []),
+        merge_task=jobspec.get("merge_task"),
+    )
+
+
+def resolve_gpus(value, pool: Optional[PoolSettings]) -> int:
+    """'disable' -> 0, 'all' -> pool capacity, int -> int.
+    Analogue of reference settings.py:4239-4251 --gpus emission."""
+    if value is None:
+        return 0
+    s = str(value).strip().lower()
+    if s in ("disable", "none", "0", "false"):
+        return 0
+    if s == "all":
+        if pool is None:
+            return MAX_GPUS_PER_NODE
+        return max(pool.gpus_dedicated + pool.gpus_low_priority, 1)
+    n = int(s)
+    if n < 0 or n > MAX_GPUS_PER_NODE:
+        raise ValueError(f"gpus out of range: {n}")
+    return n
+
+
+def task_settings(taskspec: Dict[str, Any], job: JobSettings,
+                  pool: Optional[PoolSettings] = None) -> TaskSettings:
+    """Compile one task spec (reference convoy/settings.py:3727
+    `task_settings`).  GPU binding is expressed as a count here; the
+    runner maps it to HIP_VISIBLE_DEVICES + /dev/kfd + /dev/dri binds
+    (the --gpus/--nv analogue)."""
+    env = dict(job.environment_variables)
+    env.update(taskspec.get("environment_variables") or {})
+
+    gpus_raw = taskspec.get("gpus", job.gpus_default)
+    gpus = resolve_gpus(gpus_raw, pool)
+
+    mi = None
+    if taskspec.get("multi_instance"):
+        m = taskspec["multi_instance"]
+        g = m.get("gang") or {}
+        mi = MultiInstanceSettings(
+            num_instances=m.get("num_instances"),
+            coordination_command=m.get("coordination_command"),
+            pre_execution_command=m.get("pre_execution_command"),
+            resource_files=list(m.get("resource_files") or []),
+            gang=GangSettings(
+                backend=g.get("backend", "rccl"),
+                gpus_per_rank=g.get("gpus_per_rank", 1),
+                master_port=g.get("master_port"),
+                rendezvous_timeout=utils.parse_timedelta(
+                    g.get("rendezvous_timeout", "00:05:00")),
+            ),
+        )
+
+    image = taskspec.get("image") or taskspec.get("docker_image") \
+        or taskspec.get("singularity_image")
+    if taskspec.get("singularity_image") and not taskspec.get("image") \
+            and not taskspec.get("docker_image"):
+        runtime = "singularity"
+    elif taskspec.get("docker_image"):
+        runtime = "docker"
+    else:
+        runtime = pool.container_runtime_default if pool else "process"
+
+    dor = taskspec.get("depends_on_range")
+    depends_on_range = (int(dor[0]), int(dor[1])) if dor else None
+
+    return TaskSettings(
+        id=taskspec.get("id"),
+        image=image,
+        runtime=runtime,
+        command=taskspec.get("command"),
+        entrypoint=taskspec.get("entrypoint"),
+        environment_variables=env,
+        gpus=gpus,
+        exclusive_gpus=taskspec.get("exclusive_gpus", gpus > 0),
+        shm_size=utils.parse_size(taskspec.get("shm_size")),
+        data_volumes=list(taskspec.get("data_volumes") or []),
+        shared_data_volumes=list(taskspec.get("shared_data_volumes") or []),
+        resource_files=list(taskspec.get("resource_files") or []),
+        input_data=_data_specs(taskspec.get("input_data")),
+        output_data=_data_specs(taskspec.get("output_data")),
+        depends_on=list(taskspec.get("depends_on") or []),
+        depends_on_range=depends_on_range,
+        max_task_retries=taskspec.get("max_task_retries",
+                                      job.max_task_retries),
+        max_wall_time=utils.parse_timedelta(taskspec.get("max_wall_time"))
+        or job.max_wall_time,
+        retention_time=utils.parse_timedelta(taskspec.get("retention_time"))
+        or job.retention_time,
+        exit_options=_exit_options(taskspec.get("exit_conditions"))
+        if taskspec.get("exit_conditions") else job.exit_options,
+        multi_instance=mi,
+        remove_container_after_exit=taskspec.get(
+            "remove_container_after_exit", True),
+        additional_docker_run_options=list(
+            taskspec.get("additional_docker_run_options") or []),
+        default_working_dir=taskspec.get("default_working_dir",
+                                         job.default_working_dir),
+        task_factory=taskspec.get("task_factory"),
+        labels=list(taskspec.get("labels") or []),
+    )
+
+
+# --------------------------------------------------------------------
+# global config / storage
+# --------------------------------------------------------------------
+@dataclasses.dataclass(frozen=True)
+class GlobalSettings:
+    storage_account: str
+    storage_entity_prefix: str
+    autogen_task_id_prefix: str
+    autogen_task_id_zfill: int
+    concurrent_source_downloads: int
+    docker_images: List[str]
+    singularity_images: List[str]
+    local_images: List[dict]
+    data_volumes: Dict[str, dict]
+    shared_data_volumes: Dict[str, dict]
+    files: List[dict]
+    fallback_registry: Optional[str]
+    delay_image_preload: bool
+
+
+def global_settings(conf: Dict[str, Any]) -> GlobalSettings:
+    bs = conf.get("batch_shipyard") or {}
+    gr = conf.get("global_resources") or {}
+    sing: List[str] = []
+    si = gr.get("singularity_images") or {}
+    for key in ("unsigned", "signed"):
+        for ent in si.get(key) or []:
+            sing.append(ent["image"])
+    return GlobalSettings(
+        storage_account=bs.get("storage_account_settings", "default"),
+        storage_entity_prefix=bs.get("storage_entity_prefix", "shipyard"),
+        autogen_task_id_prefix=_get(bs, "autogenerated_task_id", "prefix",
+                                    default="task-"),
+        autogen_task_id_zfill=_get(bs, "autogenerated_task_id",
+                                   "zfill_width", default=5),
+        concurrent_source_downloads=_get(conf, "data_replication",
+                                         "concurrent_source_downloads",
+                                         default=4) or 4,
+        docker_images=list(gr.get("docker_images") or []),
+        singularity_images=sing,
+        local_images=list(gr.get("local_images") or []),
+        data_volumes=dict(_get(gr, "volumes", "data_volumes", default={})),
+        shared_data_volumes=dict(_get(gr, "volumes", "shared_data_volumes",
+                                      default={})),
+        files=list(gr.get("files") or []),
+        fallback_registry=bs.get("fallback_registry"),
+        delay_image_preload=bs.get("delay_image_preload", False),
+    )
+
+
+@dataclasses.dataclass(frozen=True)
+class StorageAccountSettings:
+    name: str
+    root: str
+    create: bool
+
+
+def storage_account_settings(creds: Dict[str, Any],
+                             name: str) -> StorageAccountSettings:
+    sa = _get(creds, "credentials", "storage", name)
+    if sa is None:
+        raise KeyError(f"storage account settings not found: {name}")
+    return StorageAccountSettings(name=name, root=sa["root"],
+                                  create=sa.get("create", True))

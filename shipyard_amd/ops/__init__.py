@@ -153,6 +153,14 @@ def lz4_decode_blocks(comp, in_off, in_len, out, out_off, out_len):
     return status
 
 
+def lz4_all_ok(status) -> bool:
+    """True iff every block decoded cleanly (uint32 reductions are not
+    implemented on CUDA in torch, so check on host)."""
+    import torch
+
+    return bool((status.cpu().to(torch.int64) == 0).all().item())
+
+
 def sha256_pages(data, page_size: int = 4096):
     """SHA-256 digest of each page of a uint8 CUDA tensor -> (n_pages, 32)."""
     import torch
