@@ -1,0 +1,751 @@
+"""LocalExecutor — the MI355X-node batch service.
+
+This is the framework's replacement for the Azure Batch service + the
+reference's convoy/batch.py client ops (reference convoy/batch.py:921
+`create_pool`, 5056 `add_jobs`, 4489 `_construct_task`, 4313
+`_add_task_collection`): pools are partitions of the node's GPUs, jobs
+and tasks live in the SQLite store, and a scheduler loop assigns ready
+tasks to free GPU slots and launches them through the runner (process /
+docker / singularity with the ROCm binder; multi-instance tasks
+gang-launch RCCL ranks).
+
+State machines:
+  pool: resizing -> active -> (deleting)
+  job:  active | disabled | completed | terminated | deleted
+  task: pending -> ready -> running -> completed | failed
+        (failed + retries left -> ready;  dep failed + block -> blocked)
+"""
+from __future__ import annotations
+
+import json
+import time
+from pathlib import Path
+from typing import Any, Dict, List, Optional, Sequence, Tuple
+
+from shipyard_amd import utils
+from shipyard_amd.config import settings as cfg
+from shipyard_amd.executor import task_factory
+from shipyard_amd.executor.store import Store
+from shipyard_amd.runner.task_runner import LaunchSpec, TaskHandle, launch
+
+logger = utils.get_logger(__name__)
+
+JOBPREP_TASK_ID = "shipyard-jobprep"
+MERGE_TASK_PREFIX = "merge-task"
+
+
+class ExecutorError(RuntimeError):
+    pass
+
+
+class LocalExecutor:
+    def __init__(self, root, detect_gpus: bool = True):
+        self.root = Path(root)
+        self.root.mkdir(parents=True, exist_ok=True)
+        self.store = Store(self.root / "store.db")
+        self._handles: Dict[Tuple[str, str], TaskHandle] = {}
+        self._detect_gpus = detect_gpus
+        self._n_host_gpus: Optional[int] = None
+
+    # ----------------------------------------------------------------
+    # host inventory
+    # ----------------------------------------------------------------
+    def host_gpu_count(self) -> int:
+        if self._n_host_gpus is None:
+            n = 0
+            if self._detect_gpus:
+                try:
+                    import torch
+
+                    if torch.cuda.is_available():
+                        n = torch.cuda.device_count()
+                except Exception:
+                    n = 0
+            self._n_host_gpus = n
+        return self._n_host_gpus
+
+    # ----------------------------------------------------------------
+    # pools (reference convoy/batch.py:921 create_pool,
+    #        convoy/fleet.py:1821 _add_pool)
+    # ----------------------------------------------------------------
+    def pool_add(self, pool_conf: Dict[str, Any],
+                 wait_ready: bool = True) -> cfg.PoolSettings:
+        ps = cfg.pool_settings(pool_conf)
+        if self.store.query_one("SELECT id FROM pools WHERE id=?", (ps.id,)):
+            raise ExecutorError(f"pool {ps.id} exists")
+        self.store.add_perf(f"pool:{ps.id}", "npstart")
+        total_gpus = ps.gpus_dedicated + ps.gpus_low_priority
+        host = self.host_gpu_count()
+        device_ids = ps.device_ids
+        if device_ids is None:
+            device_ids = list(range(total_gpus))
+        if len(device_ids) < total_gpus:
+            raise ExecutorError("device_ids shorter than gpu slot count")
+        # oversubscription guard (the analogue of the reference's
+        # vm_size capability checks, settings.py:4231-4289)
+        if host and total_gpus > host * ps.max_tasks_per_gpu:
+            raise ExecutorError(
+                f"pool {ps.id} wants {total_gpus} gpu slots but host has "
+                f"{host} GPUs x {ps.max_tasks_per_gpu} tasks/gpu")
+        self.store.execute(
+            "INSERT INTO pools (id, spec_json, state, created_at, "
+            "gpus_dedicated, gpus_low_priority, max_tasks_per_gpu, cpu_slots)"
+            " VALUES (?,?,?,?,?,?,?,?)",
+            (ps.id, json.dumps(pool_conf), "resizing", time.time(),
+             ps.gpus_dedicated, ps.gpus_low_priority, ps.max_tasks_per_gpu,
+             ps.cpu_slots))
+        self._create_slots(ps, device_ids)
+        # nodeprep-analogue: rocm verify + start task commands
+        if wait_ready:
+            self.wait_for_pool_ready(ps.id)
+        return ps
+
+    def _create_slots(self, ps: cfg.PoolSettings,
+                      device_ids: Sequence[int]) -> None:
+        rows = []
+        slot = 0
+        # one slot per (gpu, task-slot) pair; dedicated slots first
+        for i in range(ps.gpus_dedicated):
+            for rep in range(ps.max_tasks_per_gpu):
+                rows.append((ps.id, slot, "gpu", device_ids[i], 1, "idle"))
+                slot += 1
+        for i in range(ps.gpus_dedicated,
+                       ps.gpus_dedicated + ps.gpus_low_priority):
+            for rep in range(ps.max_tasks_per_gpu):
+                rows.append((ps.id, slot, "gpu", device_ids[i], 0, "idle"))
+                slot += 1
+        for _ in range(ps.cpu_slots):
+            rows.append((ps.id, slot, "cpu", None, 1, "idle"))
+            slot += 1
+        self.store.executemany(
+            "INSERT INTO slots (pool_id, slot_id, kind, device_id, "
+            "dedicated, state) VALUES (?,?,?,?,?,?)", rows)
+
+    def wait_for_pool_ready(self, pool_id: str,
+                            timeout: float = 300.0) -> None:
+        """Node-ready state machine, local edition (reference
+        convoy/batch.py:625 _block_for_nodes_ready): verify ROCm when
+        requested, run start-task commands, mark active."""
+        row = self.store.query_one("SELECT * FROM pools WHERE id=?",
+                                   (pool_id,))
+        if row is None:
+            raise ExecutorError(f"no pool {pool_id}")
+        conf = json.loads(row["spec_json"])
+        ps = cfg.pool_settings(conf)
+        t0 = time.time()
+        try:
+            if ps.rocm_verify and (ps.gpus_dedicated + ps.gpus_low_priority):
+                self._verify_rocm(ps)
+            for cmd in ps.start_task_pre + ps.start_task_post:
+                rc, out, err = utils.subprocess_with_output(
+                    ["/bin/bash", "-c", cmd], timeout=timeout)
+                if rc != 0:
+                    raise ExecutorError(
+                        f"start task failed ({cmd!r}): {err.strip()}")
+        except ExecutorError:
+            if ps.restart_slot_on_start_task_failed:
+                logger.warning("start task failed; retrying once")
+                for cmd in ps.start_task_pre + ps.start_task_post:
+                    utils.subprocess_with_output(["/bin/bash", "-c", cmd])
+            else:
+                self.store.execute(
+                    "UPDATE pools SET state='starttaskfailed' WHERE id=?",
+                    (pool_id,))
+                raise
+        self.store.execute("UPDATE pools SET state='active' WHERE id=?",
+                           (pool_id,))
+        self.store.add_perf(f"pool:{pool_id}", "npend",
+                            {"elapsed_s": time.time() - t0})
+
+    def _verify_rocm(self, ps: cfg.PoolSettings) -> None:
+        host = self.host_gpu_count()
+        need = ps.gpus_dedicated + ps.gpus_low_priority
+        if host == 0:
+            raise ExecutorError(
+                f"pool {ps.id} requires {need} GPUs but none are visible "
+                "(set node_configuration.rocm.verify: false for CPU runs)")
+
+    def pool_list(self) -> List[dict]:
+        return [dict(r) for r in self.store.query(
+            "SELECT id, state, gpus_dedicated, gpus_low_priority, "
+            "max_tasks_per_gpu, cpu_slots, created_at FROM pools")]
+
+    def pool_del(self, pool_id: str, force: bool = False) -> None:
+        running = self.store.query(
+            "SELECT t.job_id, t.id FROM tasks t JOIN jobs j "
+            "ON t.job_id = j.id WHERE j.pool_id=? AND t.state='running'",
+            (pool_id,))
+        if running and not force:
+            raise ExecutorError(f"pool {pool_id} has running tasks")
+        for r in running:
+            h = self._handles.pop((r["job_id"], r["id"]), None)
+            if h:
+                h.kill()
+        with self.store.transaction() as conn:
+            conn.execute("DELETE FROM slots WHERE pool_id=?", (pool_id,))
+            conn.execute(
+                "DELETE FROM tasks WHERE job_id IN "
+                "(SELECT id FROM jobs WHERE pool_id=?)", (pool_id,))
+            conn.execute("DELETE FROM jobs WHERE pool_id=?", (pool_id,))
+            conn.execute("DELETE FROM pools WHERE id=?", (pool_id,))
+
+    def pool_resize(self, pool_id: str, dedicated: Optional[int] = None,
+                    low_priority: Optional[int] = None) -> None:
+        """Resize GPU slot counts (reference convoy/batch.py:1372
+        resize_pool).  Shrinking only reclaims idle slots."""
+        row = self.store.query_one("SELECT * FROM pools WHERE id=?",
+                                   (pool_id,))
+        if row is None:
+            raise ExecutorError(f"no pool {pool_id}")
+        conf = json.loads(row["spec_json"])
+        ps = cfg.pool_settings(conf)
+        new_ded = dedicated if dedicated is not None else ps.gpus_dedicated
+        new_low = (low_priority if low_priority is not None
+                   else ps.gpus_low_priority)
+        host = self.host_gpu_count()
+        if host and new_ded + new_low > host * ps.max_tasks_per_gpu:
+            raise ExecutorError("resize exceeds host GPU capacity")
+        conf["pool_specification"]["gpus"]["dedicated"] = new_ded
+        conf["pool_specification"]["gpus"]["low_priority"] = new_low
+        with self.store.transaction() as conn:
+            conn.execute("DELETE FROM slots WHERE pool_id=? "
+                         "AND state='idle'", (pool_id,))
+            conn.execute(
+                "UPDATE pools SET spec_json=?, gpus_dedicated=?, "
+                "gpus_low_priority=? WHERE id=?",
+                (json.dumps(conf), new_ded, new_low, pool_id))
+        ps2 = cfg.pool_settings(conf)
+        device_ids = ps2.device_ids or list(range(new_ded + new_low))
+        # recreate idle slots (busy ones keep their ids)
+        busy = {r["slot_id"] for r in self.store.query(
+            "SELECT slot_id FROM slots WHERE pool_id=?", (pool_id,))}
+        rows = []
+        slot = (max(busy) + 1) if busy else 0
+        busy_devs = [r["device_id"] for r in self.store.query(
+            "SELECT device_id FROM slots WHERE pool_id=? AND kind='gpu'",
+            (pool_id,))]
+        for i in range(new_ded + new_low):
+            dev = device_ids[i]
+            count_existing = busy_devs.count(dev)
+            for _ in range(ps2.max_tasks_per_gpu - count_existing):
+                rows.append((pool_id, slot, "gpu", dev,
+                             1 if i < new_ded else 0, "idle"))
+                slot += 1
+        existing_cpu = len(self.store.query(
+            "SELECT slot_id FROM slots WHERE pool_id=? AND kind='cpu'",
+            (pool_id,)))
+        for _ in range(ps2.cpu_slots - existing_cpu):
+            rows.append((pool_id, slot, "cpu", None, 1, "idle"))
+            slot += 1
+        if rows:
+            self.store.executemany(
+                "INSERT INTO slots (pool_id, slot_id, kind, device_id, "
+                "dedicated, state) VALUES (?,?,?,?,?,?)", rows)
+
+    def pool_stats(self, pool_id: str) -> dict:
+        """reference convoy/batch.py:1460 pool_stats"""
+        slots = self.store.query(
+            "SELECT state, COUNT(*) n FROM slots WHERE pool_id=? "
+            "GROUP BY state", (pool_id,))
+        tasks = self.store.query(
+            "SELECT t.state, COUNT(*) n FROM tasks t JOIN jobs j ON "
+            "t.job_id=j.id WHERE j.pool_id=? GROUP BY t.state", (pool_id,))
+        return {
+            "pool_id": pool_id,
+            "slots": {r["state"]: r["n"] for r in slots},
+            "tasks": {r["state"]: r["n"] for r in tasks},
+        }
+
+    def _pool_settings(self, pool_id: str) -> cfg.PoolSettings:
+        row = self.store.query_one("SELECT spec_json FROM pools WHERE id=?",
+                                   (pool_id,))
+        if row is None:
+            raise ExecutorError(f"no pool {pool_id}")
+        return cfg.pool_settings(json.loads(row["spec_json"]))
+
+    def pool_root(self, pool_id: str) -> Path:
+        return self.root / "pools" / pool_id
+
+    # ----------------------------------------------------------------
+    # jobs (reference convoy/batch.py:5056 add_jobs)
+    # ----------------------------------------------------------------
+    def jobs_add(self, jobs_conf: Dict[str, Any], pool_id: str) -> List[str]:
+        ps = self._pool_settings(pool_id)
+        added = []
+        for jobspec in jobs_conf["job_specifications"]:
+            js = cfg.job_settings(jobspec)
+            if self.store.query_one("SELECT id FROM jobs WHERE id=?",
+                                    (js.id,)):
+                raise ExecutorError(f"job {js.id} exists")
+            self.store.execute(
+                "INSERT INTO jobs (id, pool_id, spec_json, state, priority,"
+                " auto_complete, created_at) VALUES (?,?,?,?,?,?,?)",
+                (js.id, pool_id, json.dumps(jobspec), "active", js.priority,
+                 int(js.auto_complete), time.time()))
+            self._add_tasks_for_job(js, jobspec, ps)
+            added.append(js.id)
+            self.store.add_event(f"job:{js.id}", "submitted",
+                                 {"pool": pool_id})
+        return added
+
+    def _autogen_id(self, js: cfg.JobSettings, seq: int) -> str:
+        return f"{js.autogen_task_id_prefix}" \
+               f"{str(seq).zfill(js.autogen_task_id_zfill)}"
+
+    def _add_tasks_for_job(self, js: cfg.JobSettings, jobspec: dict,
+                           ps: cfg.PoolSettings) -> None:
+        """Expand task factories, assign autogenerated ids, wire
+        dependencies + merge task (reference convoy/batch.py:5160-5713)."""
+        seq = 0
+        all_ids: List[str] = []
+        rows = []
+        dep_rows = []
+        now = time.time()
+
+        def compile_one(taskspec: dict) -> Tuple[str, dict, List[str]]:
+            nonlocal seq
+            ts = cfg.task_settings(taskspec, js, ps)
+            tid = ts.id or self._autogen_id(js, seq)
+            deps = list(ts.depends_on)
+            if ts.depends_on_range:
+                lo, hi = ts.depends_on_range
+                deps += [str(i) for i in range(lo, hi + 1)]
+            compiled = dict(taskspec)
+            compiled["id"] = tid
+            return tid, compiled, deps
+
+        if js.job_preparation_command:
+            rows.append((js.id, JOBPREP_TASK_ID,
+                         json.dumps({"id": JOBPREP_TASK_ID,
+                                     "command": js.job_preparation_command,
+                                     "gpus": "disable"}),
+                         "pending", now, seq))
+            seq += 1
+
+        for taskspec in js.tasks:
+            if taskspec.get("task_factory"):
+                expanded = task_factory.generate_tasks(taskspec)
+            else:
+                expanded = [taskspec]
+            for spec1 in expanded:
+                tid, compiled, deps = compile_one(spec1)
+                if any(tid == r[1] for r in rows):
+                    raise ExecutorError(f"duplicate task id {tid}")
+                rows.append((js.id, tid, json.dumps(compiled), "pending",
+                             now, seq))
+                seq += 1
+                all_ids.append(tid)
+                for d in deps:
+                    dep_rows.append((js.id, tid, d))
+                if js.job_preparation_command:
+                    dep_rows.append((js.id, tid, JOBPREP_TASK_ID))
+
+        if js.merge_task is not None:
+            mt = dict(js.merge_task)
+            mtid = mt.get("id") or f"{MERGE_TASK_PREFIX}-{len(all_ids):05d}"
+            mt["id"] = mtid
+            rows.append((js.id, mtid, json.dumps(mt), "pending", now, seq))
+            seq += 1
+            for d in all_ids:
+                dep_rows.append((js.id, mtid, d))
+
+        self.store.executemany(
+            "INSERT INTO tasks (job_id, id, spec_json, state, submit_time,"
+            " seq) VALUES (?,?,?,?,?,?)", rows)
+        if dep_rows:
+            self.store.executemany(
+                "INSERT OR IGNORE INTO task_deps (job_id, task_id, "
+                "depends_on) VALUES (?,?,?)", dep_rows)
+
+    def jobs_list(self) -> List[dict]:
+        return [dict(r) for r in self.store.query(
+            "SELECT id, pool_id, state, priority, created_at FROM jobs")]
+
+    def tasks_list(self, job_id: str) -> List[dict]:
+        return [dict(r) for r in self.store.query(
+            "SELECT id, state, exit_code, retries, start_time, end_time "
+            "FROM tasks WHERE job_id=? ORDER BY seq", (job_id,))]
+
+    def job_disable(self, job_id: str) -> None:
+        self.store.execute(
+            "UPDATE jobs SET state='disabled' WHERE id=? AND "
+            "state='active'", (job_id,))
+
+    def job_enable(self, job_id: str) -> None:
+        self.store.execute(
+            "UPDATE jobs SET state='active' WHERE id=? AND "
+            "state='disabled'", (job_id,))
+
+    def job_terminate(self, job_id: str, wait: bool = True) -> None:
+        for (jid, tid), h in list(self._handles.items()):
+            if jid == job_id:
+                h.kill()
+        with self.store.transaction() as conn:
+            conn.execute(
+                "UPDATE tasks SET state='cancelled' WHERE job_id=? AND "
+                "state IN ('pending','ready','blocked')", (job_id,))
+            conn.execute(
+                "UPDATE jobs SET state='terminated', completed_at=? "
+                "WHERE id=?", (time.time(), job_id))
+
+    def job_del(self, job_id: str) -> None:
+        self.job_terminate(job_id)
+        with self.store.transaction() as conn:
+            conn.execute("DELETE FROM tasks WHERE job_id=?", (job_id,))
+            conn.execute("DELETE FROM task_deps WHERE job_id=?", (job_id,))
+            conn.execute("DELETE FROM jobs WHERE id=?", (job_id,))
+
+    def job_stats(self, job_id: Optional[str] = None) -> dict:
+        """reference convoy/batch.py:1972 job_stats"""
+        where, params = ("WHERE job_id=?", (job_id,)) if job_id else ("", ())
+        rows = self.store.query(
+            f"SELECT state, COUNT(*) n, AVG(end_time - start_time) avg_s "
+            f"FROM tasks {where} GROUP BY state", params)
+        out = {"tasks": {r["state"]: r["n"] for r in rows}}
+        durs = self.store.query(
+            f"SELECT start_time, end_time, submit_time FROM tasks {where}",
+            params)
+        run = [r["end_time"] - r["start_time"] for r in durs
+               if r["end_time"] and r["start_time"]]
+        wait = [r["start_time"] - r["submit_time"] for r in durs
+                if r["start_time"]]
+        if run:
+            out["run_time_s"] = {"mean": sum(run) / len(run),
+                                 "max": max(run), "min": min(run)}
+        if wait:
+            out["wait_time_s"] = {"mean": sum(wait) / len(wait),
+                                  "max": max(wait), "min": min(wait)}
+        return out
+
+    # ----------------------------------------------------------------
+    # scheduler
+    # ----------------------------------------------------------------
+    def schedule_once(self) -> int:
+        """One scheduling pass: collect finished tasks, promote
+        dependency-satisfied tasks, assign ready tasks to idle slots,
+        launch.  Returns number of state transitions made."""
+        n = 0
+        n += self._collect_finished()
+        n += self._promote_pending()
+        n += self._assign_and_launch()
+        self._complete_auto_jobs()
+        return n
+
+    def run_until_idle(self, timeout: Optional[float] = None,
+                       poll: float = 0.02) -> None:
+        deadline = None if timeout is None else time.monotonic() + timeout
+        while True:
+            self.schedule_once()
+            busy = self.store.query_one(
+                "SELECT COUNT(*) n FROM tasks t JOIN jobs j ON t.job_id=j.id"
+                " WHERE t.state IN ('pending','ready','running') "
+                "AND j.state='active'")
+            if busy["n"] == 0:
+                return
+            if deadline is not None and time.monotonic() > deadline:
+                raise TimeoutError("executor did not go idle")
+            time.sleep(poll)
+
+    def _collect_finished(self) -> int:
+        n = 0
+        for (jid, tid), h in list(self._handles.items()):
+            rc = h.poll()
+            if rc is None:
+                continue
+            del self._handles[(jid, tid)]
+            self._finish_task(jid, tid, rc, h)
+            n += 1
+        return n
+
+    def _finish_task(self, jid: str, tid: str, rc: int,
+                     h: TaskHandle) -> None:
+        row = self.store.query_one(
+            "SELECT * FROM tasks WHERE job_id=? AND id=?", (jid, tid))
+        spec = json.loads(row["spec_json"])
+        slots = json.loads(row["slots_json"] or "[]")
+        with self.store.transaction() as conn:
+            for slot_id in slots:
+                conn.execute(
+                    "UPDATE slots SET state='idle', task_ref=NULL WHERE "
+                    "pool_id=(SELECT pool_id FROM jobs WHERE id=?) AND "
+                    "slot_id=?", (jid, slot_id))
+        js = self._job_settings(jid)
+        ps = self._pool_settings(self._job_pool(jid))
+        ts = cfg.task_settings(spec, js, ps)
+        if rc == 0:
+            self.store.execute(
+                "UPDATE tasks SET state='completed', exit_code=0, "
+                "end_time=? WHERE job_id=? AND id=?",
+                (time.time(), jid, tid))
+            self.store.add_event(f"task:{jid}/{tid}", "completed")
+            return
+        retries = row["retries"]
+        max_retries = ts.max_task_retries
+        if max_retries == -1 or retries < max_retries:
+            self.store.execute(
+                "UPDATE tasks SET state='ready', retries=? "
+                "WHERE job_id=? AND id=?", (retries + 1, jid, tid))
+            self.store.add_event(f"task:{jid}/{tid}", "retry",
+                                 {"rc": rc, "attempt": retries + 1})
+            return
+        self.store.execute(
+            "UPDATE tasks SET state='failed', exit_code=?, end_time=? "
+            "WHERE job_id=? AND id=?", (rc, time.time(), jid, tid))
+        self.store.add_event(f"task:{jid}/{tid}", "failed",
+                             {"rc": rc, "timed_out": h.timed_out})
+        # exit conditions (reference convoy/batch.py:4858-4929)
+        eo = ts.exit_options
+        if eo.job_action == "terminate":
+            self.job_terminate(jid)
+        elif eo.job_action == "disable":
+            self.job_disable(jid)
+        if eo.dependency_action == "block":
+            self._block_dependents(jid, tid)
+        else:  # satisfy: dependents may proceed as if completed
+            pass
+
+    def _block_dependents(self, jid: str, tid: str) -> None:
+        """Transitively block tasks that depend on a failed task."""
+        frontier = [tid]
+        while frontier:
+            cur = frontier.pop()
+            rows = self.store.query(
+                "SELECT task_id FROM task_deps WHERE job_id=? AND "
+                "depends_on=?", (jid, cur))
+            for r in rows:
+                dep = r["task_id"]
+                updated = self.store.execute(
+                    "UPDATE tasks SET state='blocked' WHERE job_id=? AND "
+                    "id=? AND state IN ('pending','ready')",
+                    (jid, dep)).rowcount
+                if updated:
+                    frontier.append(dep)
+
+    def _promote_pending(self) -> int:
+        """pending -> ready when all dependencies are satisfied."""
+        rows = self.store.query(
+            "SELECT t.job_id, t.id FROM tasks t JOIN jobs j ON "
+            "t.job_id=j.id WHERE t.state='pending' AND j.state='active'")
+        n = 0
+        for r in rows:
+            jid, tid = r["job_id"], r["id"]
+            deps = self.store.query(
+                "SELECT depends_on FROM task_deps WHERE job_id=? AND "
+                "task_id=?", (jid, tid))
+            ok = True
+            for d in deps:
+                dep_row = self.store.query_one(
+                    "SELECT state, spec_json FROM tasks WHERE job_id=? AND "
+                    "id=?", (jid, d["depends_on"]))
+                if dep_row is None:
+                    ok = False  # dangling dep: stays pending
+                    break
+                st = dep_row["state"]
+                if st == "completed":
+                    continue
+                if st == "failed":
+                    # satisfy-on-failure handled in _finish_task via not
+                    # blocking; treat failed+satisfy as satisfied
+                    spec = json.loads(dep_row["spec_json"])
+                    eo = spec.get("exit_conditions") or {}
+                    da = (((eo.get("default") or {}).get("exit_options")
+                           or {}).get("dependency_action", "block"))
+                    if da == "satisfy":
+                        continue
+                    ok = False
+                    break
+                ok = False
+                break
+            if ok:
+                self.store.execute(
+                    "UPDATE tasks SET state='ready' WHERE job_id=? AND "
+                    "id=? AND state='pending'", (jid, tid))
+                n += 1
+        return n
+
+    def _job_settings(self, jid: str) -> cfg.JobSettings:
+        row = self.store.query_one("SELECT spec_json FROM jobs WHERE id=?",
+                                   (jid,))
+        return cfg.job_settings(json.loads(row["spec_json"]))
+
+    def _job_pool(self, jid: str) -> str:
+        row = self.store.query_one("SELECT pool_id FROM jobs WHERE id=?",
+                                   (jid,))
+        return row["pool_id"]
+
+    def _resolve_num_instances(self, value, ps: cfg.PoolSettings) -> int:
+        if isinstance(value, int):
+            return value
+        s = str(value)
+        if s in ("pool_current_dedicated", "pool_specification_vm_count_"
+                 "dedicated"):
+            return max(ps.gpus_dedicated, 1)
+        if s in ("pool_current_low_priority",
+                 "pool_specification_vm_count_low_priority"):
+            return max(ps.gpus_low_priority, 1)
+        return int(s)
+
+    def _assign_and_launch(self) -> int:
+        ready = self.store.query(
+            "SELECT t.job_id, t.id, t.spec_json, j.pool_id, j.priority "
+            "FROM tasks t JOIN jobs j ON t.job_id=j.id "
+            "WHERE t.state='ready' AND j.state='active' "
+            "ORDER BY j.priority DESC, t.submit_time ASC, t.seq ASC")
+        n = 0
+        for r in ready:
+            jid, tid, pool_id = r["job_id"], r["id"], r["pool_id"]
+            spec = json.loads(r["spec_json"])
+            ps = self._pool_settings(pool_id)
+            js = self._job_settings(jid)
+            ts = cfg.task_settings(spec, js, ps)
+
+            if ts.multi_instance is not None:
+                ranks = self._resolve_num_instances(
+                    ts.multi_instance.num_instances, ps)
+                gpus_needed = ranks * ts.multi_instance.gang.gpus_per_rank
+            else:
+                ranks = 1
+                gpus_needed = ts.gpus
+
+            slots = self._try_allocate(pool_id, gpus_needed,
+                                       cpu_ok=(gpus_needed == 0))
+            if slots is None:
+                continue
+            slot_ids = [s["slot_id"] for s in slots]
+            device_ids = [s["device_id"] for s in slots
+                          if s["kind"] == "gpu"]
+            self.store.execute(
+                "UPDATE tasks SET state='running', start_time=?, "
+                "slots_json=? WHERE job_id=? AND id=?",
+                (time.time(), json.dumps(slot_ids), jid, tid))
+            try:
+                handle = self._launch_task(ps, js, ts, jid, tid, device_ids,
+                                           ranks)
+            except Exception as exc:  # launch failure = task failure
+                logger.error("launch failed for %s/%s: %s", jid, tid, exc)
+                self.store.execute(
+                    "UPDATE tasks SET state='failed', exit_code=-1, "
+                    "end_time=? WHERE job_id=? AND id=?",
+                    (time.time(), jid, tid))
+                self._release_slots(pool_id, slot_ids)
+                continue
+            self._handles[(jid, tid)] = handle
+            self.store.add_event(f"task:{jid}/{tid}", "launched",
+                                 {"slots": slot_ids, "devices": device_ids})
+            n += 1
+        return n
+
+    def _try_allocate(self, pool_id: str, gpus: int,
+                      cpu_ok: bool) -> Optional[List[dict]]:
+        with self.store.transaction() as conn:
+            if gpus > 0:
+                rows = list(conn.execute(
+                    "SELECT slot_id, kind, device_id FROM slots WHERE "
+                    "pool_id=? AND state='idle' AND kind='gpu' "
+                    "ORDER BY dedicated DESC, device_id ASC LIMIT ?",
+                    (pool_id, gpus)))
+                if len(rows) < gpus:
+                    return None
+                # prefer distinct devices for multi-gpu tasks
+                if gpus > 1:
+                    seen = set()
+                    distinct = []
+                    for row in conn.execute(
+                            "SELECT slot_id, kind, device_id FROM slots "
+                            "WHERE pool_id=? AND state='idle' AND "
+                            "kind='gpu' ORDER BY dedicated DESC, "
+                            "device_id ASC", (pool_id,)):
+                        if row["device_id"] not in seen:
+                            seen.add(row["device_id"])
+                            distinct.append(row)
+                    if len(distinct) >= gpus:
+                        rows = distinct[:gpus]
+            else:
+                rows = list(conn.execute(
+                    "SELECT slot_id, kind, device_id FROM slots WHERE "
+                    "pool_id=? AND state='idle' "
+                    "ORDER BY kind='gpu', slot_id LIMIT 1", (pool_id,)))
+                if not rows:
+                    return None
+                rows = rows[:1]
+            for row in rows:
+                conn.execute(
+                    "UPDATE slots SET state='busy' WHERE pool_id=? AND "
+                    "slot_id=?", (pool_id, row["slot_id"]))
+            return [dict(row) for row in rows]
+
+    def _release_slots(self, pool_id: str, slot_ids: List[int]) -> None:
+        for sid in slot_ids:
+            self.store.execute(
+                "UPDATE slots SET state='idle', task_ref=NULL WHERE "
+                "pool_id=? AND slot_id=?", (pool_id, sid))
+
+    def _launch_task(self, ps: cfg.PoolSettings, js: cfg.JobSettings,
+                     ts: cfg.TaskSettings, jid: str, tid: str,
+                     device_ids: List[int], ranks: int) -> TaskHandle:
+        mi = ts.multi_instance
+        env = dict(ps.environment_variables)
+        env.update(ts.environment_variables)
+        spec = LaunchSpec(
+            pool_id=ps.id,
+            job_id=jid,
+            task_id=tid,
+            command=ts.command or "",
+            runtime=ts.runtime if ts.image else "process",
+            image=ts.image,
+            entrypoint=ts.entrypoint,
+            env=env,
+            device_ids=device_ids,
+            shm_size=ts.shm_size,
+            docker_options=ts.additional_docker_run_options,
+            remove_container=ts.remove_container_after_exit,
+            num_instances=ranks if mi else 1,
+            gang_backend=mi.gang.backend if mi else "rccl",
+            gpus_per_rank=mi.gang.gpus_per_rank if mi else 1,
+            master_port=mi.gang.master_port if mi else None,
+            pre_execution_command=mi.pre_execution_command if mi else None,
+            max_wall_time_s=(ts.max_wall_time.total_seconds()
+                             if ts.max_wall_time else None),
+        )
+        return launch(spec, self.pool_root(ps.id))
+
+    def _complete_auto_jobs(self) -> None:
+        rows = self.store.query(
+            "SELECT id FROM jobs WHERE state='active' AND auto_complete=1")
+        for r in rows:
+            jid = r["id"]
+            remaining = self.store.query_one(
+                "SELECT COUNT(*) n FROM tasks WHERE job_id=? AND state IN "
+                "('pending','ready','running','blocked')", (jid,))
+            total = self.store.query_one(
+                "SELECT COUNT(*) n FROM tasks WHERE job_id=?", (jid,))
+            if total["n"] > 0 and remaining["n"] == 0:
+                self.store.execute(
+                    "UPDATE jobs SET state='completed', completed_at=? "
+                    "WHERE id=?", (time.time(), jid))
+
+    # ----------------------------------------------------------------
+    # files (reference convoy/batch.py:3243 stream_file_and_wait)
+    # ----------------------------------------------------------------
+    def task_file(self, pool_id: str, job_id: str, task_id: str,
+                  name: str = "stdout.txt") -> Path:
+        return (self.pool_root(pool_id) / "jobs" / job_id / "tasks" /
+                task_id / name)
+
+    def stream_task_file(self, job_id: str, task_id: str,
+                         name: str = "stdout.txt",
+                         timeout: float = 60.0) -> str:
+        pool_id = self._job_pool(job_id)
+        path = self.task_file(pool_id, job_id, task_id, name)
+        deadline = time.monotonic() + timeout
+        while True:
+            self.schedule_once()
+            row = self.store.query_one(
+                "SELECT state FROM tasks WHERE job_id=? AND id=?",
+                (job_id, task_id))
+            if row and row["state"] in ("completed", "failed", "cancelled"):
+                break
+            if time.monotonic() > deadline:
+                raise TimeoutError("task did not finish")
+            time.sleep(0.05)
+        return path.read_text() if path.exists() else ""

@@ -1,0 +1,188 @@
+"""SQLite state store for the local executor.
+
+The MI355X-native replacement for the Azure Batch service's durable
+state + the reference's storage tables/queues (reference
+convoy/storage.py:68-88 `_STORAGE_CONTAINERS`): pools, GPU slots, jobs,
+tasks, dependency edges, perf events (cascade-style), and the federation
+action queue all live in one SQLite file under the object-store root.
+
+Thread-safe via one connection per store guarded by an RLock; WAL mode
+so a monitoring reader can attach concurrently.
+"""
+from __future__ import annotations
+
+import json
+import sqlite3
+import threading
+import time
+from pathlib import Path
+from typing import Any, Dict, Iterable, List, Optional
+
+_SCHEMA = """
+CREATE TABLE IF NOT EXISTS pools (
+    id TEXT PRIMARY KEY,
+    spec_json TEXT NOT NULL,
+    state TEXT NOT NULL DEFAULT 'resizing',
+    created_at REAL NOT NULL,
+    gpus_dedicated INTEGER NOT NULL,
+    gpus_low_priority INTEGER NOT NULL,
+    max_tasks_per_gpu INTEGER NOT NULL DEFAULT 1,
+    cpu_slots INTEGER NOT NULL DEFAULT 0
+);
+CREATE TABLE IF NOT EXISTS slots (
+    pool_id TEXT NOT NULL,
+    slot_id INTEGER NOT NULL,
+    kind TEXT NOT NULL,              -- gpu | cpu
+    device_id INTEGER,               -- HIP device index for gpu slots
+    dedicated INTEGER NOT NULL DEFAULT 1,
+    state TEXT NOT NULL DEFAULT 'idle',  -- idle|busy|offline|starting
+    task_ref TEXT,
+    PRIMARY KEY (pool_id, slot_id)
+);
+CREATE TABLE IF NOT EXISTS jobs (
+    id TEXT PRIMARY KEY,
+    pool_id TEXT NOT NULL,
+    spec_json TEXT NOT NULL,
+    state TEXT NOT NULL DEFAULT 'active',
+    priority INTEGER NOT NULL DEFAULT 0,
+    auto_complete INTEGER NOT NULL DEFAULT 0,
+    created_at REAL NOT NULL,
+    completed_at REAL
+);
+CREATE TABLE IF NOT EXISTS tasks (
+    job_id TEXT NOT NULL,
+    id TEXT NOT NULL,
+    spec_json TEXT NOT NULL,
+    state TEXT NOT NULL DEFAULT 'pending',
+    exit_code INTEGER,
+    retries INTEGER NOT NULL DEFAULT 0,
+    submit_time REAL NOT NULL,
+    start_time REAL,
+    end_time REAL,
+    slots_json TEXT,
+    seq INTEGER NOT NULL,
+    PRIMARY KEY (job_id, id)
+);
+CREATE TABLE IF NOT EXISTS task_deps (
+    job_id TEXT NOT NULL,
+    task_id TEXT NOT NULL,
+    depends_on TEXT NOT NULL,
+    PRIMARY KEY (job_id, task_id, depends_on)
+);
+CREATE TABLE IF NOT EXISTS events (
+    ts REAL NOT NULL,
+    source TEXT NOT NULL,
+    category TEXT NOT NULL,
+    payload TEXT
+);
+CREATE TABLE IF NOT EXISTS perf (
+    ts REAL NOT NULL,
+    source TEXT NOT NULL,
+    event TEXT NOT NULL,
+    payload TEXT
+);
+CREATE TABLE IF NOT EXISTS fed_queue (
+    id INTEGER PRIMARY KEY AUTOINCREMENT,
+    federation_id TEXT NOT NULL,
+    action TEXT NOT NULL,
+    state TEXT NOT NULL DEFAULT 'queued',  -- queued|processing|done|blocked
+    enqueued_at REAL NOT NULL,
+    attempts INTEGER NOT NULL DEFAULT 0,
+    not_before REAL NOT NULL DEFAULT 0
+);
+CREATE TABLE IF NOT EXISTS kv (
+    key TEXT PRIMARY KEY,
+    value TEXT
+);
+CREATE INDEX IF NOT EXISTS idx_tasks_state ON tasks (state);
+CREATE INDEX IF NOT EXISTS idx_jobs_pool ON jobs (pool_id);
+"""
+
+
+class Store:
+    def __init__(self, path) -> None:
+        self.path = Path(path)
+        self.path.parent.mkdir(parents=True, exist_ok=True)
+        self._lock = threading.RLock()
+        self._conn = sqlite3.connect(str(self.path), check_same_thread=False)
+        self._conn.row_factory = sqlite3.Row
+        with self._lock:
+            self._conn.execute("PRAGMA journal_mode=WAL")
+            self._conn.execute("PRAGMA synchronous=NORMAL")
+            self._conn.executescript(_SCHEMA)
+            self._conn.commit()
+
+    def close(self) -> None:
+        with self._lock:
+            self._conn.close()
+
+    # -- generic helpers ---------------------------------------------
+    def execute(self, sql: str, params: Iterable = ()) -> sqlite3.Cursor:
+        with self._lock:
+            cur = self._conn.execute(sql, tuple(params))
+            self._conn.commit()
+            return cur
+
+    def executemany(self, sql: str, rows: Iterable[Iterable]) -> None:
+        with self._lock:
+            self._conn.executemany(sql, [tuple(r) for r in rows])
+            self._conn.commit()
+
+    def query(self, sql: str, params: Iterable = ()) -> List[sqlite3.Row]:
+        with self._lock:
+            return list(self._conn.execute(sql, tuple(params)))
+
+    def query_one(self, sql: str, params: Iterable = ()) -> Optional[sqlite3.Row]:
+        rows = self.query(sql, params)
+        return rows[0] if rows else None
+
+    def transaction(self):
+        """Context manager serializing a multi-statement transaction."""
+        return _Txn(self)
+
+    # -- events / perf (cascade-style, reference cascade.py:197-226) --
+    def add_event(self, source: str, category: str,
+                  payload: Optional[Dict[str, Any]] = None) -> None:
+        self.execute(
+            "INSERT INTO events (ts, source, category, payload) "
+            "VALUES (?,?,?,?)",
+            (time.time(), source, category,
+             json.dumps(payload) if payload else None))
+
+    def add_perf(self, source: str, event: str,
+                 payload: Optional[Dict[str, Any]] = None,
+                 ts: Optional[float] = None) -> None:
+        self.execute(
+            "INSERT INTO perf (ts, source, event, payload) VALUES (?,?,?,?)",
+            (ts or time.time(), source, event,
+             json.dumps(payload) if payload else None))
+
+    # -- kv ----------------------------------------------------------
+    def kv_set(self, key: str, value: str) -> None:
+        self.execute(
+            "INSERT INTO kv (key, value) VALUES (?,?) "
+            "ON CONFLICT(key) DO UPDATE SET value=excluded.value",
+            (key, value))
+
+    def kv_get(self, key: str) -> Optional[str]:
+        row = self.query_one("SELECT value FROM kv WHERE key=?", (key,))
+        return row["value"] if row else None
+
+
+class _Txn:
+    def __init__(self, store: Store):
+        self.store = store
+
+    def __enter__(self):
+        self.store._lock.acquire()
+        return self.store._conn
+
+    def __exit__(self, exc_type, exc, tb):
+        try:
+            if exc_type is None:
+                self.store._conn.commit()
+            else:
+                self.store._conn.rollback()
+        finally:
+            self.store._lock.release()
+        return False

@@ -1,0 +1,120 @@
+"""Container/process runtime abstraction with the ROCm GPU binder.
+
+Analogue of the reference's docker/singularity run-option compiler
+(reference convoy/settings.py:3727-4305 — `--gpus`/`--nv`,
+IB device binds, `--shm-size`) re-designed for MI355X:
+
+  * GPU binding = ``HIP_VISIBLE_DEVICES`` (+ ``--device=/dev/kfd
+    --device=/dev/dri/renderD<N>`` and the video/render groups for
+    docker) — the ROCm analogue of nvidia-container-toolkit.
+  * The ``process`` runtime runs tasks as host subprocesses — the
+    native path on a single MI355X node (and the only one available in
+    this image, which ships no dockerd); docker/singularity synthesis is
+    kept at parity for hosts that have them.
+  * Gang ranks (multi-instance tasks) get RANK/LOCAL_RANK/WORLD_SIZE/
+    MASTER_* instead of ``$AZ_BATCH_HOST_LIST`` + mpirun.
+"""
+from __future__ import annotations
+
+import os
+import shlex
+import shutil
+from typing import Dict, List, Optional, Sequence
+
+
+def runtime_available(runtime: str) -> bool:
+    if runtime == "process":
+        return True
+    return shutil.which(runtime) is not None
+
+
+def rocm_render_device(gpu_index: int) -> str:
+    """Render node for HIP device N (convention: renderD128 + N)."""
+    return f"/dev/dri/renderD{128 + gpu_index}"
+
+
+def gpu_env(device_ids: Sequence[int]) -> Dict[str, str]:
+    """The ROCm binder env for a task granted these HIP devices."""
+    env = {
+        "HSA_ENABLE_IPC_MODE_LEGACY": os.environ.get(
+            "HSA_ENABLE_IPC_MODE_LEGACY", "0"),
+    }
+    if device_ids:
+        env["HIP_VISIBLE_DEVICES"] = ",".join(str(d) for d in device_ids)
+        env["ROCR_VISIBLE_DEVICES"] = env["HIP_VISIBLE_DEVICES"]
+    else:
+        # CPU-only task: hide all GPUs from it
+        env["HIP_VISIBLE_DEVICES"] = ""
+        env["ROCR_VISIBLE_DEVICES"] = ""
+    return env
+
+
+def docker_run_command(image: str, command: Optional[str],
+                       name: str,
+                       device_ids: Sequence[int],
+                       env_file: Optional[str] = None,
+                       shm_size: Optional[int] = None,
+                       working_dir: Optional[str] = None,
+                       volumes: Optional[List[str]] = None,
+                       extra_options: Optional[List[str]] = None,
+                       entrypoint: Optional[str] = None,
+                       remove: bool = True,
+                       detach: bool = False) -> List[str]:
+    """Synthesize `docker run` with the ROCm binder (the --gpus
+    analogue, reference convoy/settings.py:4239-4251)."""
+    cmd = ["docker", "run", "--name", name]
+    if remove:
+        cmd.append("--rm")
+    if detach:
+        cmd.append("-d")
+    if env_file:
+        cmd += ["--env-file", env_file]
+    if device_ids:
+        cmd += ["--device=/dev/kfd"]
+        for d in device_ids:
+            cmd += [f"--device={rocm_render_device(d)}"]
+        cmd += ["--group-add", "video", "--group-add", "render",
+                "--security-opt", "seccomp=unconfined"]
+        cmd += ["-e", "HIP_VISIBLE_DEVICES=" +
+                ",".join(str(i) for i in range(len(device_ids)))]
+    if shm_size:
+        cmd += [f"--shm-size={shm_size}"]
+    if working_dir:
+        cmd += ["-w", working_dir]
+    for vol in volumes or []:
+        cmd += ["-v", vol]
+    if entrypoint:
+        cmd += ["--entrypoint", entrypoint]
+    cmd += extra_options or []
+    cmd.append(image)
+    if command:
+        cmd += shlex.split(command)
+    return cmd
+
+
+def singularity_run_command(image: str, command: Optional[str],
+                            device_ids: Sequence[int],
+                            exec_cmd: str = "exec",
+                            working_dir: Optional[str] = None,
+                            volumes: Optional[List[str]] = None,
+                            extra_options: Optional[List[str]] = None
+                            ) -> List[str]:
+    """`singularity exec --rocm` synthesis (the --nv analogue)."""
+    cmd = ["singularity", exec_cmd]
+    if device_ids:
+        cmd.append("--rocm")
+    if working_dir:
+        cmd += ["--pwd", working_dir]
+    for vol in volumes or []:
+        cmd += ["--bind", vol]
+    cmd += extra_options or []
+    cmd.append(image)
+    if command:
+        cmd += shlex.split(command)
+    return cmd
+
+
+def process_run_command(command: str) -> List[str]:
+    """The native path: task command under bash with pipefail (the
+    reference's wrap_commands_in_shell contract, convoy/util.py:368)."""
+    return ["/bin/bash", "-c", f"set -o pipefail; {command}"]

@@ -1,0 +1,312 @@
+"""Task execution: the node-side runner contract, in-process.
+
+Analogue of the reference's node-side bash agent
+(reference scripts/shipyard_task_runner.sh:1-63: prologue -> env dump ->
+runtime exec -> epilogue with SHIPYARD_TASK_RESULT) re-designed for the
+local MI355X executor:
+
+  * every task gets a directory tree under the pool's storage root
+    (``tasks/<job>/<task>/{wd,stdout.txt,stderr.txt,.shipyard_env}``);
+  * the ``SHIPYARD_*`` env contract is preserved (task/job/pool ids,
+    dirs, runtime, user command) plus the ROCm binder env
+    (HIP_VISIBLE_DEVICES) and, for gang ranks, the torch.distributed
+    rendezvous env (RANK/LOCAL_RANK/WORLD_SIZE/MASTER_*);
+  * multi-instance tasks spawn N rank processes directly (no mpirun):
+    teardown of wedged ranks is the runner's job — when any rank exits
+    non-zero the rest get SIGTERM then SIGKILL (the local answer to the
+    reference's clean_mi_jobs + docker kill, convoy/batch.py:2322,2630).
+"""
+from __future__ import annotations
+
+import os
+import signal
+import subprocess
+import time
+from dataclasses import dataclass, field
+from pathlib import Path
+from typing import Dict, List, Optional, Sequence
+
+from shipyard_amd import utils
+from shipyard_amd.runner import runtime as rt
+
+logger = utils.get_logger(__name__)
+
+ENV_EXCLUDE = {"LS_COLORS", "PS1", "PROMPT_COMMAND"}
+
+
+@dataclass
+class TaskPaths:
+    task_dir: Path
+    working_dir: Path
+    stdout: Path
+    stderr: Path
+    env_file: Path
+
+    @classmethod
+    def create(cls, root: Path, job_id: str, task_id: str,
+               rank: Optional[int] = None) -> "TaskPaths":
+        base = root / "jobs" / job_id / "tasks" / task_id
+        if rank is not None:
+            base = base / f"rank{rank:03d}"
+        wd = base / "wd"
+        wd.mkdir(parents=True, exist_ok=True)
+        return cls(task_dir=base, working_dir=wd,
+                   stdout=base / "stdout.txt", stderr=base / "stderr.txt",
+                   env_file=base / ".shipyard_env")
+
+
+@dataclass
+class LaunchSpec:
+    """Everything the runner needs to exec one task (or one gang)."""
+    pool_id: str
+    job_id: str
+    task_id: str
+    command: str
+    runtime: str = "process"           # process | docker | singularity
+    image: Optional[str] = None
+    entrypoint: Optional[str] = None
+    env: Dict[str, str] = field(default_factory=dict)
+    device_ids: List[int] = field(default_factory=list)
+    shm_size: Optional[int] = None
+    volumes: List[str] = field(default_factory=list)
+    docker_options: List[str] = field(default_factory=list)
+    remove_container: bool = True
+    # gang
+    num_instances: int = 1
+    gang_backend: str = "rccl"
+    gpus_per_rank: int = 1
+    master_port: Optional[int] = None
+    pre_execution_command: Optional[str] = None
+    max_wall_time_s: Optional[float] = None
+
+
+@dataclass
+class RankProc:
+    rank: int
+    proc: subprocess.Popen
+    paths: TaskPaths
+    _files: tuple = ()
+
+
+class TaskHandle:
+    """A running task: 1 process, or N gang rank processes."""
+
+    def __init__(self, spec: LaunchSpec, ranks: List[RankProc],
+                 start_time: float):
+        self.spec = spec
+        self.ranks = ranks
+        self.start_time = start_time
+        self.end_time: Optional[float] = None
+        self.exit_code: Optional[int] = None
+        self.timed_out = False
+        self._first_fail: Optional[int] = None
+
+    def poll(self) -> Optional[int]:
+        """None while running; aggregate exit code when done.
+
+        Gang semantics: success iff every rank exited 0.  On first
+        non-zero exit the remaining ranks are torn down; the reported
+        exit code is the ORIGINATING rank's, not the SIGTERM codes of
+        the torn-down peers."""
+        codes = [r.proc.poll() for r in self.ranks]
+        if any(c is not None and c != 0 for c in codes):
+            if self._first_fail is None:
+                self._first_fail = next(c for c in codes
+                                        if c is not None and c != 0)
+            self._teardown()
+            codes = [r.proc.poll() for r in self.ranks]
+        if any(c is None for c in codes):
+            if (self.spec.max_wall_time_s is not None
+                    and time.monotonic() - self.start_time >
+                    self.spec.max_wall_time_s):
+                self.timed_out = True
+                self._teardown()
+            return None
+        self._close_files()
+        if self.exit_code is None:
+            self.end_time = time.time()
+            if self._first_fail is not None:
+                self.exit_code = self._first_fail
+            else:
+                self.exit_code = next((c for c in codes if c != 0), 0)
+        return self.exit_code
+
+    def wait(self, timeout: Optional[float] = None,
+             poll_interval: float = 0.02) -> int:
+        deadline = None if timeout is None else time.monotonic() + timeout
+        while True:
+            rc = self.poll()
+            if rc is not None:
+                return rc
+            if deadline is not None and time.monotonic() > deadline:
+                raise TimeoutError(
+                    f"task {self.spec.job_id}/{self.spec.task_id} "
+                    f"did not finish in {timeout}s")
+            time.sleep(poll_interval)
+
+    def kill(self) -> None:
+        self._teardown()
+
+    def _teardown(self) -> None:
+        """SIGTERM the whole process group of every live rank; escalate
+        to SIGKILL after a grace period."""
+        live = [r for r in self.ranks if r.proc.poll() is None]
+        for r in live:
+            try:
+                os.killpg(r.proc.pid, signal.SIGTERM)
+            except (ProcessLookupError, PermissionError):
+                pass
+        deadline = time.monotonic() + 5.0
+        while time.monotonic() < deadline:
+            if all(r.proc.poll() is not None for r in live):
+                break
+            time.sleep(0.05)
+        for r in live:
+            if r.proc.poll() is None:
+                try:
+                    os.killpg(r.proc.pid, signal.SIGKILL)
+                except (ProcessLookupError, PermissionError):
+                    pass
+        for r in live:
+            try:
+                r.proc.wait(timeout=5)
+            except subprocess.TimeoutExpired:
+                logger.error("rank %d of %s/%s unkillable", r.rank,
+                             self.spec.job_id, self.spec.task_id)
+        self._close_files()
+
+    def _close_files(self) -> None:
+        for r in self.ranks:
+            for f in r._files:
+                try:
+                    f.close()
+                except Exception:
+                    pass
+            r._files = ()
+
+
+def _base_env(spec: LaunchSpec, paths: TaskPaths,
+              pool_root: Path) -> Dict[str, str]:
+    env = {k: v for k, v in os.environ.items() if k not in ENV_EXCLUDE}
+    env.update({
+        "SHIPYARD_POOL_ID": spec.pool_id,
+        "SHIPYARD_JOB_ID": spec.job_id,
+        "SHIPYARD_TASK_ID": spec.task_id,
+        "SHIPYARD_TASK_DIR": str(paths.task_dir),
+        "SHIPYARD_TASK_WORKING_DIR": str(paths.working_dir),
+        "SHIPYARD_NODE_SHARED_DIR": str(pool_root / "shared"),
+        "SHIPYARD_JOB_SHARED_DIR": str(pool_root / "jobs" / spec.job_id /
+                                       "shared"),
+        "SHIPYARD_RUNTIME": spec.runtime,
+        "SHIPYARD_USER_CMD": spec.command,
+        # AZ_BATCH_* compatibility aliases for recipe portability
+        "AZ_BATCH_TASK_ID": spec.task_id,
+        "AZ_BATCH_JOB_ID": spec.job_id,
+        "AZ_BATCH_TASK_DIR": str(paths.task_dir),
+        "AZ_BATCH_TASK_WORKING_DIR": str(paths.working_dir),
+        "AZ_BATCH_NODE_SHARED_DIR": str(pool_root / "shared"),
+    })
+    env.update({k: str(v) for k, v in spec.env.items()})
+    return env
+
+
+def _dump_env(env: Dict[str, str], path: Path) -> None:
+    """Env dump minus exclusions (reference shipyard_task_runner.sh:24)."""
+    with open(path, "w") as f:
+        for k in sorted(env):
+            if k.startswith(("BASH_FUNC", "SHIPYARD_ENV_EXCLUDE")):
+                continue
+            f.write(f"{k}={env[k]}\n")
+
+
+def _spawn(cmd: List[str], env: Dict[str, str], paths: TaskPaths,
+           rank: int) -> RankProc:
+    out = open(paths.stdout, "ab")
+    err = open(paths.stderr, "ab")
+    proc = subprocess.Popen(
+        cmd, cwd=str(paths.working_dir), env=env, stdout=out, stderr=err,
+        start_new_session=True)  # own pgid → killable as a group
+    return RankProc(rank=rank, proc=proc, paths=paths, _files=(out, err))
+
+
+def launch(spec: LaunchSpec, pool_root: Path) -> TaskHandle:
+    """Launch a task (single process or gang) and return its handle."""
+    (pool_root / "shared").mkdir(parents=True, exist_ok=True)
+    (pool_root / "jobs" / spec.job_id / "shared").mkdir(parents=True,
+                                                        exist_ok=True)
+    start = time.monotonic()
+    ranks: List[RankProc] = []
+    n = max(1, spec.num_instances)
+
+    command = spec.command
+    if spec.pre_execution_command:
+        command = f"{spec.pre_execution_command}; {command}"
+
+    if n == 1:
+        paths = TaskPaths.create(pool_root, spec.job_id, spec.task_id)
+        env = _base_env(spec, paths, pool_root)
+        env.update(rt.gpu_env(spec.device_ids))
+        _dump_env(env, paths.env_file)
+        cmd = _runtime_cmd(spec, command, paths, env)
+        ranks.append(_spawn(cmd, env, paths, 0))
+    else:
+        port = spec.master_port or _free_port()
+        # split granted devices across ranks
+        per_rank = spec.gpus_per_rank
+        for rank in range(n):
+            paths = TaskPaths.create(pool_root, spec.job_id, spec.task_id,
+                                     rank=rank)
+            env = _base_env(spec, paths, pool_root)
+            devs = spec.device_ids[rank * per_rank:(rank + 1) * per_rank] \
+                if per_rank else []
+            env.update(rt.gpu_env(devs))
+            env.update({
+                "RANK": str(rank),
+                "LOCAL_RANK": "0" if devs else str(rank),
+                "WORLD_SIZE": str(n),
+                "MASTER_ADDR": "127.0.0.1",
+                "MASTER_PORT": str(port),
+                "SHIPYARD_GANG_SIZE": str(n),
+                "SHIPYARD_GANG_RANK": str(rank),
+                "SHIPYARD_GANG_BACKEND": spec.gang_backend,
+            })
+            _dump_env(env, paths.env_file)
+            cmd = _runtime_cmd(spec, command, paths, env)
+            ranks.append(_spawn(cmd, env, paths, rank))
+
+    handle = TaskHandle(spec, ranks, start)
+    return handle
+
+
+def _runtime_cmd(spec: LaunchSpec, command: str, paths: TaskPaths,
+                 env: Dict[str, str]) -> List[str]:
+    if spec.runtime == "process":
+        return rt.process_run_command(command)
+    if spec.runtime == "docker":
+        if not rt.runtime_available("docker"):
+            raise RuntimeError("docker runtime requested but not installed")
+        name = f"shipyard-{spec.job_id}-{spec.task_id}"
+        # local visible devices are remapped 0..k-1 inside the container
+        return rt.docker_run_command(
+            image=spec.image or "", command=command, name=name,
+            device_ids=spec.device_ids, env_file=str(paths.env_file),
+            shm_size=spec.shm_size, working_dir="/work",
+            volumes=[f"{paths.working_dir}:/work"] + spec.volumes,
+            extra_options=spec.docker_options, entrypoint=spec.entrypoint,
+            remove=spec.remove_container)
+    if spec.runtime == "singularity":
+        if not rt.runtime_available("singularity"):
+            raise RuntimeError(
+                "singularity runtime requested but not installed")
+        return rt.singularity_run_command(
+            image=spec.image or "", command=command,
+            device_ids=spec.device_ids, volumes=spec.volumes)
+    raise ValueError(f"unknown runtime {spec.runtime}")
+
+
+def _free_port() -> int:
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]

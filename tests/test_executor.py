@@ -1,0 +1,286 @@
+"""Executor lifecycle tests (CPU; process runtime).
+
+Covers the seam the reference leaves to the Azure Batch service
+(SURVEY.md §4): pool/job/task state machines, dependencies, retries,
+exit conditions, merge tasks, gang launch/teardown.
+"""
+import json
+import time
+
+import pytest
+
+from shipyard_amd.executor import ExecutorError, LocalExecutor
+
+
+def make_pool(ex, pool_id="p1", cpu_slots=2, gpus=0):
+    return ex.pool_add({"pool_specification": {
+        "id": pool_id,
+        "gpus": {"dedicated": gpus, "low_priority": 0},
+        "cpu_slots": cpu_slots,
+        "node_configuration": {"rocm": {"verify": False}},
+    }})
+
+
+@pytest.fixture()
+def ex(tmp_path):
+    ex = LocalExecutor(tmp_path / "root", detect_gpus=False)
+    yield ex
+    ex.store.close()
+
+
+def job(jid, tasks, **kw):
+    return {"job_specifications": [dict(id=jid, tasks=tasks, **kw)]}
+
+
+def test_pool_lifecycle(ex):
+    make_pool(ex, "pool-a", cpu_slots=2)
+    pools = ex.pool_list()
+    assert pools[0]["id"] == "pool-a" and pools[0]["state"] == "active"
+    stats = ex.pool_stats("pool-a")
+    assert stats["slots"]["idle"] == 2
+    ex.pool_del("pool-a")
+    assert ex.pool_list() == []
+
+
+def test_duplicate_pool_rejected(ex):
+    make_pool(ex, "dup")
+    with pytest.raises(ExecutorError):
+        make_pool(ex, "dup")
+
+
+def test_single_task_runs(ex, tmp_path):
+    make_pool(ex)
+    ex.jobs_add(job("j1", [{"id": "hello",
+                            "command": "echo hello-world"}]), "p1")
+    ex.run_until_idle(timeout=30)
+    tasks = ex.tasks_list("j1")
+    assert tasks[0]["state"] == "completed"
+    out = ex.task_file("p1", "j1", "hello").read_text()
+    assert "hello-world" in out
+
+
+def test_env_contract(ex):
+    make_pool(ex)
+    ex.jobs_add(job("jenv", [{
+        "id": "env",
+        "command": "echo $SHIPYARD_JOB_ID/$SHIPYARD_TASK_ID/$SHIPYARD_RUNTIME",
+        "environment_variables": {"MYVAR": "42"},
+    }]), "p1")
+    ex.run_until_idle(timeout=30)
+    out = ex.task_file("p1", "jenv", "env").read_text()
+    assert "jenv/env/process" in out
+
+
+def test_dependencies_order(ex):
+    make_pool(ex, cpu_slots=4)
+    ex.jobs_add(job("jdep", [
+        {"id": "a", "command": "sleep 0.2; date +%s.%N > a.txt"},
+        {"id": "b", "command": "date +%s.%N > b.txt",
+         "depends_on": ["a"]},
+    ]), "p1")
+    ex.run_until_idle(timeout=30)
+    tasks = {t["id"]: t for t in ex.tasks_list("jdep")}
+    assert tasks["a"]["state"] == tasks["b"]["state"] == "completed"
+    assert tasks["b"]["start_time"] >= tasks["a"]["end_time"]
+
+
+def test_depends_on_range(ex):
+    make_pool(ex, cpu_slots=4)
+    ex.jobs_add(job("jrange", [
+        {"id": "1", "command": "true"},
+        {"id": "2", "command": "true"},
+        {"id": "3", "command": "true"},
+        {"id": "final", "command": "true", "depends_on_range": [1, 3]},
+    ]), "p1")
+    ex.run_until_idle(timeout=30)
+    states = {t["id"]: t["state"] for t in ex.tasks_list("jrange")}
+    assert all(s == "completed" for s in states.values())
+
+
+def test_failed_dep_blocks(ex):
+    make_pool(ex)
+    ex.jobs_add(job("jblock", [
+        {"id": "bad", "command": "false"},
+        {"id": "child", "command": "true", "depends_on": ["bad"]},
+        {"id": "grandchild", "command": "true", "depends_on": ["child"]},
+    ]), "p1")
+    ex.run_until_idle(timeout=30)
+    states = {t["id"]: t["state"] for t in ex.tasks_list("jblock")}
+    assert states == {"bad": "failed", "child": "blocked",
+                      "grandchild": "blocked"}
+
+
+def test_dependency_action_satisfy(ex):
+    make_pool(ex)
+    ex.jobs_add(job("jsat", [
+        {"id": "bad", "command": "false",
+         "exit_conditions": {"default": {"exit_options": {
+             "dependency_action": "satisfy"}}}},
+        {"id": "child", "command": "true", "depends_on": ["bad"]},
+    ]), "p1")
+    ex.run_until_idle(timeout=30)
+    states = {t["id"]: t["state"] for t in ex.tasks_list("jsat")}
+    assert states == {"bad": "failed", "child": "completed"}
+
+
+def test_retries(ex, tmp_path):
+    make_pool(ex)
+    marker = tmp_path / "attempts"
+    # fail twice then succeed
+    cmd = (f"echo x >> {marker}; [ $(wc -l < {marker}) -ge 3 ]")
+    ex.jobs_add(job("jretry", [{"id": "r", "command": cmd,
+                                "max_task_retries": 5}]), "p1")
+    ex.run_until_idle(timeout=30)
+    t = ex.tasks_list("jretry")[0]
+    assert t["state"] == "completed"
+    assert t["retries"] == 2
+
+
+def test_retries_exhausted(ex):
+    make_pool(ex)
+    ex.jobs_add(job("jfail", [{"id": "f", "command": "false",
+                               "max_task_retries": 2}]), "p1")
+    ex.run_until_idle(timeout=30)
+    t = ex.tasks_list("jfail")[0]
+    assert t["state"] == "failed" and t["retries"] == 2
+
+
+def test_exit_condition_terminate_job(ex):
+    make_pool(ex, cpu_slots=1)
+    ex.jobs_add(job("jterm", [
+        {"id": "boom", "command": "false",
+         "exit_conditions": {"default": {"exit_options": {
+             "job_action": "terminate"}}}},
+        {"id": "later", "command": "sleep 30"},
+    ]), "p1")
+    ex.run_until_idle(timeout=30)
+    jobs = {j["id"]: j for j in ex.jobs_list()}
+    assert jobs["jterm"]["state"] == "terminated"
+
+
+def test_max_wall_time_kills(ex):
+    make_pool(ex)
+    ex.jobs_add(job("jwall", [{"id": "slow", "command": "sleep 60",
+                               "max_wall_time": "00:00:01",
+                               "max_task_retries": 0}]), "p1")
+    t0 = time.monotonic()
+    ex.run_until_idle(timeout=60)
+    assert time.monotonic() - t0 < 30
+    t = ex.tasks_list("jwall")[0]
+    assert t["state"] == "failed"
+
+
+def test_merge_task_runs_last(ex):
+    make_pool(ex, cpu_slots=4)
+    ex.jobs_add({"job_specifications": [{
+        "id": "jmerge",
+        "tasks": [{"id": "t1", "command": "true"},
+                  {"id": "t2", "command": "true"}],
+        "merge_task": {"id": "merge", "command": "true"},
+    }]}, "p1")
+    ex.run_until_idle(timeout=30)
+    tasks = {t["id"]: t for t in ex.tasks_list("jmerge")}
+    assert tasks["merge"]["state"] == "completed"
+    assert tasks["merge"]["start_time"] >= max(
+        tasks["t1"]["end_time"], tasks["t2"]["end_time"])
+
+
+def test_job_preparation_runs_first(ex):
+    make_pool(ex, cpu_slots=2)
+    ex.jobs_add({"job_specifications": [{
+        "id": "jprep",
+        "job_preparation": {"command": "touch $SHIPYARD_JOB_SHARED_DIR/ready"},
+        "tasks": [{"id": "t",
+                   "command": "test -f $SHIPYARD_JOB_SHARED_DIR/ready"}],
+    }]}, "p1")
+    ex.run_until_idle(timeout=30)
+    states = {t["id"]: t["state"] for t in ex.tasks_list("jprep")}
+    assert states["t"] == "completed"
+
+
+def test_autogenerated_task_ids(ex):
+    make_pool(ex, cpu_slots=2)
+    ex.jobs_add(job("jauto", [{"command": "true"},
+                              {"command": "true"}]), "p1")
+    ids = [t["id"] for t in ex.tasks_list("jauto")]
+    assert ids == ["task-00000", "task-00001"]
+
+
+def test_priority_ordering(ex):
+    make_pool(ex, cpu_slots=1)
+    ex.jobs_add(job("jlow", [{"id": "low", "command": "true"}],
+                    priority=0), "p1")
+    ex.jobs_add(job("jhigh", [{"id": "high", "command": "true"}],
+                    priority=10), "p1")
+    # first assignment pass must pick the high-priority job's task
+    ex.schedule_once()
+    running = ex.store.query(
+        "SELECT job_id FROM tasks WHERE state IN ('running','completed')")
+    assert {r["job_id"] for r in running} == {"jhigh"}
+    ex.run_until_idle(timeout=30)
+
+
+def test_gang_task_env_and_teardown(ex, tmp_path):
+    make_pool(ex, cpu_slots=4)
+    # 3-rank gloo gang: each rank writes its RANK; rank 1 fails -> all die
+    ex.jobs_add(job("jgang", [{
+        "id": "gang",
+        "command": "echo rank=$RANK world=$WORLD_SIZE; "
+                   "if [ \"$RANK\" = 1 ]; then exit 3; else sleep 20; fi",
+        "max_task_retries": 0,
+        "multi_instance": {
+            "num_instances": 3,
+            "gang": {"backend": "gloo", "gpus_per_rank": 0},
+        },
+    }]), "p1")
+    t0 = time.monotonic()
+    ex.run_until_idle(timeout=60)
+    assert time.monotonic() - t0 < 15, "wedged ranks not torn down"
+    t = ex.tasks_list("jgang")[0]
+    assert t["state"] == "failed" and t["exit_code"] == 3
+    base = ex.pool_root("p1") / "jobs" / "jgang" / "tasks" / "gang"
+    for rank in range(3):
+        out = (base / f"rank{rank:03d}" / "stdout.txt").read_text()
+        assert f"rank={rank} world=3" in out
+
+
+def test_job_terminate_kills_running(ex):
+    make_pool(ex)
+    ex.jobs_add(job("jkill", [{"id": "s", "command": "sleep 60"}]), "p1")
+    for _ in range(100):
+        ex.schedule_once()
+        if ex.tasks_list("jkill")[0]["state"] == "running":
+            break
+        time.sleep(0.02)
+    ex.job_terminate("jkill")
+    jobs = {j["id"]: j["state"] for j in ex.jobs_list()}
+    assert jobs["jkill"] == "terminated"
+
+
+def test_auto_complete(ex):
+    make_pool(ex)
+    ex.jobs_add(job("jac", [{"id": "t", "command": "true"}],
+                    auto_complete=True), "p1")
+    ex.run_until_idle(timeout=30)
+    jobs = {j["id"]: j["state"] for j in ex.jobs_list()}
+    assert jobs["jac"] == "completed"
+
+
+def test_task_stats(ex):
+    make_pool(ex)
+    ex.jobs_add(job("jstat", [{"id": "t", "command": "sleep 0.05"}]), "p1")
+    ex.run_until_idle(timeout=30)
+    st = ex.job_stats("jstat")
+    assert st["tasks"]["completed"] == 1
+    assert st["run_time_s"]["mean"] > 0
+
+
+def test_gpu_task_rejected_on_cpu_pool(ex):
+    make_pool(ex, cpu_slots=1)
+    ex.jobs_add(job("jgpu", [{"id": "g", "command": "true",
+                              "gpus": 2}]), "p1")
+    # never schedulable: no gpu slots; stays ready
+    for _ in range(5):
+        ex.schedule_once()
+    t = ex.tasks_list("jgpu")[0]
+    assert t["state"] == "ready"
