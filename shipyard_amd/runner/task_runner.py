@@ -207,6 +207,12 @@ def _base_env(spec: LaunchSpec, paths: TaskPaths,
         "AZ_BATCH_NODE_SHARED_DIR": str(pool_root / "shared"),
     })
     env.update({k: str(v) for k, v in spec.env.items()})
+    # make the framework importable inside tasks (the analogue of
+    # nodeprep installing shipyard on every node)
+    pkg_root = str(Path(__file__).resolve().parents[2])
+    pp = env.get("PYTHONPATH", "")
+    if pkg_root not in pp.split(os.pathsep):
+        env["PYTHONPATH"] = (pkg_root + (os.pathsep + pp if pp else ""))
     return env
 
 
