@@ -1,0 +1,96 @@
+"""Integrity manifests for bulk data (CPU + GPU paths).
+
+The reference computes whole-file MD5/SHA256 on the client CPU around
+its transfers (reference convoy/util.py:461-508, data.py:770-797
+split-chunk reassembly checks).  Here a manifest is chunk-granular
+(CRC32C) with an optional SHA-256 page digest root, computable on either
+the CPU (pure python, for CPU-only hosts/tests) or the MI355X
+(shipyard_amd.ops kernels at HBM rate).
+"""
+from __future__ import annotations
+
+import hashlib
+import json
+from dataclasses import dataclass, field
+from pathlib import Path
+from typing import List, Optional
+
+from shipyard_amd.ops import gf2
+
+DEFAULT_CHUNK = 1 << 20  # 1 MiB
+
+
+@dataclass
+class Manifest:
+    length: int
+    chunk_size: int
+    chunk_crc32c: List[int]
+    sha256_root: Optional[str] = None
+
+    def to_json(self) -> str:
+        return json.dumps({
+            "length": self.length,
+            "chunk_size": self.chunk_size,
+            "chunk_crc32c": self.chunk_crc32c,
+            "sha256_root": self.sha256_root,
+        })
+
+    @classmethod
+    def from_json(cls, s: str) -> "Manifest":
+        d = json.loads(s)
+        return cls(length=d["length"], chunk_size=d["chunk_size"],
+                   chunk_crc32c=list(d["chunk_crc32c"]),
+                   sha256_root=d.get("sha256_root"))
+
+
+def compute_cpu(data: bytes, chunk_size: int = DEFAULT_CHUNK,
+                with_sha_root: bool = True) -> Manifest:
+    crcs = []
+    sha = hashlib.sha256()
+    for off in range(0, len(data), chunk_size):
+        chunk = data[off:off + chunk_size]
+        crcs.append(gf2.crc32c(chunk))
+        if with_sha_root:
+            sha.update(hashlib.sha256(chunk).digest())
+    return Manifest(length=len(data), chunk_size=chunk_size,
+                    chunk_crc32c=crcs,
+                    sha256_root=sha.hexdigest() if with_sha_root else None)
+
+
+def compute_gpu(data_tensor, chunk_size: int = DEFAULT_CHUNK,
+                with_sha_root: bool = True) -> Manifest:
+    """GPU manifest of a uint8 CUDA tensor: CRC32C chunks + SHA-256
+    chunk digests (page_size = chunk_size), root = sha256 over the
+    concatenated chunk digests."""
+    from shipyard_amd import ops
+
+    n = data_tensor.numel()
+    crcs = [int(x) for x in
+            ops.crc32c_chunks(data_tensor, chunk_size=chunk_size).tolist()]
+    root = None
+    if with_sha_root:
+        pages = ops.sha256_pages(data_tensor, page_size=chunk_size)
+        root = hashlib.sha256(pages.cpu().numpy().tobytes()).hexdigest()
+    return Manifest(length=n, chunk_size=chunk_size, chunk_crc32c=crcs,
+                    sha256_root=root)
+
+
+def verify(manifest: Manifest, other: Manifest) -> bool:
+    return (manifest.length == other.length
+            and manifest.chunk_size == other.chunk_size
+            and manifest.chunk_crc32c == other.chunk_crc32c
+            and (manifest.sha256_root is None or other.sha256_root is None
+                 or manifest.sha256_root == other.sha256_root))
+
+
+def manifest_path(data_path: Path) -> Path:
+    return Path(str(data_path) + ".manifest.json")
+
+
+def write_manifest(data_path: Path, m: Manifest) -> None:
+    manifest_path(data_path).write_text(m.to_json())
+
+
+def read_manifest(data_path: Path) -> Optional[Manifest]:
+    p = manifest_path(data_path)
+    return Manifest.from_json(p.read_text()) if p.exists() else None

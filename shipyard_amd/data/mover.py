@@ -1,0 +1,264 @@
+"""Bulk data mover: ingress/egress with split + parallel streams.
+
+Behavioral re-implementation of the reference's data movement layer
+(reference convoy/data.py:981 `ingress_data`, 567-738
+`_multinode_transfer` min-bucket bin-packing + file splitting + per-node
+workers + Mbit/s report, 219/447 task input/output processing) for the
+local MI355X node:
+
+  * destinations are local directories (shared volumes / scratch) or
+    the object store — transport is parallel worker threads instead of
+    scp/rsync (those remain available for remote destinations);
+  * large files split into chunks moved by N workers and reassembled
+    (pwrite into a preallocated target — no remote `cat` needed);
+  * integrity is the chunked CRC32C manifest (GPU-verifiable on
+    ingest via shipyard_amd.ops);
+  * throughput is logged in Mbit/s exactly like the reference
+    (convoy/data.py:731-737).
+"""
+from __future__ import annotations
+
+import concurrent.futures as cf
+import fnmatch
+import os
+import time
+from dataclasses import dataclass
+from pathlib import Path
+from typing import Dict, List, Optional, Sequence, Tuple
+
+from shipyard_amd import utils
+from shipyard_amd.data import integrity
+from shipyard_amd.data.storage import ObjectStore
+
+logger = utils.get_logger(__name__)
+
+DEFAULT_SPLIT_MB = 128
+DEFAULT_WORKERS = 4
+
+
+@dataclass
+class TransferResult:
+    files: int
+    bytes: int
+    seconds: float
+    verified: bool
+
+    @property
+    def mbit_s(self) -> float:
+        return (self.bytes * 8 / 1e6 / self.seconds) if self.seconds else 0.0
+
+
+def _gather_files(source_path: Path, include: Sequence[str],
+                  exclude: Sequence[str]) -> List[Tuple[Path, str]]:
+    out: List[Tuple[Path, str]] = []
+    src = Path(source_path)
+    if src.is_file():
+        return [(src, src.name)]
+    for p in sorted(src.rglob("*")):
+        if not p.is_file():
+            continue
+        rel = p.relative_to(src).as_posix()
+        if include and not any(fnmatch.fnmatch(rel, pat) for pat in include):
+            continue
+        if exclude and any(fnmatch.fnmatch(rel, pat) for pat in exclude):
+            continue
+        out.append((p, rel))
+    return out
+
+
+def _bin_pack(files: List[Tuple[Path, str]], buckets: int,
+              split_bytes: Optional[int]
+              ) -> List[List[Tuple[Path, str, int, int]]]:
+    """Min-bucket bin-packing with large-file splitting (reference
+    convoy/data.py:634-665).  Returns per-worker lists of
+    (src, rel, offset, length) chunks."""
+    chunks: List[Tuple[Path, str, int, int]] = []
+    for p, rel in files:
+        size = p.stat().st_size
+        if split_bytes and size > split_bytes:
+            off = 0
+            while off < size:
+                ln = min(split_bytes, size - off)
+                chunks.append((p, rel, off, ln))
+                off += ln
+        else:
+            chunks.append((p, rel, 0, size))
+    # largest-first into the least-loaded bucket
+    buckets_: List[List[Tuple[Path, str, int, int]]] = \
+        [[] for _ in range(max(buckets, 1))]
+    loads = [0] * len(buckets_)
+    for chunk in sorted(chunks, key=lambda c: -c[3]):
+        i = loads.index(min(loads))
+        buckets_[i].append(chunk)
+        loads[i] += chunk[3]
+    return [b for b in buckets_ if b]
+
+
+def _copy_chunk(src: Path, dst: Path, offset: int, length: int) -> int:
+    """pwrite chunk copy — the local replacement for the reference's
+    split-file scp + remote `cat` reassembly (data.py:770-797)."""
+    with open(src, "rb") as f:
+        f.seek(offset)
+        data = f.read(length)
+    with open(dst, "r+b") as f:
+        f.seek(offset)
+        f.write(data)
+    return len(data)
+
+
+def ingress_directory(source_path, dest_path,
+                      include: Sequence[str] = (),
+                      exclude: Sequence[str] = (),
+                      workers: int = DEFAULT_WORKERS,
+                      split_mb: Optional[int] = DEFAULT_SPLIT_MB,
+                      verify: bool = False) -> TransferResult:
+    """Parallel local ingress: source tree -> destination directory."""
+    files = _gather_files(Path(source_path), include, exclude)
+    dest = Path(dest_path)
+    dest.mkdir(parents=True, exist_ok=True)
+    split_bytes = split_mb * (1 << 20) if split_mb else None
+
+    # preallocate targets so chunk workers can pwrite concurrently
+    for p, rel in files:
+        dst = dest / rel
+        dst.parent.mkdir(parents=True, exist_ok=True)
+        with open(dst, "wb") as f:
+            f.truncate(p.stat().st_size)
+
+    plan = _bin_pack(files, workers, split_bytes)
+    t0 = time.perf_counter()
+    total = 0
+    with cf.ThreadPoolExecutor(max_workers=max(len(plan), 1)) as pool:
+        futs = []
+        for bucket in plan:
+            def run(bucket=bucket):
+                n = 0
+                for src, rel, off, ln in bucket:
+                    n += _copy_chunk(src, dest / rel, off, ln)
+                return n
+            futs.append(pool.submit(run))
+        for f in futs:
+            total += f.result()
+    elapsed = time.perf_counter() - t0
+
+    ok = True
+    if verify:
+        for p, rel in files:
+            src_m = integrity.compute_cpu(p.read_bytes(),
+                                          with_sha_root=False)
+            dst_m = integrity.compute_cpu((dest / rel).read_bytes(),
+                                          with_sha_root=False)
+            if not integrity.verify(src_m, dst_m):
+                ok = False
+                raise ValueError(f"ingress verify failed: {rel}")
+    res = TransferResult(files=len(files), bytes=total, seconds=elapsed,
+                         verified=verify and ok)
+    logger.info("ingress: %d files %d bytes in %.3fs = %.2f Mbit/s",
+                res.files, res.bytes, res.seconds, res.mbit_s)
+    return res
+
+
+def ingress_to_object_store(source_path, store: ObjectStore,
+                            remote_path: str,
+                            include: Sequence[str] = (),
+                            exclude: Sequence[str] = (),
+                            pack: bool = True,
+                            workers: int = DEFAULT_WORKERS
+                            ) -> TransferResult:
+    """Ingress into the object store, SYSHARD-packing each file (the
+    blobxfer-analogue staging path, reference convoy/data.py:879-951)."""
+    files = _gather_files(Path(source_path), include, exclude)
+    t0 = time.perf_counter()
+    total = 0
+
+    def one(item):
+        p, rel = item
+        store.upload_file(p, f"{remote_path}/{rel}", pack=pack,
+                          manifest=not pack)
+        return p.stat().st_size
+
+    with cf.ThreadPoolExecutor(max_workers=workers) as pool:
+        for n in pool.map(one, files):
+            total += n
+    elapsed = time.perf_counter() - t0
+    res = TransferResult(files=len(files), bytes=total, seconds=elapsed,
+                         verified=pack)
+    logger.info("object-store ingress: %d files %d bytes in %.3fs = "
+                "%.2f Mbit/s", res.files, res.bytes, res.seconds, res.mbit_s)
+    return res
+
+
+def egress_from_object_store(store: ObjectStore, remote_path: str,
+                             dest_path,
+                             include: Sequence[str] = (),
+                             exclude: Sequence[str] = (),
+                             workers: int = DEFAULT_WORKERS,
+                             verify: bool = True) -> TransferResult:
+    names = list(store.list(remote_path, include=list(include) or None,
+                            exclude=list(exclude) or None))
+    dest = Path(dest_path)
+    t0 = time.perf_counter()
+    total = 0
+
+    def one(rel):
+        data = store.download_bytes(rel, verify=verify)
+        out_rel = rel[len(remote_path):].lstrip("/")
+        if out_rel.endswith(".syshard"):
+            out_rel = out_rel[:-len(".syshard")]
+        dst = dest / out_rel
+        dst.parent.mkdir(parents=True, exist_ok=True)
+        dst.write_bytes(data)
+        return len(data)
+
+    with cf.ThreadPoolExecutor(max_workers=workers) as pool:
+        for n in pool.map(one, names):
+            total += n
+    elapsed = time.perf_counter() - t0
+    res = TransferResult(files=len(names), bytes=total, seconds=elapsed,
+                         verified=verify)
+    logger.info("egress: %d files %d bytes in %.3fs = %.2f Mbit/s",
+                res.files, res.bytes, res.seconds, res.mbit_s)
+    return res
+
+
+# ---------------------------------------------------------------------
+# task-level input/output data (reference convoy/data.py:219/447)
+# ---------------------------------------------------------------------
+def process_input_data(store_roots: Dict[str, ObjectStore],
+                       specs, default_dest: Path) -> None:
+    """Materialize a task's input_data before launch."""
+    for ds in specs:
+        spec = ds.spec
+        if ds.kind == "local_storage":
+            store = store_roots[spec.get("storage_account_settings",
+                                         "default")]
+            dest = Path(utils.expand_env(spec.get("local_path")
+                                         or str(default_dest)))
+            egress_from_object_store(
+                store, spec["remote_path"], dest,
+                include=spec.get("include") or (),
+                exclude=spec.get("exclude") or (),
+                verify=spec.get("verify", True))
+
+
+def process_output_data(store_roots: Dict[str, ObjectStore],
+                        specs, task_wd: Path,
+                        task_succeeded: bool) -> None:
+    """Persist a task's output_data after exit."""
+    for ds in specs:
+        spec = ds.spec
+        cond = spec.get("condition", "taskcompletion")
+        if cond == "tasksuccess" and not task_succeeded:
+            continue
+        if cond == "taskfailure" and task_succeeded:
+            continue
+        if ds.kind == "local_storage":
+            store = store_roots[spec.get("storage_account_settings",
+                                         "default")]
+            src = Path(utils.expand_env(spec.get("local_path")
+                                        or str(task_wd)))
+            ingress_to_object_store(
+                src, store, spec["remote_path"],
+                include=spec.get("include") or (),
+                exclude=spec.get("exclude") or (),
+                pack=spec.get("encode", False))

@@ -1,0 +1,111 @@
+"""Local object store — the Azure Blob/File analogue.
+
+The reference stores resource files, global resources and task data in
+Azure Storage containers (reference convoy/storage.py:68-88).  Here a
+storage account is a directory root on the node's NVMe; containers are
+subdirectories; blobs are files.  Upload/download optionally pack
+through SYSHARD (LZ4 blocks + CRC32C manifest) so the GPU decode path
+can consume them directly.
+"""
+from __future__ import annotations
+
+import fnmatch
+import shutil
+from pathlib import Path
+from typing import Iterator, List, Optional, Tuple
+
+from shipyard_amd.data import integrity, shardfmt
+
+
+class ObjectStore:
+    def __init__(self, root, create: bool = True):
+        self.root = Path(root)
+        if create:
+            self.root.mkdir(parents=True, exist_ok=True)
+
+    def _path(self, remote_path: str) -> Path:
+        p = (self.root / remote_path.lstrip("/")).resolve()
+        if not str(p).startswith(str(self.root.resolve())):
+            raise ValueError(f"path escapes store root: {remote_path}")
+        return p
+
+    def exists(self, remote_path: str) -> bool:
+        return self._path(remote_path).exists()
+
+    def upload_bytes(self, remote_path: str, data: bytes,
+                     pack: bool = False,
+                     manifest: bool = False) -> Path:
+        dst = self._path(remote_path)
+        dst.parent.mkdir(parents=True, exist_ok=True)
+        if pack:
+            dst = dst.with_suffix(dst.suffix + ".syshard")
+            dst.write_bytes(shardfmt.pack(data))
+        else:
+            dst.write_bytes(data)
+        if manifest:
+            integrity.write_manifest(dst, integrity.compute_cpu(data))
+        return dst
+
+    def upload_file(self, local_path, remote_path: str,
+                    pack: bool = False, manifest: bool = False) -> Path:
+        data = Path(local_path).read_bytes() if (pack or manifest) else None
+        if pack or manifest:
+            return self.upload_bytes(remote_path, data, pack=pack,
+                                     manifest=manifest)
+        dst = self._path(remote_path)
+        dst.parent.mkdir(parents=True, exist_ok=True)
+        shutil.copy2(local_path, dst)
+        return dst
+
+    def download_bytes(self, remote_path: str,
+                       verify: bool = True) -> bytes:
+        p = self._path(remote_path)
+        if not p.exists() and p.with_suffix(p.suffix + ".syshard").exists():
+            p = p.with_suffix(p.suffix + ".syshard")
+        data = p.read_bytes()
+        if p.suffix == ".syshard" or data[:8] == shardfmt.MAGIC:
+            return shardfmt.unpack_cpu(data, verify=verify)
+        if verify:
+            m = integrity.read_manifest(p)
+            if m is not None:
+                got = integrity.compute_cpu(data, m.chunk_size)
+                if not integrity.verify(m, got):
+                    raise ValueError(f"integrity failure: {remote_path}")
+        return data
+
+    def download_file(self, remote_path: str, local_path,
+                      verify: bool = True) -> Path:
+        data = self.download_bytes(remote_path, verify=verify)
+        dst = Path(local_path)
+        dst.parent.mkdir(parents=True, exist_ok=True)
+        dst.write_bytes(data)
+        return dst
+
+    def delete(self, remote_path: str) -> bool:
+        p = self._path(remote_path)
+        if p.is_dir():
+            shutil.rmtree(p)
+            return True
+        if p.exists():
+            p.unlink()
+            return True
+        return False
+
+    def list(self, prefix: str = "",
+             include: Optional[List[str]] = None,
+             exclude: Optional[List[str]] = None) -> Iterator[str]:
+        base = self._path(prefix) if prefix else self.root
+        if not base.exists():
+            return
+        for p in sorted(base.rglob("*")):
+            if not p.is_file() or p.name.endswith(".manifest.json"):
+                continue
+            rel = p.relative_to(self.root).as_posix()
+            name = p.relative_to(base).as_posix()
+            if include and not any(fnmatch.fnmatch(name, pat)
+                                   for pat in include):
+                continue
+            if exclude and any(fnmatch.fnmatch(name, pat)
+                               for pat in exclude):
+                continue
+            yield rel

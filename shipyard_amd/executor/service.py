@@ -24,6 +24,8 @@ from typing import Any, Dict, List, Optional, Sequence, Tuple
 
 from shipyard_amd import utils
 from shipyard_amd.config import settings as cfg
+from shipyard_amd.data import mover
+from shipyard_amd.data.storage import ObjectStore
 from shipyard_amd.executor import task_factory
 from shipyard_amd.executor.store import Store
 from shipyard_amd.runner.task_runner import LaunchSpec, TaskHandle, launch
@@ -39,13 +41,21 @@ class ExecutorError(RuntimeError):
 
 
 class LocalExecutor:
-    def __init__(self, root, detect_gpus: bool = True):
+    def __init__(self, root, detect_gpus: bool = True,
+                 credentials_conf: Optional[Dict[str, Any]] = None):
         self.root = Path(root)
         self.root.mkdir(parents=True, exist_ok=True)
         self.store = Store(self.root / "store.db")
         self._handles: Dict[Tuple[str, str], TaskHandle] = {}
         self._detect_gpus = detect_gpus
         self._n_host_gpus: Optional[int] = None
+        # object stores (Azure Storage analogue); "default" always exists
+        self.stores: Dict[str, ObjectStore] = {
+            "default": ObjectStore(self.root / "objects")}
+        for name, sa in ((credentials_conf or {}).get("credentials", {})
+                         .get("storage", {}) or {}).items():
+            self.stores[name] = ObjectStore(sa["root"],
+                                            create=sa.get("create", True))
 
     # ----------------------------------------------------------------
     # host inventory
@@ -472,6 +482,15 @@ class LocalExecutor:
         js = self._job_settings(jid)
         ps = self._pool_settings(self._job_pool(jid))
         ts = cfg.task_settings(spec, js, ps)
+        if ts.output_data:
+            try:
+                self._process_output_data(ps, jid, tid, ts.output_data,
+                                          rc == 0)
+            except Exception as exc:
+                logger.error("output_data failed for %s/%s: %s", jid, tid,
+                             exc)
+                if rc == 0:
+                    rc = -2
         if rc == 0:
             self.store.execute(
                 "UPDATE tasks SET state='completed', exit_code=0, "
@@ -680,12 +699,53 @@ class LocalExecutor:
                 "UPDATE slots SET state='idle', task_ref=NULL WHERE "
                 "pool_id=? AND slot_id=?", (pool_id, sid))
 
+    def _task_wd(self, pool_id: str, jid: str, tid: str) -> Path:
+        return self.pool_root(pool_id) / "jobs" / jid / "tasks" / tid / "wd"
+
+    def _process_input_data(self, ps: cfg.PoolSettings, jid: str, tid: str,
+                            specs) -> None:
+        """Materialize input_data before launch: object-store egress
+        (blobxfer analogue) and local_batch task-output pulls (the
+        cargo/task_file_mover.py analogue)."""
+        dest = self._task_wd(ps.id, jid, tid)
+        dest.mkdir(parents=True, exist_ok=True)
+        for ds in specs:
+            spec = ds.spec
+            if ds.kind == "local_storage":
+                store = self.stores[spec.get("storage_account_settings",
+                                             "default")]
+                d = Path(utils.expand_env(spec.get("local_path") or
+                                          str(dest)))
+                mover.egress_from_object_store(
+                    store, spec["remote_path"], d,
+                    include=spec.get("include") or (),
+                    exclude=spec.get("exclude") or (),
+                    verify=spec.get("verify", True))
+            elif ds.kind == "local_batch":
+                src_job, src_task = spec["job_id"], spec["task_id"]
+                src_pool = self._job_pool(src_job)
+                src = self._task_wd(src_pool, src_job, src_task)
+                d = dest / (spec.get("destination") or "")
+                mover.ingress_directory(
+                    src, d, include=spec.get("include") or (),
+                    exclude=spec.get("exclude") or ())
+
+    def _process_output_data(self, ps: cfg.PoolSettings, jid: str,
+                             tid: str, specs, ok: bool) -> None:
+        wd = self._task_wd(ps.id, jid, tid)
+        mover.process_output_data(self.stores, specs, wd, ok)
+
     def _launch_task(self, ps: cfg.PoolSettings, js: cfg.JobSettings,
                      ts: cfg.TaskSettings, jid: str, tid: str,
                      device_ids: List[int], ranks: int) -> TaskHandle:
         mi = ts.multi_instance
         env = dict(ps.environment_variables)
         env.update(ts.environment_variables)
+        if ts.input_data or js.input_data:
+            self._process_input_data(ps, jid, tid,
+                                     list(js.input_data) + list(ts.input_data))
+            env["SHIPYARD_TASK_INPUT_DIR"] = str(
+                self._task_wd(ps.id, jid, tid))
         spec = LaunchSpec(
             pool_id=ps.id,
             job_id=jid,
