@@ -1,0 +1,56 @@
+"""Perf event dump with inter-event deltas.
+
+Analogue of the reference's cascade perf tooling (reference
+cascade/perf.py:55-103 `process_event` printing deltas and
+cascade/graph.py timelines) over the store's perf table.
+"""
+from __future__ import annotations
+
+import json
+from typing import List, Optional
+
+from shipyard_amd.executor.store import Store
+
+
+def events(store: Store, source_prefix: str = "") -> List[dict]:
+    rows = store.query(
+        "SELECT ts, source, event, payload FROM perf "
+        "WHERE source LIKE ? ORDER BY ts", (source_prefix + "%",))
+    out = []
+    prev_ts = {}
+    for r in rows:
+        payload = json.loads(r["payload"]) if r["payload"] else {}
+        delta = None
+        if r["source"] in prev_ts:
+            delta = r["ts"] - prev_ts[r["source"]]
+        prev_ts[r["source"]] = r["ts"]
+        out.append({"ts": r["ts"], "source": r["source"],
+                    "event": r["event"], "delta_s": delta, **payload})
+    return out
+
+
+def timeline(store: Store) -> dict:
+    """Per-source first->last span (alloc->ready chart data, the
+    cascade/graph.py analogue)."""
+    evs = events(store)
+    spans = {}
+    for e in evs:
+        s = spans.setdefault(e["source"], {"start": e["ts"],
+                                           "end": e["ts"], "events": 0})
+        s["start"] = min(s["start"], e["ts"])
+        s["end"] = max(s["end"], e["ts"])
+        s["events"] += 1
+    for s in spans.values():
+        s["span_s"] = s["end"] - s["start"]
+    return spans
+
+
+def dump(store: Store, source_prefix: str = "") -> str:
+    lines = []
+    for e in events(store, source_prefix):
+        delta = f" (+{e['delta_s']:.3f}s)" if e.get("delta_s") else ""
+        extra = {k: v for k, v in e.items()
+                 if k not in ("ts", "source", "event", "delta_s")}
+        lines.append(f"{e['ts']:.3f} {e['source']} {e['event']}{delta} "
+                     f"{json.dumps(extra) if extra else ''}".rstrip())
+    return "\n".join(lines)

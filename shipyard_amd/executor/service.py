@@ -79,7 +79,9 @@ class LocalExecutor:
     #        convoy/fleet.py:1821 _add_pool)
     # ----------------------------------------------------------------
     def pool_add(self, pool_conf: Dict[str, Any],
-                 wait_ready: bool = True) -> cfg.PoolSettings:
+                 wait_ready: bool = True,
+                 config_conf: Optional[Dict[str, Any]] = None
+                 ) -> cfg.PoolSettings:
         ps = cfg.pool_settings(pool_conf)
         if self.store.query_one("SELECT id FROM pools WHERE id=?", (ps.id,)):
             raise ExecutorError(f"pool {ps.id} exists")
@@ -108,7 +110,32 @@ class LocalExecutor:
         # nodeprep-analogue: rocm verify + start task commands
         if wait_ready:
             self.wait_for_pool_ready(ps.id)
+        # cascade-analogue: preload global resources into the pool's
+        # image cache (reference fleet.py:1821 _add_pool ->
+        # cascade distribute_global_resources)
+        if config_conf is not None:
+            gs = cfg.global_settings(config_conf)
+            imgs = [im["name"] for im in gs.local_images]
+            if (imgs or gs.docker_images) and not gs.delay_image_preload:
+                rep = self.replicator(ps.id,
+                                      concurrency=gs.
+                                      concurrent_source_downloads,
+                                      account=gs.storage_account)
+                if ps.block_until_all_global_resources_loaded:
+                    rep.distribute(local_images=imgs,
+                                   docker_images=gs.docker_images)
         return ps
+
+    def replicator(self, pool_id: str, concurrency: int = 4,
+                   account: str = "default"):
+        from shipyard_amd.cascade.replicator import Replicator
+
+        store = self.stores.get(account) or self.stores["default"]
+        return Replicator(
+            store, self.pool_root(pool_id) / "images",
+            concurrency=concurrency,
+            perf_cb=lambda src, ev, payload: self.store.add_perf(
+                src, ev, payload))
 
     def _create_slots(self, ps: cfg.PoolSettings,
                       device_ids: Sequence[int]) -> None:

@@ -1,0 +1,85 @@
+"""Cascade-analogue tests: lease arbitration, image pack/stage, perf
+events, pool preload integration."""
+import os
+import threading
+import time
+
+from shipyard_amd.cascade import perf as cperf
+from shipyard_amd.cascade.replicator import LeaseSlots, Replicator
+from shipyard_amd.data.storage import ObjectStore
+from shipyard_amd.executor import LocalExecutor
+
+
+def test_lease_slots_bound_concurrency(tmp_path):
+    slots = 2
+    active = []
+    peak = []
+    lock = threading.Lock()
+
+    def worker(i):
+        ls = LeaseSlots(tmp_path, "img", slots)
+        with ls:
+            with lock:
+                active.append(i)
+                peak.append(len(active))
+            time.sleep(0.05)
+            with lock:
+                active.remove(i)
+
+    threads = [threading.Thread(target=worker, args=(i,)) for i in range(6)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert max(peak) <= slots
+
+
+def test_pack_and_stage_image(tmp_path):
+    store = ObjectStore(tmp_path / "store")
+    src = tmp_path / "imgsrc"
+    (src / "bin").mkdir(parents=True)
+    (src / "bin" / "app").write_bytes(b"#!/bin/sh\necho hi\n" * 1000)
+    (src / "layer.dat").write_bytes(os.urandom(200_000))
+
+    events = []
+    rep = Replicator(store, tmp_path / "cache", concurrency=2,
+                     perf_cb=lambda s, e, p: events.append((s, e)))
+    meta = rep.pack_image("myimg", src)
+    assert len(meta["layers"]) == 2
+
+    res = rep.stage_image("myimg", use_gpu=False)
+    assert not res["cached"]
+    assert (tmp_path / "cache" / "myimg" / "bin" / "app").read_bytes() == \
+        (src / "bin" / "app").read_bytes()
+    assert (tmp_path / "cache" / "myimg" / "layer.dat").read_bytes() == \
+        (src / "layer.dat").read_bytes()
+    # second stage hits the cache
+    res2 = rep.stage_image("myimg", use_gpu=False)
+    assert res2["cached"]
+    assert ("image:myimg", "pull-start") in events
+    assert ("image:myimg", "pull-end") in events
+
+
+def test_distribute_and_perf_table(tmp_path):
+    ex = LocalExecutor(tmp_path / "root", detect_gpus=False)
+    src = tmp_path / "src"
+    src.mkdir()
+    (src / "data.bin").write_bytes(b"payload " * 10000)
+    rep = ex.replicator("pre")  # pack before pool exists is fine
+    rep.pack_image("img-a", src)
+
+    ex.pool_add(
+        {"pool_specification": {
+            "id": "pc", "gpus": {"dedicated": 0}, "cpu_slots": 1,
+            "node_configuration": {"rocm": {"verify": False}}}},
+        config_conf={
+            "batch_shipyard": {"storage_account_settings": "default"},
+            "global_resources": {
+                "local_images": [{"name": "img-a", "source": "prepacked"}]},
+        })
+    assert (ex.pool_root("pc") / "images" / "img-a" / "data.bin").exists()
+    dump = cperf.dump(ex.store)
+    assert "pull-end" in dump and "gr-done" in dump
+    spans = cperf.timeline(ex.store)
+    assert any(s.startswith("image:img-a") for s in spans)
+    ex.store.close()
