@@ -1,0 +1,230 @@
+"""Federation: constraint-matched job scheduling across pools.
+
+Behavioral re-implementation of the reference's federation proxy daemon
+(reference federation/federation.py:2727 FederationProcessor, 2230
+`find_target_pool_for_job`, 1709 `_filter_pool_with_hard_constraints`,
+2084 `_greedy_best_fit_match_for_job`, 3095 `process_federation_queue`,
+1332 blocked-action requeue with backoff) as an in-process scheduler:
+the queue is the store's fed_queue table, federations are named sets of
+local pools, and jobs route to the best-fit pool at dequeue time.
+"""
+from __future__ import annotations
+
+import json
+import time
+from dataclasses import dataclass
+from typing import Any, Dict, List, Optional
+
+from shipyard_amd import utils
+from shipyard_amd.config import settings as cfg
+
+logger = utils.get_logger(__name__)
+
+MAX_ATTEMPTS = 10
+BACKOFF_BASE_S = 0.5
+
+
+@dataclass
+class PoolSnapshot:
+    pool_id: str
+    state: str
+    gpus_dedicated: int
+    gpus_low_priority: int
+    idle_gpu_slots: int
+    idle_cpu_slots: int
+    backlog: int
+    autoscale_enabled: bool
+
+
+class Federation:
+    def __init__(self, fed_id: str, pools: List[str],
+                 force_unique_job_ids: bool = False):
+        self.id = fed_id
+        self.pools = list(pools)
+        self.force_unique_job_ids = force_unique_job_ids
+
+
+class FederationProcessor:
+    def __init__(self, executor, federations: Dict[str, Federation]):
+        self.ex = executor
+        self.federations = federations
+
+    @classmethod
+    def from_config(cls, executor, fed_conf: Dict[str, Any]
+                    ) -> "FederationProcessor":
+        feds = {}
+        for fid, spec in (fed_conf.get("federation", {})
+                          .get("federations", {}) or {}).items():
+            feds[fid] = Federation(
+                fid, spec["pools"],
+                force_unique_job_ids=spec.get("force_unique_job_ids",
+                                              False))
+        return cls(executor, feds)
+
+    # -- submission (reference storage.py:1276 add_job_to_federation) --
+    def submit_job(self, federation_id: str,
+                   jobs_conf: Dict[str, Any]) -> int:
+        if federation_id not in self.federations:
+            raise KeyError(f"unknown federation {federation_id}")
+        cur = self.ex.store.execute(
+            "INSERT INTO fed_queue (federation_id, action, enqueued_at) "
+            "VALUES (?,?,?)",
+            (federation_id,
+             json.dumps({"kind": "add_job", "jobs": jobs_conf}),
+             time.time()))
+        return cur.lastrowid
+
+    # -- pool snapshots ------------------------------------------------
+    def _snapshot(self, pool_id: str) -> Optional[PoolSnapshot]:
+        row = self.ex.store.query_one(
+            "SELECT * FROM pools WHERE id=?", (pool_id,))
+        if row is None:
+            return None
+        idle_gpu = self.ex.store.query_one(
+            "SELECT COUNT(*) n FROM slots WHERE pool_id=? AND "
+            "state='idle' AND kind='gpu'", (pool_id,))["n"]
+        idle_cpu = self.ex.store.query_one(
+            "SELECT COUNT(*) n FROM slots WHERE pool_id=? AND "
+            "state='idle' AND kind='cpu'", (pool_id,))["n"]
+        backlog = self.ex.store.query_one(
+            "SELECT COUNT(*) n FROM tasks t JOIN jobs j ON t.job_id=j.id "
+            "WHERE j.pool_id=? AND t.state IN ('pending','ready')",
+            (pool_id,))["n"]
+        spec = json.loads(row["spec_json"])
+        ps = cfg.pool_settings(spec)
+        return PoolSnapshot(
+            pool_id=pool_id, state=row["state"],
+            gpus_dedicated=row["gpus_dedicated"],
+            gpus_low_priority=row["gpus_low_priority"],
+            idle_gpu_slots=idle_gpu, idle_cpu_slots=idle_cpu,
+            backlog=backlog, autoscale_enabled=ps.autoscale.enabled)
+
+    # -- constraint filtering (reference federation.py:1709) ----------
+    def _passes_hard_constraints(self, snap: PoolSnapshot,
+                                 constraints: Optional[dict],
+                                 job_gpus: int) -> bool:
+        if snap.state != "active":
+            return False
+        total_gpus = snap.gpus_dedicated + snap.gpus_low_priority
+        if job_gpus > 0 and total_gpus < job_gpus:
+            return False
+        if not constraints:
+            return True
+        pc = constraints.get("pool") or {}
+        if pc.get("autoscale"):
+            a = pc["autoscale"]
+            if not a.get("allow", True) and snap.autoscale_enabled:
+                return False
+            if a.get("exclusive") and not snap.autoscale_enabled:
+                return False
+        if pc.get("low_priority_nodes"):
+            lp = pc["low_priority_nodes"]
+            if not lp.get("allow", True) and snap.gpus_low_priority > 0:
+                return False
+            if lp.get("exclusive") and snap.gpus_low_priority == 0:
+                return False
+        mab = pc.get("max_active_task_backlog") or {}
+        ratio = mab.get("ratio")
+        if ratio is not None:
+            slots = max(snap.gpus_dedicated + snap.gpus_low_priority, 1)
+            if snap.backlog / slots > ratio and not (
+                    mab.get("autoscale_exempt", True)
+                    and snap.autoscale_enabled):
+                return False
+        cn = constraints.get("compute_node") or {}
+        if cn.get("gpus") and snap.gpus_dedicated + \
+                snap.gpus_low_priority < cn["gpus"]:
+            return False
+        return True
+
+    # -- greedy best fit (reference federation.py:2084) ----------------
+    def find_target_pool_for_job(self, fed: Federation,
+                                 jobspec: dict) -> Optional[str]:
+        js = cfg.job_settings(jobspec)
+        job_gpus = 0
+        for t in js.tasks:
+            g = cfg.resolve_gpus(t.get("gpus", js.gpus_default), None)
+            mi = t.get("multi_instance")
+            if mi:
+                gpr = ((mi.get("gang") or {}).get("gpus_per_rank", 1))
+                ni = mi.get("num_instances", 1)
+                if isinstance(ni, int):
+                    g = max(g, ni * gpr)
+            job_gpus = max(job_gpus, g)
+        candidates = []
+        for pid in fed.pools:
+            snap = self._snapshot(pid)
+            if snap is None:
+                continue
+            if not self._passes_hard_constraints(
+                    snap, js.federation_constraints, job_gpus):
+                continue
+            can_run_now = (snap.idle_gpu_slots >= job_gpus
+                           if job_gpus else
+                           (snap.idle_cpu_slots + snap.idle_gpu_slots) > 0)
+            # greedy best fit: runnable-now first, then least backlog,
+            # then tightest fit (fewest idle slots that still satisfy)
+            candidates.append((not can_run_now, snap.backlog,
+                               snap.idle_gpu_slots, pid))
+        if not candidates:
+            return None
+        candidates.sort()
+        return candidates[0][3]
+
+    # -- queue processing (reference federation.py:3095) ---------------
+    def process_queue_once(self) -> int:
+        now = time.time()
+        rows = self.ex.store.query(
+            "SELECT * FROM fed_queue WHERE state IN ('queued','blocked') "
+            "AND not_before<=? ORDER BY id", (now,))
+        n = 0
+        for r in rows:
+            fid = r["federation_id"]
+            fed = self.federations.get(fid)
+            action = json.loads(r["action"])
+            if fed is None:
+                self._mark(r["id"], "done")  # drop unknown federation
+                continue
+            if action["kind"] != "add_job":
+                self._mark(r["id"], "done")
+                continue
+            ok = True
+            placed = []
+            for jobspec in action["jobs"]["job_specifications"]:
+                if fed.force_unique_job_ids and self.ex.store.query_one(
+                        "SELECT id FROM jobs WHERE id=?", (jobspec["id"],)):
+                    ok = False
+                    break
+                target = self.find_target_pool_for_job(fed, jobspec)
+                if target is None:
+                    ok = False
+                    break
+                placed.append((target, jobspec))
+            if not ok:
+                self._requeue(r)
+                continue
+            for target, jobspec in placed:
+                self.ex.jobs_add({"job_specifications": [jobspec]}, target)
+                self.ex.store.add_event(
+                    f"fed:{fid}", "job-scheduled",
+                    {"job": jobspec["id"], "pool": target})
+            self._mark(r["id"], "done")
+            n += 1
+        return n
+
+    def _mark(self, qid: int, state: str) -> None:
+        self.ex.store.execute(
+            "UPDATE fed_queue SET state=? WHERE id=?", (state, qid))
+
+    def _requeue(self, row) -> None:
+        """Blocked-action backoff (reference federation.py:1332-1364)."""
+        attempts = row["attempts"] + 1
+        if attempts >= MAX_ATTEMPTS:
+            self._mark(row["id"], "failed")
+            logger.warning("federation action %d failed permanently",
+                           row["id"])
+            return
+        delay = BACKOFF_BASE_S * (2 ** min(attempts, 8))
+        self.ex.store.execute(
+            "UPDATE fed_queue SET state='blocked', attempts=?, not_before=?"
+            " WHERE id=?", (attempts, time.time() + delay, row["id"]))

