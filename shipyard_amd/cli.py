@@ -1,0 +1,555 @@
+"""shipyard — CLI for the MI355X-native batch framework.
+
+Mirrors the reference's click command tree (reference shipyard.py:
+1001-3136: account/fs/storage/keyvault/cert/pool/jobs/data/diag/misc/
+monitor/fed/slurm) against the local executor.  Config files resolve
+from --configdir / per-file options / SHIPYARD_*_CONF env, strictly
+validated before any action (reference shipyard.py:441-580).
+"""
+from __future__ import annotations
+
+import json
+import os
+import sys
+import time
+from pathlib import Path
+
+import click
+
+from shipyard_amd import utils
+from shipyard_amd.config import ConfigBundle, ConfigType
+
+logger = utils.get_logger("shipyard")
+
+
+class CliContext:
+    def __init__(self):
+        self.configdir = None
+        self.root = None
+        self.raw = False
+        self._bundle = None
+        self._executor = None
+        self.overrides = {}
+
+    @property
+    def bundle(self) -> ConfigBundle:
+        if self._bundle is None:
+            self._bundle = ConfigBundle.from_dir(self.configdir or ".",
+                                                 **self.overrides)
+        return self._bundle
+
+    def conf(self, ctype: ConfigType, required: bool = True):
+        doc = self.bundle.get(ctype)
+        if doc is None and required:
+            raise click.ClickException(
+                f"{ctype.value}.yaml not found (use --configdir or "
+                f"SHIPYARD_{ctype.value.upper()}_CONF)")
+        return doc
+
+    @property
+    def executor(self):
+        if self._executor is None:
+            from shipyard_amd.executor import LocalExecutor
+
+            root = self.root or os.environ.get("SHIPYARD_ROOT")
+            if root is None:
+                creds = self.bundle.get(ConfigType.credentials)
+                if creds:
+                    sa = (creds.get("credentials", {}).get("storage") or
+                          {}).get("default")
+                    if sa:
+                        root = sa["root"]
+            if root is None:
+                root = os.path.expanduser("~/.shipyard_amd")
+            self._executor = LocalExecutor(
+                root, credentials_conf=self.bundle.get(
+                    ConfigType.credentials))
+        return self._executor
+
+    def emit(self, obj) -> None:
+        if self.raw:
+            click.echo(json.dumps(obj, indent=2, default=str))
+        else:
+            click.echo(json.dumps(obj, indent=2, default=str))
+
+
+pass_ctx = click.make_pass_decorator(CliContext, ensure=True)
+
+
+def _common(f):
+    f = click.option("--configdir", envvar="SHIPYARD_CONFIGDIR",
+                     help="directory with config yaml files")(f)
+    f = click.option("--root", envvar="SHIPYARD_ROOT",
+                     help="executor state root")(f)
+    f = click.option("--raw", is_flag=True, help="raw json output")(f)
+    return f
+
+
+def _apply(ctx, configdir, root, raw):
+    if configdir:
+        ctx.configdir = configdir
+    if root:
+        ctx.root = root
+    ctx.raw = ctx.raw or raw
+
+
+@click.group()
+@click.version_option(version="0.1.0", prog_name="shipyard-amd")
+def cli():
+    """Batch Shipyard for MI355X: local GPU-pool batch orchestration."""
+
+
+# ---------------------------------------------------------------- account
+@cli.group()
+def account():
+    """Node/account info."""
+
+
+@account.command("info")
+@_common
+@pass_ctx
+def account_info(ctx, configdir, root, raw):
+    """Show node inventory (the `account info` analogue)."""
+    _apply(ctx, configdir, root, raw)
+    info = {"gpus": 0, "rocm": None, "hostname": os.uname().nodename}
+    try:
+        import torch
+
+        if torch.cuda.is_available():
+            info["gpus"] = torch.cuda.device_count()
+            info["device_name"] = torch.cuda.get_device_name(0)
+        info["rocm"] = getattr(torch.version, "hip", None)
+    except Exception:
+        pass
+    ctx.emit(info)
+
+
+# ---------------------------------------------------------------- pool
+@cli.group()
+def pool():
+    """Pool lifecycle (GPU slot partitions)."""
+
+
+@pool.command("add")
+@_common
+@pass_ctx
+def pool_add(ctx, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ps = ctx.executor.pool_add(ctx.conf(ConfigType.pool),
+                               config_conf=ctx.conf(ConfigType.config,
+                                                    required=False))
+    ctx.emit({"pool": ps.id, "state": "active",
+              "gpus": ps.gpus_dedicated + ps.gpus_low_priority})
+
+
+@pool.command("list")
+@_common
+@pass_ctx
+def pool_list(ctx, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ctx.emit(ctx.executor.pool_list())
+
+
+@pool.command("del")
+@click.option("--poolid", help="pool id (default: from pool.yaml)")
+@click.option("--force", is_flag=True)
+@_common
+@pass_ctx
+def pool_del(ctx, poolid, force, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    pid = poolid or ctx.conf(ConfigType.pool)["pool_specification"]["id"]
+    ctx.executor.pool_del(pid, force=force)
+    ctx.emit({"deleted": pid})
+
+
+@pool.command("resize")
+@click.option("--poolid")
+@click.option("--dedicated", type=int)
+@click.option("--low-priority", type=int)
+@_common
+@pass_ctx
+def pool_resize(ctx, poolid, dedicated, low_priority, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    pid = poolid or ctx.conf(ConfigType.pool)["pool_specification"]["id"]
+    ctx.executor.pool_resize(pid, dedicated=dedicated,
+                             low_priority=low_priority)
+    ctx.emit(ctx.executor.pool_stats(pid))
+
+
+@pool.command("stats")
+@click.option("--poolid")
+@_common
+@pass_ctx
+def pool_stats(ctx, poolid, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    pid = poolid or ctx.conf(ConfigType.pool)["pool_specification"]["id"]
+    ctx.emit(ctx.executor.pool_stats(pid))
+
+
+@pool.group("images")
+def pool_images():
+    """Global resource (image) ops on a pool."""
+
+
+@pool_images.command("update")
+@click.option("--poolid")
+@_common
+@pass_ctx
+def pool_images_update(ctx, poolid, configdir, root, raw):
+    """Re-distribute global resources (reference fleet.py:2241)."""
+    _apply(ctx, configdir, root, raw)
+    pid = poolid or ctx.conf(ConfigType.pool)["pool_specification"]["id"]
+    from shipyard_amd.config.settings import global_settings
+
+    gs = global_settings(ctx.conf(ConfigType.config))
+    rep = ctx.executor.replicator(pid,
+                                  concurrency=gs.concurrent_source_downloads,
+                                  account=gs.storage_account)
+    res = rep.distribute(
+        local_images=[im["name"] for im in gs.local_images],
+        docker_images=gs.docker_images)
+    ctx.emit(res)
+
+
+# ---------------------------------------------------------------- jobs
+@cli.group()
+def jobs():
+    """Job and task operations."""
+
+
+@jobs.command("add")
+@click.option("--poolid")
+@click.option("--tail", help="stream this file of the last task")
+@click.option("--wait", is_flag=True, help="run scheduler until idle")
+@_common
+@pass_ctx
+def jobs_add(ctx, poolid, tail, wait, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    pid = poolid or ctx.conf(ConfigType.pool)["pool_specification"]["id"]
+    added = ctx.executor.jobs_add(ctx.conf(ConfigType.jobs), pid)
+    if wait or tail:
+        ctx.executor.run_until_idle()
+    out = {"jobs": added}
+    if tail:
+        jid = added[-1]
+        tasks = ctx.executor.tasks_list(jid)
+        if tasks:
+            out["tail"] = ctx.executor.task_file(
+                pid, jid, tasks[-1]["id"], tail).read_text()
+    ctx.emit(out)
+
+
+@jobs.command("list")
+@_common
+@pass_ctx
+def jobs_list(ctx, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ctx.emit(ctx.executor.jobs_list())
+
+
+@jobs.command("term")
+@click.option("--jobid", required=True)
+@_common
+@pass_ctx
+def jobs_term(ctx, jobid, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ctx.executor.job_terminate(jobid)
+    ctx.emit({"terminated": jobid})
+
+
+@jobs.command("del")
+@click.option("--jobid", required=True)
+@_common
+@pass_ctx
+def jobs_del(ctx, jobid, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ctx.executor.job_del(jobid)
+    ctx.emit({"deleted": jobid})
+
+
+@jobs.command("disable")
+@click.option("--jobid", required=True)
+@_common
+@pass_ctx
+def jobs_disable(ctx, jobid, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ctx.executor.job_disable(jobid)
+    ctx.emit({"disabled": jobid})
+
+
+@jobs.command("enable")
+@click.option("--jobid", required=True)
+@_common
+@pass_ctx
+def jobs_enable(ctx, jobid, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ctx.executor.job_enable(jobid)
+    ctx.emit({"enabled": jobid})
+
+
+@jobs.command("stats")
+@click.option("--jobid")
+@_common
+@pass_ctx
+def jobs_stats(ctx, jobid, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ctx.emit(ctx.executor.job_stats(jobid))
+
+
+@jobs.group("tasks")
+def jobs_tasks():
+    """Task-level operations."""
+
+
+@jobs_tasks.command("list")
+@click.option("--jobid", required=True)
+@_common
+@pass_ctx
+def tasks_list(ctx, jobid, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ctx.emit(ctx.executor.tasks_list(jobid))
+
+
+# ---------------------------------------------------------------- data
+@cli.group()
+def data():
+    """Data movement: ingress/egress/stream."""
+
+
+@data.command("ingress")
+@_common
+@pass_ctx
+def data_ingress(ctx, configdir, root, raw):
+    """Ingress global_resources.files (reference data.py:981)."""
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.config.settings import global_settings
+    from shipyard_amd.data import mover
+
+    gs = global_settings(ctx.conf(ConfigType.config))
+    results = []
+    for f in gs.files:
+        src = f["source"]["path"]
+        dst = f["destination"]
+        xfer = (dst.get("data_transfer") or {})
+        method = xfer.get("method", "local_copy")
+        include = f["source"].get("include") or ()
+        exclude = f["source"].get("exclude") or ()
+        if method == "object_store" or dst.get("remote_path"):
+            store = ctx.executor.stores.get(
+                dst.get("storage_account_settings", "default"))
+            res = mover.ingress_to_object_store(
+                src, store, dst.get("remote_path", "ingest"),
+                include=include, exclude=exclude,
+                pack=xfer.get("compress", True))
+        else:
+            sdv = dst.get("shared_data_volume")
+            rel = dst.get("relative_destination_path") or ""
+            base = Path(ctx.executor.root) / "volumes" / (sdv or "default")
+            res = mover.ingress_directory(
+                src, base / rel, include=include, exclude=exclude,
+                workers=xfer.get("max_parallel_transfers_per_node", 4),
+                split_mb=xfer.get("split_files_megabytes", 128),
+                verify=xfer.get("verify", False))
+        results.append({"source": src, "files": res.files,
+                        "bytes": res.bytes, "mbit_s": round(res.mbit_s, 2)})
+    ctx.emit(results)
+
+
+@data.group("files")
+def data_files():
+    """Task file access."""
+
+
+@data_files.command("stream")
+@click.option("--filespec", required=True,
+              help="jobid,taskid[,filename]")
+@_common
+@pass_ctx
+def files_stream(ctx, filespec, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    parts = filespec.split(",")
+    jid, tid = parts[0], parts[1]
+    name = parts[2] if len(parts) > 2 else "stdout.txt"
+    click.echo(ctx.executor.stream_task_file(jid, tid, name))
+
+
+# ---------------------------------------------------------------- storage
+@cli.group()
+def storage():
+    """Object store operations."""
+
+
+@storage.command("clear")
+@click.option("--prefix", default="")
+@_common
+@pass_ctx
+def storage_clear(ctx, prefix, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    st = ctx.executor.stores["default"]
+    n = 0
+    for name in list(st.list(prefix)):
+        st.delete(name)
+        n += 1
+    ctx.emit({"deleted": n})
+
+
+@storage.command("list")
+@click.option("--prefix", default="")
+@_common
+@pass_ctx
+def storage_list(ctx, prefix, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ctx.emit(list(ctx.executor.stores["default"].list(prefix)))
+
+
+# ---------------------------------------------------------------- diag
+@cli.group()
+def diag():
+    """Diagnostics: events and perf timelines."""
+
+
+@diag.command("events")
+@click.option("--prefix", default="")
+@_common
+@pass_ctx
+def diag_events(ctx, prefix, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.cascade import perf
+
+    click.echo(perf.dump(ctx.executor.store, prefix))
+
+
+@diag.command("timeline")
+@_common
+@pass_ctx
+def diag_timeline(ctx, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.cascade import perf
+
+    ctx.emit(perf.timeline(ctx.executor.store))
+
+
+# ---------------------------------------------------------------- monitor
+@cli.group()
+def monitor():
+    """Monitoring: exporter + file_sd targets."""
+
+
+@monitor.command("scrape")
+@_common
+@pass_ctx
+def monitor_scrape(ctx, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.monitor.exporter import Exporter
+
+    click.echo(Exporter(store=ctx.executor.store).scrape().decode())
+
+
+@monitor.command("start")
+@click.option("--port", type=int, default=9400)
+@_common
+@pass_ctx
+def monitor_start(ctx, port, configdir, root, raw):  # pragma: no cover
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.monitor.exporter import Exporter
+
+    Exporter(store=ctx.executor.store, port=port).serve_forever()
+
+
+@monitor.command("targets")
+@click.option("--outdir", required=True)
+@_common
+@pass_ctx
+def monitor_targets(ctx, outdir, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.monitor import heimdall
+
+    files = heimdall.write_file_sd(ctx.executor.store, outdir)
+    ctx.emit([str(f) for f in files])
+
+
+# ---------------------------------------------------------------- fed
+@cli.group()
+def fed():
+    """Federation: multi-pool constraint scheduling."""
+
+
+@fed.command("jobs-add")
+@click.option("--federation-id", required=True)
+@_common
+@pass_ctx
+def fed_jobs_add(ctx, federation_id, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.federation.scheduler import FederationProcessor
+
+    fp = FederationProcessor.from_config(
+        ctx.executor, ctx.conf(ConfigType.federation))
+    qid = fp.submit_job(federation_id, ctx.conf(ConfigType.jobs))
+    ctx.emit({"queued": qid})
+
+
+@fed.command("process")
+@_common
+@pass_ctx
+def fed_process(ctx, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.federation.scheduler import FederationProcessor
+
+    fp = FederationProcessor.from_config(
+        ctx.executor, ctx.conf(ConfigType.federation))
+    n = fp.process_queue_once()
+    ctx.emit({"processed": n})
+
+
+# ---------------------------------------------------------------- daemon
+@cli.command("daemon")
+@click.option("--idle-exit", is_flag=True,
+              help="exit when no work remains")
+@click.option("--interval", type=float, default=0.05)
+@_common
+@pass_ctx
+def daemon(ctx, idle_exit, interval, configdir, root, raw):
+    """Scheduler loop: task scheduling + autoscale + federation queue +
+    recurrences (the local stand-in for the Azure Batch service)."""
+    _apply(ctx, configdir, root, raw)
+    ex = ctx.executor
+    from shipyard_amd.config import settings as cfg
+    from shipyard_amd.executor.autoscale import AutoscaleController
+
+    controllers = {}
+    fp = None
+    fed_conf = ctx.conf(ConfigType.federation, required=False)
+    if fed_conf:
+        from shipyard_amd.federation.scheduler import FederationProcessor
+
+        fp = FederationProcessor.from_config(ex, fed_conf)
+    logger.info("daemon running (idle_exit=%s)", idle_exit)
+    while True:
+        ex.schedule_once()
+        now = time.time()
+        for p in ex.pool_list():
+            pid = p["id"]
+            if pid not in controllers:
+                controllers[pid] = AutoscaleController(
+                    ex, pid, ex._pool_settings(pid).autoscale)
+            controllers[pid].maybe_evaluate(now)
+        if fp is not None:
+            fp.process_queue_once()
+        if idle_exit:
+            busy = ex.store.query_one(
+                "SELECT COUNT(*) n FROM tasks WHERE state IN "
+                "('pending','ready','running')")
+            qd = ex.store.query_one(
+                "SELECT COUNT(*) n FROM fed_queue WHERE state IN "
+                "('queued','blocked')")
+            if busy["n"] == 0 and qd["n"] == 0:
+                break
+        time.sleep(interval)
+
+
+def main():  # console entry
+    cli(prog_name="shipyard")
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,123 @@
+"""CLI end-to-end tests (the min end-to-end slice of SURVEY.md §7.3:
+pool add -> jobs add -> files stream -> jobs del, all local)."""
+import json
+from pathlib import Path
+
+import pytest
+import yaml
+from click.testing import CliRunner
+
+from shipyard_amd.cli import cli
+
+
+@pytest.fixture()
+def configdir(tmp_path):
+    cfg = tmp_path / "conf"
+    cfg.mkdir()
+    (cfg / "credentials.yaml").write_text(yaml.safe_dump({
+        "credentials": {"storage": {"default": {
+            "root": str(tmp_path / "sroot")}}},
+    }))
+    (cfg / "config.yaml").write_text(yaml.safe_dump({
+        "batch_shipyard": {"storage_account_settings": "default"},
+    }))
+    (cfg / "pool.yaml").write_text(yaml.safe_dump({
+        "pool_specification": {
+            "id": "clipool",
+            "gpus": {"dedicated": 0, "low_priority": 0},
+            "cpu_slots": 2,
+            "node_configuration": {"rocm": {"verify": False}},
+        }}))
+    (cfg / "jobs.yaml").write_text(yaml.safe_dump({
+        "job_specifications": [{
+            "id": "clijob",
+            "tasks": [{"id": "t1", "command": "echo cli-slice-ok"}],
+        }]}))
+    return cfg
+
+
+def run(args, configdir, tmp_path, catch=False):
+    runner = CliRunner()
+    res = runner.invoke(cli, args + ["--configdir", str(configdir),
+                                     "--root", str(tmp_path / "xroot")],
+                        catch_exceptions=catch)
+    return res
+
+
+def test_end_to_end_slice(configdir, tmp_path):
+    r = run(["pool", "add"], configdir, tmp_path)
+    assert r.exit_code == 0, r.output
+    assert json.loads(r.output)["pool"] == "clipool"
+
+    r = run(["pool", "list"], configdir, tmp_path)
+    assert "clipool" in r.output
+
+    r = run(["jobs", "add", "--wait"], configdir, tmp_path)
+    assert r.exit_code == 0, r.output
+
+    r = run(["data", "files", "stream", "--filespec", "clijob,t1"],
+            configdir, tmp_path)
+    assert "cli-slice-ok" in r.output
+
+    r = run(["jobs", "stats", "--jobid", "clijob"], configdir, tmp_path)
+    assert json.loads(r.output)["tasks"]["completed"] == 1
+
+    r = run(["jobs", "del", "--jobid", "clijob"], configdir, tmp_path)
+    assert r.exit_code == 0
+
+    r = run(["pool", "del"], configdir, tmp_path)
+    assert r.exit_code == 0
+
+
+def test_validation_rejects_bad_config(configdir, tmp_path):
+    (configdir / "pool.yaml").write_text(yaml.safe_dump({
+        "pool_specification": {"id": "x", "gpus": {"dedicated": 1},
+                               "bogus": True}}))
+    r = run(["pool", "add"], configdir, tmp_path, catch=True)
+    assert r.exit_code != 0
+    assert isinstance(r.exception, Exception)
+
+
+def test_account_info(configdir, tmp_path):
+    r = run(["account", "info"], configdir, tmp_path)
+    assert r.exit_code == 0
+    assert "gpus" in json.loads(r.output)
+
+
+def test_daemon_idle_exit_with_fed(configdir, tmp_path):
+    (configdir / "federation.yaml").write_text(yaml.safe_dump({
+        "federation": {"federations": {"f1": {"pools": ["clipool"]}}}}))
+    r = run(["pool", "add"], configdir, tmp_path)
+    assert r.exit_code == 0
+    r = run(["fed", "jobs-add", "--federation-id", "f1"], configdir,
+            tmp_path)
+    assert r.exit_code == 0, r.output
+    r = run(["daemon", "--idle-exit"], configdir, tmp_path)
+    assert r.exit_code == 0, r.output
+    r = run(["jobs", "stats", "--jobid", "clijob"], configdir, tmp_path)
+    assert json.loads(r.output)["tasks"]["completed"] == 1
+
+
+def test_monitor_scrape(configdir, tmp_path):
+    r = run(["pool", "add"], configdir, tmp_path)
+    r = run(["monitor", "scrape"], configdir, tmp_path)
+    assert r.exit_code == 0
+    assert "shipyard_executor_metric" in r.output
+
+
+def test_storage_and_ingress(configdir, tmp_path):
+    src = tmp_path / "ingest-src"
+    src.mkdir()
+    (src / "x.dat").write_bytes(b"abc" * 1000)
+    conf = yaml.safe_load((configdir / "config.yaml").read_text())
+    conf["global_resources"] = {"files": [{
+        "source": {"path": str(src)},
+        "destination": {"storage_account_settings": "default",
+                        "remote_path": "staged",
+                        "data_transfer": {"method": "object_store"}},
+    }]}
+    (configdir / "config.yaml").write_text(yaml.safe_dump(conf))
+    r = run(["data", "ingress"], configdir, tmp_path)
+    assert r.exit_code == 0, r.output
+    r = run(["storage", "list", "--prefix", "staged"], configdir, tmp_path)
+    assert "x.dat" in r.output
