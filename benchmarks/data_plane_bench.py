@@ -1,0 +1,169 @@
+#!/usr/bin/env python3
+"""Data-plane kernel benchmark on MI355X: CRC32C / SHA-256 / LZ4 decode.
+
+Measures the HIP hot paths that replace the reference's CPU-side
+hashing + dockerd inflate (BASELINE.md metric #3: "HIP data-mover hot
+path ... throughput reported vs CPU path").  Run under rocprofv3 for
+per-kernel evidence:
+
+  rocprofv3 --stats -d gpurun_out/prof -- python benchmarks/data_plane_bench.py
+
+Emits one JSON line per op: {op, gbytes, seconds, GBps, cpu_GBps_est}.
+"""
+from __future__ import annotations
+
+import json
+import random
+import time
+
+import torch
+
+from shipyard_amd import ops
+from shipyard_amd.data import lz4py
+from shipyard_amd.ops import gf2
+
+
+def timeit(fn, warmup=2, iters=5):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters
+
+
+def bench_crc(size=1 << 30, chunk=1 << 20):
+    data = torch.randint(0, 256, (size,), dtype=torch.uint8, device="cuda")
+    mats = torch.tensor(gf2.level_matrices(chunk), dtype=torch.int64).to(
+        torch.uint32).cuda()
+    out = torch.empty(size // chunk, dtype=torch.uint32, device="cuda")
+    import ctypes
+
+    lib = ops._load()
+
+    def run():
+        lib.sy_crc32c_chunks(
+            ctypes.c_void_p(data.data_ptr()), ctypes.c_uint64(size),
+            ctypes.c_uint32(chunk), ctypes.c_void_p(mats.data_ptr()),
+            ctypes.c_void_p(out.data_ptr()),
+            ctypes.c_uint64(size // chunk), ops._stream())
+
+    sec = timeit(run)
+    # cpu single-thread reference on a 4 MiB sample
+    sample = bytes(data[:1 << 22].cpu().numpy().tobytes())
+    t0 = time.perf_counter()
+    gf2.crc32c(sample)
+    cpu_sec_per_byte = (time.perf_counter() - t0) / len(sample)
+    return {"op": "crc32c_chunks", "gbytes": size / 1e9, "seconds": sec,
+            "GBps": size / sec / 1e9,
+            "cpu_ref_GBps": 1e-9 / cpu_sec_per_byte,
+            "cpu_ref": "pure-python table crc (lower bound)"}
+
+
+def bench_sha(size=1 << 30, page=4096):
+    data = torch.randint(0, 256, (size,), dtype=torch.uint8, device="cuda")
+    out = torch.empty((size // page, 32), dtype=torch.uint8, device="cuda")
+    import ctypes
+
+    lib = ops._load()
+
+    def run():
+        lib.sy_sha256_pages(
+            ctypes.c_void_p(data.data_ptr()), ctypes.c_uint64(size),
+            ctypes.c_uint32(page), ctypes.c_void_p(out.data_ptr()),
+            ctypes.c_uint64(size // page), ops._stream())
+
+    sec = timeit(run)
+    import hashlib
+
+    sample = bytes(data[:1 << 22].cpu().numpy().tobytes())
+    t0 = time.perf_counter()
+    hashlib.sha256(sample)
+    cpu_sec_per_byte = (time.perf_counter() - t0) / len(sample)
+    return {"op": "sha256_pages", "gbytes": size / 1e9, "seconds": sec,
+            "GBps": size / sec / 1e9,
+            "cpu_ref_GBps": 1e-9 / cpu_sec_per_byte,
+            "cpu_ref": "hashlib (openssl) single thread"}
+
+
+def bench_lz4(total_raw=1 << 30, distinct=64, block_raw=64 * 1024):
+    # author `distinct` compressible blocks once on CPU, replicate the
+    # block table to reach total_raw decoded bytes
+    random.seed(7)
+    comp_blobs = []
+    for i in range(distinct):
+        raw = bytes(random.choices(b"abcdefghijklmnop", k=block_raw))
+        comp_blobs.append(lz4py.compress_block(raw))
+    offs, lens = [], []
+    payload = bytearray()
+    for c in comp_blobs:
+        offs.append(len(payload))
+        lens.append(len(c))
+        payload += c
+        payload += b"\x00" * (-len(payload) % 16)
+    n_blocks = total_raw // block_raw
+    in_off = [offs[i % distinct] for i in range(n_blocks)]
+    in_len = [lens[i % distinct] for i in range(n_blocks)]
+    out_off = [i * block_raw for i in range(n_blocks)]
+    out_len = [block_raw] * n_blocks
+
+    dev = torch.device("cuda")
+    d_comp = torch.frombuffer(payload, dtype=torch.uint8).to(dev)
+    t64 = lambda v: torch.tensor(v, dtype=torch.int64, device=dev)
+    t32 = lambda v: torch.tensor(v, dtype=torch.int64).to(torch.uint32).to(dev)
+    d_in_off, d_in_len = t64(in_off), t32(in_len)
+    d_out_off, d_out_len = t64(out_off), t32(out_len)
+    d_out = torch.empty(total_raw, dtype=torch.uint8, device=dev)
+    d_status = torch.empty(n_blocks, dtype=torch.uint32, device=dev)
+    import ctypes
+
+    lib = ops._load()
+
+    def run():
+        lib.sy_lz4_decode_blocks(
+            ctypes.c_void_p(d_comp.data_ptr()),
+            ctypes.c_void_p(d_in_off.data_ptr()),
+            ctypes.c_void_p(d_in_len.data_ptr()),
+            ctypes.c_void_p(d_out.data_ptr()),
+            ctypes.c_void_p(d_out_off.data_ptr()),
+            ctypes.c_void_p(d_out_len.data_ptr()),
+            ctypes.c_void_p(d_status.data_ptr()),
+            ctypes.c_uint32(n_blocks), ops._stream())
+
+    sec = timeit(run, warmup=1, iters=3)
+    assert ops.lz4_all_ok(d_status)
+    # verify one replicated block decodes correctly
+    got = bytes(d_out[:block_raw].cpu().numpy().tobytes())
+    want = lz4py.decompress_block(comp_blobs[0], block_raw)
+    assert got == want, "decode mismatch"
+    t0 = time.perf_counter()
+    for i in range(4):
+        lz4py.decompress_block(comp_blobs[i], block_raw)
+    cpu_sec_per_byte = (time.perf_counter() - t0) / (4 * block_raw)
+    ratio = sum(lens[:distinct]) / (distinct * block_raw)
+    return {"op": "lz4_decode_blocks", "gbytes": total_raw / 1e9,
+            "seconds": sec, "GBps": total_raw / sec / 1e9,
+            "comp_ratio": round(ratio, 3),
+            "cpu_ref_GBps": 1e-9 / cpu_sec_per_byte,
+            "cpu_ref": "pure-python decoder (lower bound)"}
+
+
+def main():
+    assert torch.cuda.is_available()
+    scale = 1  # full size on the box
+    results = [
+        bench_crc(size=(1 << 30) // scale),
+        bench_sha(size=(1 << 30) // scale),
+        bench_lz4(total_raw=(1 << 30) // scale),
+    ]
+    for r in results:
+        r["GBps"] = round(r["GBps"], 2)
+        r["cpu_ref_GBps"] = round(r["cpu_ref_GBps"], 4)
+        r["seconds"] = round(r["seconds"], 5)
+        print(json.dumps(r), flush=True)
+
+
+if __name__ == "__main__":
+    main()
