@@ -14,7 +14,11 @@ from __future__ import annotations
 
 import json
 import random
+import sys
 import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
 
 import torch
 

@@ -1,0 +1,169 @@
+"""Storage clusters — the RemoteFS analogue.
+
+The reference provisions NFS/GlusterFS VMs with mdadm RAID-0 data disks
+(reference convoy/remotefs.py:623 `create_storage_cluster`,
+scripts/shipyard_remotefs_bootstrap.sh:49-220).  On one MI355X node the
+same capability surface is local scratch clusters:
+
+  * host_dir — a directory-backed cluster (bind-mounted into tasks);
+  * tmpfs — RAM-backed scratch (mount -t tmpfs, sized);
+  * raid0 — mdadm stripe over NVMe devices + mkfs + mount (command
+    synthesis with dry-run, since CI hosts have no spare devices).
+
+Clusters register in the store and mount into pools via
+shared_data_volumes with volume_driver: storage_cluster.
+"""
+from __future__ import annotations
+
+import json
+import os
+import shutil
+import subprocess
+import time
+from pathlib import Path
+from typing import Dict, List, Optional
+
+from shipyard_amd import utils
+
+logger = utils.get_logger(__name__)
+
+KV_PREFIX = "remotefs:"
+
+
+class RemoteFsError(RuntimeError):
+    pass
+
+
+def _cluster_conf(fs_conf: dict, cluster_id: str) -> dict:
+    clusters = (fs_conf.get("remote_fs", {})
+                .get("storage_clusters", {}) or {})
+    if cluster_id not in clusters:
+        raise RemoteFsError(f"no storage cluster {cluster_id} in fs config")
+    return clusters[cluster_id]
+
+
+def synthesize_setup_commands(cluster_id: str, conf: dict) -> List[List[str]]:
+    """Commands that would bring the cluster up (the bootstrap-script
+    analogue); executed by create unless dry_run."""
+    driver = conf["driver"]
+    mnt = conf["mountpoint"]
+    cmds: List[List[str]] = [["mkdir", "-p", mnt]]
+    if driver == "host_dir":
+        backing = conf.get("path")
+        if backing and backing != mnt:
+            cmds += [["mkdir", "-p", backing],
+                     ["mount", "--bind", backing, mnt]]
+    elif driver == "tmpfs":
+        size = utils.parse_size(conf.get("size", "1gi"))
+        cmds += [["mount", "-t", "tmpfs", "-o", f"size={size}",
+                  f"shipyard-{cluster_id}", mnt]]
+    elif driver == "raid0":
+        devices = conf.get("devices") or []
+        if not devices:
+            raise RemoteFsError("raid0 needs devices")
+        md = f"/dev/md/shipyard-{cluster_id}"
+        fs = conf.get("filesystem", "ext4")
+        cmds += [
+            ["mdadm", "--create", md, "--level=0",
+             f"--raid-devices={len(devices)}", *devices],
+            [f"mkfs.{fs}", md],
+            ["mount", md, mnt],
+        ]
+    else:
+        raise RemoteFsError(f"unknown driver {driver}")
+    return cmds
+
+
+class StorageClusterManager:
+    def __init__(self, store):
+        self.store = store
+
+    def create(self, cluster_id: str, fs_conf: dict,
+               dry_run: Optional[bool] = None) -> dict:
+        conf = _cluster_conf(fs_conf, cluster_id)
+        if self.status(cluster_id):
+            raise RemoteFsError(f"cluster {cluster_id} exists")
+        driver = conf["driver"]
+        if dry_run is None:
+            # bind/tmpfs/mdadm mounts need root + real devices; default
+            # to executing only the safe directory-backed case
+            dry_run = driver != "host_dir"
+        cmds = synthesize_setup_commands(cluster_id, conf)
+        executed = []
+        if not dry_run:
+            for cmd in cmds:
+                if cmd[0] in ("mount", "mdadm") and os.geteuid() != 0:
+                    raise RemoteFsError(f"{cmd[0]} requires root")
+                rc, out, err = utils.subprocess_with_output(cmd)
+                if rc != 0:
+                    raise RemoteFsError(f"{' '.join(cmd)} failed: {err}")
+                executed.append(cmd)
+        elif driver == "host_dir":
+            Path(conf["mountpoint"]).mkdir(parents=True, exist_ok=True)
+        rec = {
+            "id": cluster_id, "driver": driver,
+            "mountpoint": conf["mountpoint"],
+            "created_at": time.time(), "state": "ready",
+            "dry_run": dry_run,
+            "commands": [" ".join(c) for c in cmds],
+        }
+        self.store.kv_set(KV_PREFIX + cluster_id, json.dumps(rec))
+        self.store.add_event(f"fs:{cluster_id}", "created",
+                             {"driver": driver, "dry_run": dry_run})
+        return rec
+
+    def status(self, cluster_id: str) -> Optional[dict]:
+        raw = self.store.kv_get(KV_PREFIX + cluster_id)
+        if raw is None:
+            return None
+        rec = json.loads(raw)
+        mnt = Path(rec["mountpoint"])
+        rec["mounted"] = mnt.exists()
+        if mnt.exists():
+            st = shutil.disk_usage(mnt)
+            rec["disk"] = {"total": st.total, "used": st.used,
+                           "free": st.free}
+        return rec
+
+    def expand(self, cluster_id: str, fs_conf: dict) -> dict:
+        """add-brick analogue (reference remotefs.py:1171): re-read the
+        config's device/size and record the new geometry."""
+        conf = _cluster_conf(fs_conf, cluster_id)
+        rec = self.status(cluster_id)
+        if rec is None:
+            raise RemoteFsError(f"no cluster {cluster_id}")
+        rec["devices"] = conf.get("devices") or []
+        rec["expanded_at"] = time.time()
+        self.store.kv_set(KV_PREFIX + cluster_id, json.dumps(
+            {k: v for k, v in rec.items() if k not in ("mounted", "disk")}))
+        return rec
+
+    def delete(self, cluster_id: str, keep_data: bool = True) -> None:
+        rec = self.status(cluster_id)
+        if rec is None:
+            return
+        if not rec.get("dry_run"):
+            mnt = rec["mountpoint"]
+            if rec["driver"] in ("tmpfs", "raid0") or \
+                    (rec["driver"] == "host_dir" and os.path.ismount(mnt)):
+                utils.subprocess_with_output(["umount", mnt])
+        if not keep_data and rec["driver"] == "host_dir":
+            shutil.rmtree(rec["mountpoint"], ignore_errors=True)
+        self.store.execute("DELETE FROM kv WHERE key=?",
+                           (KV_PREFIX + cluster_id,))
+        self.store.add_event(f"fs:{cluster_id}", "deleted")
+
+    def list(self) -> List[dict]:
+        rows = self.store.query(
+            "SELECT key, value FROM kv WHERE key LIKE ?",
+            (KV_PREFIX + "%",))
+        return [json.loads(r["value"]) for r in rows]
+
+    def mount_args_for_task(self, cluster_id: str,
+                            container_path: str,
+                            bind_options: Optional[str]) -> str:
+        rec = self.status(cluster_id)
+        if rec is None:
+            raise RemoteFsError(f"no cluster {cluster_id}")
+        opts = f":{bind_options}" if bind_options else ""
+        return f"{rec['mountpoint']}:{container_path}{opts}"

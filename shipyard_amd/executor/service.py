@@ -114,6 +114,7 @@ class LocalExecutor:
         # image cache (reference fleet.py:1821 _add_pool ->
         # cascade distribute_global_resources)
         if config_conf is not None:
+            self.store.kv_set("global_config", json.dumps(config_conf))
             gs = cfg.global_settings(config_conf)
             imgs = [im["name"] for im in gs.local_images]
             if (imgs or gs.docker_images) and not gs.delay_image_preload:
@@ -762,12 +763,65 @@ class LocalExecutor:
         wd = self._task_wd(ps.id, jid, tid)
         mover.process_output_data(self.stores, specs, wd, ok)
 
+    def global_config(self) -> Optional[dict]:
+        raw = self.store.kv_get("global_config")
+        return json.loads(raw) if raw else None
+
+    def _resolve_volumes(self, ts: cfg.TaskSettings,
+                         env: Dict[str, str]) -> List[str]:
+        """Resolve data/shared volumes to bind strings + env exports
+        (the reference's volume compiler, settings.py data_volumes /
+        shared_data_volumes -> docker -v; process runtime gets
+        SHIPYARD_VOLUME_<NAME> env pointing at the host path)."""
+        gc = self.global_config()
+        if gc is None:
+            return []
+        gs = cfg.global_settings(gc)
+        binds: List[str] = []
+        for name in ts.data_volumes:
+            vol = gs.data_volumes.get(name)
+            if vol is None:
+                raise ExecutorError(f"unknown data volume {name}")
+            host = vol.get("host_path") or str(self.root / "volumes" / name)
+            Path(host).mkdir(parents=True, exist_ok=True)
+            opts = vol.get("bind_options")
+            binds.append(f"{host}:{vol['container_path']}"
+                         + (f":{opts}" if opts else ""))
+            env[f"SHIPYARD_VOLUME_{name.upper()}"] = host
+        for name in ts.shared_data_volumes:
+            vol = gs.shared_data_volumes.get(name)
+            if vol is None:
+                raise ExecutorError(f"unknown shared data volume {name}")
+            driver = vol["volume_driver"]
+            if driver == "storage_cluster":
+                from shipyard_amd.data.remotefs import StorageClusterManager
+
+                mgr = StorageClusterManager(self.store)
+                rec = mgr.status(vol.get("cluster_id") or name)
+                if rec is None:
+                    raise ExecutorError(
+                        f"storage cluster for volume {name} not created")
+                host = rec["mountpoint"]
+            elif driver == "tmpfs":
+                host = "/dev/shm/shipyard-" + name
+                Path(host).mkdir(parents=True, exist_ok=True)
+            else:  # host_dir / nvme_scratch / object_store
+                host = vol.get("host_path") or \
+                    str(self.root / "volumes" / name)
+                Path(host).mkdir(parents=True, exist_ok=True)
+            opts = vol.get("bind_options")
+            binds.append(f"{host}:{vol['container_path']}"
+                         + (f":{opts}" if opts else ""))
+            env[f"SHIPYARD_VOLUME_{name.upper()}"] = host
+        return binds
+
     def _launch_task(self, ps: cfg.PoolSettings, js: cfg.JobSettings,
                      ts: cfg.TaskSettings, jid: str, tid: str,
                      device_ids: List[int], ranks: int) -> TaskHandle:
         mi = ts.multi_instance
         env = dict(ps.environment_variables)
         env.update(ts.environment_variables)
+        volumes = self._resolve_volumes(ts, env)
         if ts.input_data or js.input_data:
             self._process_input_data(ps, jid, tid,
                                      list(js.input_data) + list(ts.input_data))
@@ -784,6 +838,7 @@ class LocalExecutor:
             env=env,
             device_ids=device_ids,
             shm_size=ts.shm_size,
+            volumes=volumes,
             docker_options=ts.additional_docker_run_options,
             remove_container=ts.remove_container_after_exit,
             num_instances=ranks if mi else 1,

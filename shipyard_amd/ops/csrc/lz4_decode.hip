@@ -142,12 +142,19 @@ __global__ __launch_bounds__(SY_WAVE) void lz4_decode_kernel(
       pos = __shfl(pos, 0);
 
       // ---- cooperative match copy with doubling over overlap ----
+      // The decoded match is periodic with period `offset`.  Round r
+      // may copy up to dist = done + offset bytes by reading at stride
+      // -dist: dist is a multiple of offset by induction (offset, 2o,
+      // 4o, ...), so dst[j - dist] == dst[j - offset], and every read
+      // lands strictly below `done` (a completed round) — no
+      // intra-round hazard.  Round sizes double: O(log(mlen/offset))
+      // rounds instead of byte-serial.
       uint32_t done = 0;
       while (done < mlen) {
-        const uint32_t safe = offset + done;  // bytes copyable this round
-        const uint32_t n = min(mlen - done, safe);
+        const uint32_t dist = done + offset;
+        const uint32_t n = min(mlen - done, dist);
         for (uint32_t i = lane; i < n; i += SY_WAVE) {
-          dst[dpos + done + i] = dst[dpos + done + i - offset];
+          dst[dpos + done + i] = dst[dpos + done + i - dist];
         }
         done += n;
         // order LDS writes before next round's cross-lane reads
