@@ -1,0 +1,178 @@
+"""Slurm elastic resume/suspend against local pools."""
+from __future__ import annotations
+
+import json
+import re
+import time
+from pathlib import Path
+from typing import Dict, List, Optional
+
+from shipyard_amd import utils
+
+logger = utils.get_logger(__name__)
+
+KV_HOSTS = "slurm:hosts"
+
+
+def expand_hostlist(spec: str) -> List[str]:
+    """Expand 'name-[0-3],other-7' style Slurm hostlists."""
+    out: List[str] = []
+    for part in _split_top(spec):
+        m = re.fullmatch(r"(.*)\[([0-9,\-]+)\](.*)", part)
+        if not m:
+            out.append(part)
+            continue
+        prefix, ranges, suffix = m.groups()
+        for r in ranges.split(","):
+            if "-" in r:
+                lo, hi = r.split("-")
+                width = len(lo)
+                for i in range(int(lo), int(hi) + 1):
+                    out.append(f"{prefix}{str(i).zfill(width)}{suffix}")
+            else:
+                out.append(f"{prefix}{r}{suffix}")
+    return out
+
+
+def _split_top(spec: str) -> List[str]:
+    parts, depth, cur = [], 0, ""
+    for ch in spec:
+        if ch == "," and depth == 0:
+            parts.append(cur)
+            cur = ""
+            continue
+        if ch == "[":
+            depth += 1
+        elif ch == "]":
+            depth -= 1
+        cur += ch
+    if cur:
+        parts.append(cur)
+    return parts
+
+
+class SlurmAdapter:
+    def __init__(self, executor, slurm_conf: dict):
+        self.ex = executor
+        self.conf = slurm_conf["slurm"]
+        self.cluster_id = self.conf["cluster_id"]
+
+    # host <-> pool mapping -------------------------------------------
+    def _partition_for_host(self, host: str) -> Optional[dict]:
+        for pname, part in (self.conf.get("elastic_partitions") or
+                            {}).items():
+            if host.startswith(f"{self.cluster_id}-{pname}-"):
+                return {"name": pname, **part}
+        return None
+
+    def _hosts(self) -> Dict[str, dict]:
+        raw = self.ex.store.kv_get(KV_HOSTS)
+        return json.loads(raw) if raw else {}
+
+    def _save_hosts(self, hosts: Dict[str, dict]) -> None:
+        self.ex.store.kv_set(KV_HOSTS, json.dumps(hosts))
+
+    # ResumeProgram ----------------------------------------------------
+    def resume(self, hostlist: str) -> List[str]:
+        hosts = expand_hostlist(hostlist)
+        assigned = self._hosts()
+        done = []
+        for host in hosts:
+            part = self._partition_for_host(host)
+            if part is None:
+                logger.warning("no partition for host %s", host)
+                continue
+            pools = part.get("batch_pools") or {}
+            for pool_id, pconf in pools.items():
+                row = self.ex.store.query_one(
+                    "SELECT gpus_dedicated, gpus_low_priority FROM pools "
+                    "WHERE id=?", (pool_id,))
+                if row is None:
+                    continue
+                maxn = pconf.get("max_compute_nodes", 8)
+                node_type = pconf.get("compute_node_type", "dedicated")
+                cur = row["gpus_dedicated"] if node_type == "dedicated" \
+                    else row["gpus_low_priority"]
+                if cur >= maxn:
+                    continue
+                kw = {"dedicated": cur + 1} if node_type == "dedicated" \
+                    else {"low_priority": cur + 1}
+                self.ex.pool_resize(pool_id, **kw)
+                assigned[host] = {"pool": pool_id, "type": node_type,
+                                  "at": time.time()}
+                done.append(host)
+                self.ex.store.add_event(f"slurm:{host}", "resumed",
+                                        {"pool": pool_id})
+                break
+        self._save_hosts(assigned)
+        return done
+
+    # SuspendProgram ---------------------------------------------------
+    def suspend(self, hostlist: str) -> List[str]:
+        hosts = expand_hostlist(hostlist)
+        assigned = self._hosts()
+        done = []
+        for host in hosts:
+            rec = assigned.pop(host, None)
+            if rec is None:
+                continue
+            row = self.ex.store.query_one(
+                "SELECT gpus_dedicated, gpus_low_priority FROM pools "
+                "WHERE id=?", (rec["pool"],))
+            if row is None:
+                continue
+            if rec["type"] == "dedicated":
+                self.ex.pool_resize(rec["pool"], dedicated=max(
+                    row["gpus_dedicated"] - 1, 0))
+            else:
+                self.ex.pool_resize(rec["pool"], low_priority=max(
+                    row["gpus_low_priority"] - 1, 0))
+            done.append(host)
+            self.ex.store.add_event(f"slurm:{host}", "suspended",
+                                    {"pool": rec["pool"]})
+        self._save_hosts(assigned)
+        return done
+
+    def resume_failed(self, hostlist: str) -> List[str]:
+        """resume-fail path (reference slurm/slurm.py:1146): treat as
+        suspend + event."""
+        hosts = self.suspend(hostlist)
+        for h in hosts:
+            self.ex.store.add_event(f"slurm:{h}", "resume-failed")
+        return hosts
+
+
+def generate_slurm_conf(slurm_conf: dict, script_dir) -> Dict[str, str]:
+    """Emit slurm.conf fragment + Resume/Suspend programs (reference
+    shipyard_slurm_master_bootstrap.sh:637-700 writes the same trio)."""
+    sc = slurm_conf["slurm"]
+    cid = sc["cluster_id"]
+    script_dir = Path(script_dir)
+    lines = []
+    for pname, part in (sc.get("elastic_partitions") or {}).items():
+        total = sum(p.get("max_compute_nodes", 0)
+                    for p in (part.get("batch_pools") or {}).values())
+        nodes = f"{cid}-{pname}-[0-{max(total - 1, 0)}]"
+        lines.append(f"NodeName={nodes} State=CLOUD")
+        opts = " ".join(part.get("other_options") or [])
+        default = " Default=YES" if part.get("default") else ""
+        lines.append(
+            f"PartitionName={pname} Nodes={nodes}{default} "
+            f"MaxTime={part.get('max_runtime_limit') or 'INFINITE'} "
+            f"State=UP {opts}".rstrip())
+    frag = "\n".join(lines) + "\n"
+    resume = ("#!/usr/bin/env bash\n"
+              f"exec python3 -m shipyard_amd.slurm_elastic resume \"$@\"\n")
+    suspend = ("#!/usr/bin/env bash\n"
+               f"exec python3 -m shipyard_amd.slurm_elastic suspend \"$@\"\n")
+    out = {}
+    for name, content in (("slurm.conf.fragment", frag),
+                          ("resume.sh", resume),
+                          ("suspend.sh", suspend)):
+        p = script_dir / name
+        p.parent.mkdir(parents=True, exist_ok=True)
+        p.write_text(content)
+        if name.endswith(".sh"):
+            p.chmod(0o755)
+        out[name] = str(p)
+    return out

@@ -1,0 +1,118 @@
+"""Slurm elastic adapter + storage cluster tests."""
+from shipyard_amd.data.remotefs import StorageClusterManager
+from shipyard_amd.executor import LocalExecutor
+from shipyard_amd.slurm_elastic import (SlurmAdapter, expand_hostlist,
+                                        generate_slurm_conf)
+
+
+def test_expand_hostlist():
+    assert expand_hostlist("a-1") == ["a-1"]
+    assert expand_hostlist("n-[0-2]") == ["n-0", "n-1", "n-2"]
+    assert expand_hostlist("n-[00-02]") == ["n-00", "n-01", "n-02"]
+    assert expand_hostlist("n-[0,3-4],m-7") == ["n-0", "n-3", "n-4", "m-7"]
+
+
+SLURM_CONF = {"slurm": {
+    "cluster_id": "sy",
+    "elastic_partitions": {
+        "gpu": {
+            "default": True,
+            "batch_pools": {
+                "spool": {"max_compute_nodes": 3,
+                          "compute_node_type": "dedicated"}},
+        }},
+}}
+
+
+def test_resume_suspend_resizes_pool(tmp_path):
+    ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+    ex.pool_add({"pool_specification": {
+        "id": "spool", "gpus": {"dedicated": 0}, "cpu_slots": 0,
+        "node_configuration": {"rocm": {"verify": False}}}})
+    ad = SlurmAdapter(ex, SLURM_CONF)
+    done = ad.resume("sy-gpu-[0-1]")
+    assert done == ["sy-gpu-0", "sy-gpu-1"]
+    row = ex.store.query_one(
+        "SELECT gpus_dedicated FROM pools WHERE id='spool'")
+    assert row["gpus_dedicated"] == 2
+    # max_compute_nodes honored
+    ad.resume("sy-gpu-2")
+    ad.resume("sy-gpu-3")
+    row = ex.store.query_one(
+        "SELECT gpus_dedicated FROM pools WHERE id='spool'")
+    assert row["gpus_dedicated"] == 3
+    done = ad.suspend("sy-gpu-[0-1]")
+    assert len(done) == 2
+    row = ex.store.query_one(
+        "SELECT gpus_dedicated FROM pools WHERE id='spool'")
+    assert row["gpus_dedicated"] == 1
+    ex.store.close()
+
+
+def test_generate_slurm_conf(tmp_path):
+    files = generate_slurm_conf(SLURM_CONF, tmp_path / "scripts")
+    frag = (tmp_path / "scripts" / "slurm.conf.fragment").read_text()
+    assert "NodeName=sy-gpu-[0-2] State=CLOUD" in frag
+    assert "PartitionName=gpu" in frag and "Default=YES" in frag
+    assert "resume.sh" in files["resume.sh"]
+
+
+def test_storage_cluster_host_dir(tmp_path):
+    ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+    fs_conf = {"remote_fs": {"storage_clusters": {
+        "scratch": {"driver": "host_dir",
+                    "mountpoint": str(tmp_path / "mnt" / "scratch")}}}}
+    mgr = StorageClusterManager(ex.store)
+    rec = mgr.create("scratch", fs_conf)
+    assert rec["state"] == "ready"
+    st = mgr.status("scratch")
+    assert st["mounted"] and "disk" in st
+    assert mgr.list()[0]["id"] == "scratch"
+    arg = mgr.mount_args_for_task("scratch", "/scratch", "rw")
+    assert arg.endswith(":/scratch:rw")
+    mgr.delete("scratch")
+    assert mgr.status("scratch") is None
+    ex.store.close()
+
+
+def test_raid0_dry_run_synthesis(tmp_path):
+    ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+    fs_conf = {"remote_fs": {"storage_clusters": {
+        "fast": {"driver": "raid0",
+                 "devices": ["/dev/nvme1n1", "/dev/nvme2n1"],
+                 "filesystem": "xfs",
+                 "mountpoint": str(tmp_path / "mnt" / "fast")}}}}
+    mgr = StorageClusterManager(ex.store)
+    rec = mgr.create("fast", fs_conf)  # dry-run by default for raid0
+    assert rec["dry_run"]
+    assert any("mdadm --create" in c for c in rec["commands"])
+    assert any("mkfs.xfs" in c for c in rec["commands"])
+    ex.store.close()
+
+
+def test_shared_volume_storage_cluster_in_task(tmp_path):
+    ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+    mnt = tmp_path / "mnt" / "sc1"
+    fs_conf = {"remote_fs": {"storage_clusters": {
+        "sc1": {"driver": "host_dir", "mountpoint": str(mnt)}}}}
+    StorageClusterManager(ex.store).create("sc1", fs_conf)
+    ex.pool_add(
+        {"pool_specification": {
+            "id": "pv", "gpus": {"dedicated": 0}, "cpu_slots": 1,
+            "node_configuration": {"rocm": {"verify": False}}}},
+        config_conf={
+            "batch_shipyard": {"storage_account_settings": "default"},
+            "global_resources": {"volumes": {"shared_data_volumes": {
+                "sc1": {"volume_driver": "storage_cluster",
+                        "container_path": "/mnt/sc1"}}}},
+        })
+    ex.jobs_add({"job_specifications": [{
+        "id": "jv",
+        "tasks": [{"id": "t",
+                   "command": "echo vol-ok > $SHIPYARD_VOLUME_SC1/out.txt",
+                   "shared_data_volumes": ["sc1"]}],
+    }]}, "pv")
+    ex.run_until_idle(timeout=30)
+    assert ex.tasks_list("jv")[0]["state"] == "completed"
+    assert (mnt / "out.txt").read_text().strip() == "vol-ok"
+    ex.store.close()
