@@ -40,8 +40,9 @@ def timeit(fn, warmup=2, iters=5):
 
 def bench_crc(size=1 << 30, chunk=1 << 20):
     data = torch.randint(0, 256, (size,), dtype=torch.uint8, device="cuda")
-    mats = torch.tensor(gf2.level_matrices(chunk), dtype=torch.int64).to(
-        torch.uint32).cuda()
+    n_chains = gf2.pick_crc_chains(chunk)
+    mats = torch.tensor(gf2.level_matrices(chunk, 256 * n_chains),
+                        dtype=torch.int64).to(torch.uint32).cuda()
     out = torch.empty(size // chunk, dtype=torch.uint32, device="cuda")
     import ctypes
 
@@ -52,7 +53,8 @@ def bench_crc(size=1 << 30, chunk=1 << 20):
             ctypes.c_void_p(data.data_ptr()), ctypes.c_uint64(size),
             ctypes.c_uint32(chunk), ctypes.c_void_p(mats.data_ptr()),
             ctypes.c_void_p(out.data_ptr()),
-            ctypes.c_uint64(size // chunk), ops._stream())
+            ctypes.c_uint64(size // chunk),
+            ctypes.c_uint32(n_chains), ops._stream())
 
     sec = timeit(run)
     # cpu single-thread reference on a 4 MiB sample

@@ -1,0 +1,185 @@
+// rccl_allreduce_bench — the framework's mpiBench analogue (C++/HIP).
+//
+// The reference ships mpiBench/OSU recipes as multi-instance MPI tasks
+// (reference recipes/mpiBench-OpenMPI/config/docker/jobs.yaml,
+// convoy/batch.py:4362 _construct_mpi_command).  This binary is the
+// MI355X-native equivalent: RCCL all-reduce over xGMI, launched either
+//
+//   * single-process over all visible GPUs (ncclCommInitAll), or
+//   * as a gang task: one process per GPU, RANK/WORLD_SIZE from the
+//     gang launcher env, ncclUniqueId exchanged through a file
+//     (SHIPYARD_NCCL_ID_FILE) instead of mpirun/$AZ_BATCH_HOST_LIST.
+//
+// Sweeps payload sizes and reports one JSON line per size:
+//   {"bytes":B,"iters":K,"us_per_op":T,"algbw_GBps":A,"busbw_GBps":S}
+// busbw = algbw * 2*(N-1)/N (nccl-tests convention; per-link xGMI
+// bound ~153 GB/s/link on the 8-GPU hive).
+//
+// Build: shipyard_amd/comm/build_native.py (hipcc, links librccl).
+
+#include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
+
+#include <chrono>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <string>
+#include <thread>
+#include <vector>
+
+#define HIP_CHECK(x)                                                       \
+  do {                                                                     \
+    hipError_t e_ = (x);                                                   \
+    if (e_ != hipSuccess) {                                                \
+      fprintf(stderr, "HIP error %s at %s:%d\n", hipGetErrorString(e_),    \
+              __FILE__, __LINE__);                                         \
+      exit(1);                                                             \
+    }                                                                      \
+  } while (0)
+
+#define NCCL_CHECK(x)                                                      \
+  do {                                                                     \
+    ncclResult_t r_ = (x);                                                 \
+    if (r_ != ncclSuccess) {                                               \
+      fprintf(stderr, "RCCL error %s at %s:%d\n", ncclGetErrorString(r_),  \
+              __FILE__, __LINE__);                                         \
+      exit(1);                                                             \
+    }                                                                      \
+  } while (0)
+
+static int env_int(const char* k, int d) {
+  const char* v = getenv(k);
+  return v ? atoi(v) : d;
+}
+
+// uniqueId exchange through the gang's shared directory
+static void exchange_id(ncclUniqueId* id, int rank) {
+  const char* path = getenv("SHIPYARD_NCCL_ID_FILE");
+  if (!path) {
+    fprintf(stderr, "SHIPYARD_NCCL_ID_FILE required for multi-process\n");
+    exit(1);
+  }
+  std::string tmp = std::string(path) + ".tmp";
+  if (rank == 0) {
+    NCCL_CHECK(ncclGetUniqueId(id));
+    FILE* f = fopen(tmp.c_str(), "wb");
+    fwrite(id, sizeof(*id), 1, f);
+    fclose(f);
+    rename(tmp.c_str(), path);
+  } else {
+    for (int i = 0; i < 6000; ++i) {
+      FILE* f = fopen(path, "rb");
+      if (f) {
+        size_t n = fread(id, 1, sizeof(*id), f);
+        fclose(f);
+        if (n == sizeof(*id)) return;
+      }
+      std::this_thread::sleep_for(std::chrono::milliseconds(50));
+    }
+    fprintf(stderr, "timed out waiting for nccl id file\n");
+    exit(1);
+  }
+}
+
+struct Gang {
+  int nranks = 1;
+  int rank = 0;            // this process's rank (multi-process mode)
+  bool multiproc = false;
+  std::vector<ncclComm_t> comms;  // 1 (multiproc) or nranks (single)
+  std::vector<hipStream_t> streams;
+  std::vector<void*> bufs;
+};
+
+int main(int argc, char** argv) {
+  size_t min_bytes = 1 << 10, max_bytes = 256ull << 20;
+  int iters = 20, warmup = 5;
+  for (int i = 1; i < argc; ++i) {
+    if (!strcmp(argv[i], "--min") && i + 1 < argc)
+      min_bytes = strtoull(argv[++i], nullptr, 10);
+    else if (!strcmp(argv[i], "--max") && i + 1 < argc)
+      max_bytes = strtoull(argv[++i], nullptr, 10);
+    else if (!strcmp(argv[i], "--iters") && i + 1 < argc)
+      iters = atoi(argv[++i]);
+    else if (!strcmp(argv[i], "--warmup") && i + 1 < argc)
+      warmup = atoi(argv[++i]);
+  }
+
+  Gang g;
+  int world = env_int("WORLD_SIZE", 0);
+  if (world > 1) {
+    g.multiproc = true;
+    g.nranks = world;
+    g.rank = env_int("RANK", 0);
+    int dev = env_int("LOCAL_RANK", g.rank);
+    int ndev = 0;
+    HIP_CHECK(hipGetDeviceCount(&ndev));
+    HIP_CHECK(hipSetDevice(dev % ndev));
+    ncclUniqueId id;
+    exchange_id(&id, g.rank);
+    g.comms.resize(1);
+    NCCL_CHECK(ncclCommInitRank(&g.comms[0], world, id, g.rank));
+    g.streams.resize(1);
+    HIP_CHECK(hipStreamCreate(&g.streams[0]));
+    g.bufs.resize(1);
+    HIP_CHECK(hipMalloc(&g.bufs[0], max_bytes));
+    HIP_CHECK(hipMemset(g.bufs[0], 1, max_bytes));
+  } else {
+    int ndev = 0;
+    HIP_CHECK(hipGetDeviceCount(&ndev));
+    g.nranks = env_int("SHIPYARD_BENCH_GPUS", ndev);
+    if (g.nranks > ndev) g.nranks = ndev;
+    g.comms.resize(g.nranks);
+    NCCL_CHECK(ncclCommInitAll(g.comms.data(), g.nranks, nullptr));
+    g.streams.resize(g.nranks);
+    g.bufs.resize(g.nranks);
+    for (int i = 0; i < g.nranks; ++i) {
+      HIP_CHECK(hipSetDevice(i));
+      HIP_CHECK(hipStreamCreate(&g.streams[i]));
+      HIP_CHECK(hipMalloc(&g.bufs[i], max_bytes));
+      HIP_CHECK(hipMemset(g.bufs[i], 1, max_bytes));
+    }
+  }
+
+  auto run_once = [&](size_t bytes) {
+    size_t count = bytes / sizeof(float);
+    if (g.multiproc) {
+      NCCL_CHECK(ncclAllReduce(g.bufs[0], g.bufs[0], count, ncclFloat,
+                               ncclSum, g.comms[0], g.streams[0]));
+    } else {
+      NCCL_CHECK(ncclGroupStart());
+      for (int i = 0; i < g.nranks; ++i) {
+        NCCL_CHECK(ncclAllReduce(g.bufs[i], g.bufs[i], count, ncclFloat,
+                                 ncclSum, g.comms[i], g.streams[i]));
+      }
+      NCCL_CHECK(ncclGroupEnd());
+    }
+  };
+  auto sync_all = [&] {
+    for (auto s : g.streams) HIP_CHECK(hipStreamSynchronize(s));
+  };
+
+  for (size_t bytes = min_bytes; bytes <= max_bytes; bytes <<= 1) {
+    for (int i = 0; i < warmup; ++i) run_once(bytes);
+    sync_all();
+    auto t0 = std::chrono::steady_clock::now();
+    for (int i = 0; i < iters; ++i) run_once(bytes);
+    sync_all();
+    auto t1 = std::chrono::steady_clock::now();
+    double us =
+        std::chrono::duration<double, std::micro>(t1 - t0).count() / iters;
+    double alg = bytes / (us * 1e-6) / 1e9;
+    double bus = g.nranks > 1 ? alg * 2.0 * (g.nranks - 1) / g.nranks : alg;
+    if (g.rank == 0) {
+      printf("{\"bytes\":%zu,\"iters\":%d,\"nranks\":%d,"
+             "\"us_per_op\":%.2f,\"algbw_GBps\":%.2f,\"busbw_GBps\":%.2f,"
+             "\"mode\":\"%s\"}\n",
+             bytes, iters, g.nranks, us, alg, bus,
+             g.multiproc ? "gang" : "single-process");
+      fflush(stdout);
+    }
+  }
+
+  for (auto c : g.comms) ncclCommDestroy(c);
+  return 0;
+}

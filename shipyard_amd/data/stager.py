@@ -1,0 +1,173 @@
+"""Shard stager: NVMe -> pinned host -> HBM3E pipeline with GPU
+decode + verify.
+
+The MI355X-native replacement for the reference's blobxfer-based task
+input staging (reference convoy/data.py:879-951) and the data hot path
+of BASELINE.md metric #3: files stream through two pinned staging
+buffers with `hipMemcpyAsync` (torch non_blocking copies) on a
+dedicated stream, overlapping disk reads with H2D; SYSHARD payloads are
+then LZ4-decoded and CRC32C-verified in HBM by the HIP kernels — the
+decoded tensor is already resident where the consuming GPU task wants
+it (288 GB HBM3E), with no CPU inflate/hash pass.
+"""
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass
+from pathlib import Path
+from typing import Dict, List, Optional
+
+from shipyard_amd import utils
+from shipyard_amd.data import shardfmt
+
+logger = utils.get_logger(__name__)
+
+
+@dataclass
+class StageResult:
+    path: str
+    file_bytes: int
+    raw_bytes: int
+    seconds: float
+    decoded: bool
+    verified: bool
+
+    @property
+    def gbps(self) -> float:
+        return self.raw_bytes / self.seconds / 1e9 if self.seconds else 0.0
+
+
+class ShardStager:
+    def __init__(self, device=None, staging_mb: int = 64,
+                 verify: bool = True):
+        import torch
+
+        self.torch = torch
+        self.device = device or torch.device("cuda",
+                                             torch.cuda.current_device())
+        self.staging_bytes = staging_mb << 20
+        self.verify = verify
+        self.stream = torch.cuda.Stream(device=self.device)
+        self._pinned = [
+            torch.empty(self.staging_bytes, dtype=torch.uint8,
+                        pin_memory=True)
+            for _ in range(2)
+        ]
+
+    def _upload(self, f, total: int):
+        """Double-buffered read->H2D of `total` bytes from open file
+        `f`; returns the device tensor."""
+        torch = self.torch
+        dev_buf = torch.empty(max(total, 1), dtype=torch.uint8,
+                              device=self.device)
+        off = 0
+        idx = 0
+        events = [torch.cuda.Event(), torch.cuda.Event()]
+        first = [True, True]
+        while off < total:
+            n = min(self.staging_bytes, total - off)
+            pin = self._pinned[idx]
+            # wait until this pinned buffer's previous H2D retired
+            if not first[idx]:
+                events[idx].synchronize()
+            first[idx] = False
+            mv = memoryview(pin.numpy())[:n]
+            got = f.readinto(mv)
+            assert got == n, "short read"
+            with torch.cuda.stream(self.stream):
+                dev_buf[off:off + n].copy_(pin[:n], non_blocking=True)
+                events[idx].record(self.stream)
+            off += n
+            idx ^= 1
+        self.stream.synchronize()
+        return dev_buf
+
+    def stage_file(self, path) -> "tuple":
+        """Stage one file into HBM.  SYSHARD files decode+verify on the
+        GPU; plain files upload as-is.  Returns (tensor, StageResult).
+        """
+        torch = self.torch
+        p = Path(path)
+        size = p.stat().st_size
+        t0 = time.perf_counter()
+        with open(p, "rb") as f:
+            head = f.read(shardfmt.HEADER.size)
+            is_shard = head[:8] == shardfmt.MAGIC
+            if not is_shard:
+                f.seek(0)
+                out = self._upload(f, size)
+                sec = time.perf_counter() - t0
+                return out, StageResult(str(p), size, size, sec,
+                                        decoded=False, verified=False)
+            _, flags, block_raw, raw_size, n_blocks = \
+                shardfmt.HEADER.unpack(head)
+            table = f.read(shardfmt.ENTRY.size * n_blocks)
+            payload_total = size - shardfmt.HEADER.size - len(table)
+            d_payload = self._upload(f, payload_total)
+
+        # parse table on host (tiny), decode + verify in HBM
+        idx = shardfmt.ShardIndex(
+            block_raw=block_raw, raw_size=raw_size,
+            blocks=[shardfmt.BlockEntry(*shardfmt.ENTRY.unpack_from(
+                table, i * shardfmt.ENTRY.size)) for i in range(n_blocks)],
+            payload_off=0)
+        out = self._decode(d_payload, idx)
+        sec = time.perf_counter() - t0
+        res = StageResult(str(p), size, raw_size, sec, decoded=True,
+                          verified=self.verify)
+        logger.info("staged %s: %d -> %d bytes in %.3fs (%.2f GB/s raw)",
+                    p.name, size, raw_size, sec, res.gbps)
+        return out, res
+
+    def _decode(self, d_payload, idx: shardfmt.ShardIndex):
+        import torch
+
+        from shipyard_amd import ops
+
+        out = torch.empty(max(idx.raw_size, 1), dtype=torch.uint8,
+                          device=self.device)
+        raw_offs = []
+        acc = 0
+        for b in idx.blocks:
+            raw_offs.append(acc)
+            acc += b.raw_len
+        lz4_blocks = [(i, b) for i, b in enumerate(idx.blocks)
+                      if not b.stored]
+        stored = [(i, b) for i, b in enumerate(idx.blocks) if b.stored]
+        with torch.cuda.stream(self.stream):
+            for i, b in stored:
+                out[raw_offs[i]:raw_offs[i] + b.raw_len] = \
+                    d_payload[b.comp_off:b.comp_off + b.comp_len]
+        self.stream.synchronize()
+        if lz4_blocks:
+            mk64 = lambda v: torch.tensor(v, dtype=torch.int64,
+                                          device=self.device)
+            mk32 = lambda v: torch.tensor(v, dtype=torch.int64).to(
+                torch.uint32).to(self.device)
+            with torch.cuda.stream(self.stream):
+                status = ops.lz4_decode_blocks(
+                    d_payload, mk64([b.comp_off for _, b in lz4_blocks]),
+                    mk32([b.comp_len for _, b in lz4_blocks]), out,
+                    mk64([raw_offs[i] for i, _ in lz4_blocks]),
+                    mk32([b.raw_len for _, b in lz4_blocks]))
+            self.stream.synchronize()
+            if not ops.lz4_all_ok(status):
+                raise ValueError("GPU decode failed in stager")
+        if self.verify and idx.blocks:
+            with torch.cuda.stream(self.stream):
+                crcs = ops.crc32c_chunks(out[:idx.raw_size],
+                                         chunk_size=idx.block_raw)
+            self.stream.synchronize()
+            want = [b.crc32c for b in idx.blocks]
+            got = [int(x) for x in crcs.tolist()]
+            if got != want:
+                raise ValueError("CRC mismatch in staged shard")
+        return out[:idx.raw_size]
+
+    def stage_many(self, paths: List) -> Dict[str, StageResult]:
+        out = {}
+        for p in paths:
+            tensor, res = self.stage_file(p)
+            out[str(p)] = res
+            del tensor
+        return out

@@ -50,7 +50,8 @@ def _load() -> ctypes.CDLL:
     lib.sy_crc32c_chunks.restype = ctypes.c_int
     lib.sy_crc32c_chunks.argtypes = [
         ctypes.c_void_p, ctypes.c_uint64, ctypes.c_uint32,
-        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_uint64, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_uint64,
+        ctypes.c_uint32, ctypes.c_void_p,
     ]
     lib.sy_lz4_decode_blocks.restype = ctypes.c_int
     lib.sy_lz4_decode_blocks.argtypes = [
@@ -91,13 +92,16 @@ def crc32c_chunks(data, chunk_size: int = 1 << 20, finish: bool = True):
     assert data.dtype == torch.uint8 and data.is_cuda and data.is_contiguous()
     n = data.numel()
     n_chunks = (n + chunk_size - 1) // chunk_size
-    mats = torch.tensor(gf2.level_matrices(chunk_size), dtype=torch.int64)
+    n_chains = gf2.pick_crc_chains(chunk_size)
+    mats = torch.tensor(gf2.level_matrices(chunk_size, 256 * n_chains),
+                        dtype=torch.int64)
     d_mats = mats.to(torch.uint32).to(data.device)
     out = torch.empty(n_chunks, dtype=torch.uint32, device=data.device)
     rc = lib.sy_crc32c_chunks(
         ctypes.c_void_p(data.data_ptr()), ctypes.c_uint64(n),
         ctypes.c_uint32(chunk_size), ctypes.c_void_p(d_mats.data_ptr()),
-        ctypes.c_void_p(out.data_ptr()), ctypes.c_uint64(n_chunks), _stream())
+        ctypes.c_void_p(out.data_ptr()), ctypes.c_uint64(n_chunks),
+        ctypes.c_uint32(n_chains), _stream())
     _check(rc, "sy_crc32c_chunks")
     raw = out.cpu()
     if not finish:

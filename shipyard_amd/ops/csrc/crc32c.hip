@@ -1,34 +1,31 @@
-// CRC32C (Castagnoli) chunked kernel for MI355X (gfx950).
+// CRC32C (Castagnoli) chunked kernel for MI355X (gfx950) — v2.
 //
 // Role in the framework: integrity manifests for the data mover and the
-// cascade-analogue layer cache.  The reference computes MD5/SHA256 on the
-// CPU around its bulk transfers (reference convoy/util.py:461-508,
-// convoy/data.py:636-660 file-split transfers); here the per-chunk
-// integrity pass runs on the GPU at HBM streaming rate so staging into
-// 288 GB HBM3E and verification are one pass.
+// cascade-analogue layer cache (the reference hashes on the CPU around
+// its bulk transfers — reference convoy/util.py:461-508,
+// convoy/data.py:636-660).  The per-chunk integrity pass runs on the
+// GPU at HBM streaming rate so staging into 288 GB HBM3E and
+// verification are one pass.
 //
-// Design (CDNA4-first, not a CPU port):
-//  * One 256-thread workgroup per chunk; each lane owns a contiguous
-//    SEG = chunk_size/256 byte segment and walks it with uint4 (16 B)
-//    loads.  Per-lane sequential walks keep one 128 B line per lane hot
-//    in the CU's L1 (64 lanes x 128 B = 8 KiB), so full lines are
-//    consumed even though lanes stride by SEG.
-//  * Slice-by-8 tables live in LDS (8 x 256 x 4 B = 8 KiB), generated at
-//    workgroup start (parallel across lanes; no host table upload).
-//  * Lane CRCs combine in a log2(256)=8 level LDS tree using GF(2)
-//    "shift by L zero bytes" operators.  The 8 per-level 32x32 bit
-//    matrices (L = SEG * 2^k) are precomputed on the host (they depend
-//    only on chunk_size) and passed in as 8*32 uint32 words.
-//  * Ragged final chunk: CRC is linear and leading zero bytes are
-//    identity under a zero register, so the short chunk is treated as
-//    zero-padded at the FRONT.  Lanes before the message start simply
-//    contribute raw CRC 0; the boundary lane skips the padded prefix.
-//    The kernel therefore emits RAW (init=0, no final xor) CRCs; the
-//    host wrapper applies init/final-xor via one GF(2) vector shift per
-//    distinct chunk length (shipyard_amd/ops/gf2.py).
+// Design (CDNA4-first):
+//  * One 256-thread workgroup per chunk.  Each lane owns NCHAINS
+//    contiguous segments and walks them with uint4 (16 B) loads,
+//    keeping NCHAINS independent CRC chains in flight — the slice-by-8
+//    table chain is serial per segment (one LDS-latency round trip per
+//    8 B), so interleaved chains are the ILP lever (v1 ran one chain
+//    per lane and measured 1.17 TB/s; the chain latency, not LDS
+//    bandwidth, was the bound).
+//  * Slice-by-8 tables live in LDS (8 KiB), generated at workgroup
+//    start.
+//  * Lane chains combine in-register with the first log2(NCHAINS)
+//    GF(2) shift operators, then a log2(256)-level LDS tree combines
+//    lanes; the 32x32 bit operators (shift by seg * 2^k zero bytes)
+//    are host-precomputed per chunk_size (shipyard_amd/ops/gf2.py).
+//  * Ragged final chunk: raw CRC (init 0) ignores leading zero bytes,
+//    so the short chunk is treated as zero-padded at the FRONT; the
+//    host wrapper applies init/final-xor per true length.
 //
-// Grid sizing: n_chunks workgroups (>> 256 for real transfers: a 1 GiB
-// staging buffer at 1 MiB chunks is 1024 workgroups over 256 CUs).
+// The kernel emits RAW (init=0, no final xor) CRCs.
 
 #include "common.h"
 
@@ -42,33 +39,31 @@ __device__ __forceinline__ uint32_t gf2_matvec(const uint32_t* __restrict__ m,
   uint32_t r = 0;
 #pragma unroll
   for (int i = 0; i < 32; ++i) {
-    // column i of the operator applies when bit i of v is set
     r ^= (v >> i & 1u) ? m[i] : 0u;
   }
   return r;
 }
 
+template <int NCHAINS, int NLEVELS>
 __global__ __launch_bounds__(kThreads) void crc32c_chunks_kernel(
     const uint8_t* __restrict__ data, uint64_t n_bytes, uint32_t chunk_size,
-    const uint32_t* __restrict__ level_mats,  // [8][32]
+    const uint32_t* __restrict__ level_mats,  // [NLEVELS][32]
     uint32_t* __restrict__ out_raw, uint64_t n_chunks) {
   __shared__ uint32_t tab[8][256];
   __shared__ uint32_t lane_crc[kThreads];
-  __shared__ uint32_t mats[8][32];
+  __shared__ uint32_t mats[NLEVELS][32];
 
   const int t = threadIdx.x;
 
   // ---- build slice-by-8 tables in LDS ----
-  // t0: one entry per thread (256 entries / 256 threads)
   {
     uint32_t c = (uint32_t)t;
 #pragma unroll
     for (int k = 0; k < 8; ++k) c = (c >> 1) ^ (kPoly & (0u - (c & 1u)));
     tab[0][t] = c;
   }
-  if (t < 8 * 32 / kThreads * kThreads && t < 256) {
-    // copy combine matrices (8*32 = 256 words, one per thread)
-    mats[t >> 5][t & 31] = level_mats[t];
+  for (int i = t; i < NLEVELS * 32; i += kThreads) {
+    mats[i / 32][i % 32] = level_mats[i];
   }
   __syncthreads();
   for (int k = 1; k < 8; ++k) {
@@ -77,56 +72,101 @@ __global__ __launch_bounds__(kThreads) void crc32c_chunks_kernel(
     __syncthreads();
   }
 
-  const uint64_t seg = chunk_size / kThreads;  // enforced %16==0 by host
+  const uint32_t n_segs = kThreads * NCHAINS;
+  const uint64_t seg = chunk_size / n_segs;  // %16==0 enforced by host
 
   for (uint64_t chunk = blockIdx.x; chunk < n_chunks; chunk += gridDim.x) {
     const uint64_t cbeg = chunk * (uint64_t)chunk_size;
     const uint64_t cend = min(cbeg + chunk_size, n_bytes);
     const uint64_t clen = cend - cbeg;
-    // virtual front padding so every chunk spans exactly chunk_size
-    const uint64_t pad = chunk_size - clen;
+    const uint64_t pad = chunk_size - clen;  // virtual front padding
 
-    // lane t's virtual segment: [t*seg, (t+1)*seg) in padded space;
-    // maps to data [cbeg + t*seg - pad, ...) where positive.
-    uint64_t vbeg = (uint64_t)t * seg;
-    uint64_t vend = vbeg + seg;
-    uint32_t crc = 0;
-    if (vend > pad) {
-      uint64_t dbeg = cbeg + (vbeg > pad ? vbeg - pad : 0);
-      uint64_t dend = cbeg + (vend - pad);
-      // head: bytes until 16-aligned relative position
-      const uint8_t* p = data + dbeg;
-      uint64_t nseg = dend - dbeg;
-      // byte-wise until pointer 16-aligned (ragged boundary lane only;
-      // full lanes are 16-aligned when data is, since seg % 16 == 0)
-      while (nseg && ((uintptr_t)p & 15u)) {
-        crc = (crc >> 8) ^ tab[0][(crc ^ *p++) & 0xffu];
-        --nseg;
-      }
-      while (nseg >= 16) {
-        const uint4 v = *reinterpret_cast<const uint4*>(p);
-        uint32_t a = v.x ^ crc, b = v.y;
-        crc = tab[7][a & 0xffu] ^ tab[6][(a >> 8) & 0xffu] ^
-              tab[5][(a >> 16) & 0xffu] ^ tab[4][a >> 24] ^
-              tab[3][b & 0xffu] ^ tab[2][(b >> 8) & 0xffu] ^
-              tab[1][(b >> 16) & 0xffu] ^ tab[0][b >> 24];
-        a = v.z ^ crc; b = v.w;
-        crc = tab[7][a & 0xffu] ^ tab[6][(a >> 8) & 0xffu] ^
-              tab[5][(a >> 16) & 0xffu] ^ tab[4][a >> 24] ^
-              tab[3][b & 0xffu] ^ tab[2][(b >> 8) & 0xffu] ^
-              tab[1][(b >> 16) & 0xffu] ^ tab[0][b >> 24];
-        p += 16;
-        nseg -= 16;
-      }
-      while (nseg) {
-        crc = (crc >> 8) ^ tab[0][(crc ^ *p++) & 0xffu];
-        --nseg;
+    uint32_t crc[NCHAINS];
+    const uint8_t* p[NCHAINS];
+    uint64_t rem[NCHAINS];
+#pragma unroll
+    for (int c = 0; c < NCHAINS; ++c) {
+      crc[c] = 0;
+      const uint64_t vbeg = ((uint64_t)t * NCHAINS + c) * seg;
+      const uint64_t vend = vbeg + seg;
+      if (vend <= pad) {
+        p[c] = nullptr;
+        rem[c] = 0;
+      } else {
+        const uint64_t dbeg = cbeg + (vbeg > pad ? vbeg - pad : 0);
+        const uint64_t dend = cbeg + (vend - pad);
+        p[c] = data + dbeg;
+        rem[c] = dend - dbeg;
       }
     }
-    lane_crc[t] = crc;
+
+    // ragged head bytes (boundary lane of the last chunk only)
+#pragma unroll
+    for (int c = 0; c < NCHAINS; ++c) {
+      while (rem[c] && ((uintptr_t)p[c] & 15u)) {
+        crc[c] = (crc[c] >> 8) ^ tab[0][(crc[c] ^ *p[c]++) & 0xffu];
+        --rem[c];
+      }
+    }
+
+    // main interleaved loop: all full lanes have rem == seg (uniform),
+    // so the hot path is branch-light; boundary lanes take the
+    // per-chain guard.
+    bool any = true;
+    while (any) {
+      any = false;
+      uint4 v[NCHAINS];
+      bool act[NCHAINS];
+#pragma unroll
+      for (int c = 0; c < NCHAINS; ++c) {
+        act[c] = rem[c] >= 16;
+        if (act[c]) v[c] = *reinterpret_cast<const uint4*>(p[c]);
+      }
+#pragma unroll
+      for (int c = 0; c < NCHAINS; ++c) {
+        if (!act[c]) continue;
+        any = true;
+        uint32_t a = v[c].x ^ crc[c], b = v[c].y;
+        crc[c] = tab[7][a & 0xffu] ^ tab[6][(a >> 8) & 0xffu] ^
+                 tab[5][(a >> 16) & 0xffu] ^ tab[4][a >> 24] ^
+                 tab[3][b & 0xffu] ^ tab[2][(b >> 8) & 0xffu] ^
+                 tab[1][(b >> 16) & 0xffu] ^ tab[0][b >> 24];
+        a = v[c].z ^ crc[c];
+        b = v[c].w;
+        crc[c] = tab[7][a & 0xffu] ^ tab[6][(a >> 8) & 0xffu] ^
+                 tab[5][(a >> 16) & 0xffu] ^ tab[4][a >> 24] ^
+                 tab[3][b & 0xffu] ^ tab[2][(b >> 8) & 0xffu] ^
+                 tab[1][(b >> 16) & 0xffu] ^ tab[0][b >> 24];
+        p[c] += 16;
+        rem[c] -= 16;
+      }
+    }
+    // tail bytes (< 16; boundary lanes only)
+#pragma unroll
+    for (int c = 0; c < NCHAINS; ++c) {
+      while (rem[c]) {
+        crc[c] = (crc[c] >> 8) ^ tab[0][(crc[c] ^ *p[c]++) & 0xffu];
+        --rem[c];
+      }
+    }
+
+    // ---- in-register combine of this lane's chains ----
+    // crc(L||R) = shift(crc(L), len(R)) ^ crc(R); level k shifts by
+    // seg * 2^k.
+    int level = 0;
+#pragma unroll
+    for (int c = 1; c < NCHAINS; c <<= 1) {
+      // combine pairs at distance c (NCHAINS is a power of two)
+      for (int base = 0; base + c < NCHAINS; base += 2 * c) {
+        crc[base] = gf2_matvec(mats[level], crc[base]) ^ crc[base + c];
+      }
+      ++level;
+    }
+
+    lane_crc[t] = crc[0];
     __syncthreads();
 
-    // ---- log-tree combine: crc(L||R) = shift(crc(L), len(R)) ^ crc(R) ----
+    // ---- log-tree combine across lanes ----
 #pragma unroll
     for (int k = 0; k < 8; ++k) {
       const int stride = 1 << k;
@@ -134,7 +174,8 @@ __global__ __launch_bounds__(kThreads) void crc32c_chunks_kernel(
       uint32_t merged = 0;
       const bool active = idx + stride < kThreads;
       if (active) {
-        merged = gf2_matvec(mats[k], lane_crc[idx]) ^ lane_crc[idx + stride];
+        merged = gf2_matvec(mats[level + k], lane_crc[idx]) ^
+                 lane_crc[idx + stride];
       }
       __syncthreads();
       if (active) lane_crc[idx] = merged;
@@ -151,13 +192,35 @@ SY_EXPORT int sy_crc32c_chunks(const void* d_data, uint64_t n_bytes,
                                uint32_t chunk_size,
                                const uint32_t* d_level_mats,
                                uint32_t* d_out_raw, uint64_t n_chunks,
-                               hipStream_t stream) {
-  if (chunk_size == 0 || chunk_size % (256 * 16) != 0) return -22;  // EINVAL
+                               uint32_t n_chains, hipStream_t stream) {
   if (n_chunks == 0) return 0;
-  // >> 256 workgroups when the payload is real; cap grid and stride.
+  const uint32_t segs = 256 * n_chains;
+  if (chunk_size == 0 || chunk_size % (segs * 16) != 0) return -22;
   uint32_t grid = (uint32_t)(n_chunks < 8192 ? n_chunks : 8192);
-  hipLaunchKernelGGL(crc32c_chunks_kernel, dim3(grid), dim3(256), 0, stream,
-                     static_cast<const uint8_t*>(d_data), n_bytes, chunk_size,
-                     d_level_mats, d_out_raw, n_chunks);
+  const uint8_t* d = static_cast<const uint8_t*>(d_data);
+  switch (n_chains) {
+    case 1:
+      hipLaunchKernelGGL((crc32c_chunks_kernel<1, 8>), dim3(grid), dim3(256),
+                         0, stream, d, n_bytes, chunk_size, d_level_mats,
+                         d_out_raw, n_chunks);
+      break;
+    case 2:
+      hipLaunchKernelGGL((crc32c_chunks_kernel<2, 9>), dim3(grid), dim3(256),
+                         0, stream, d, n_bytes, chunk_size, d_level_mats,
+                         d_out_raw, n_chunks);
+      break;
+    case 4:
+      hipLaunchKernelGGL((crc32c_chunks_kernel<4, 10>), dim3(grid), dim3(256),
+                         0, stream, d, n_bytes, chunk_size, d_level_mats,
+                         d_out_raw, n_chunks);
+      break;
+    case 8:
+      hipLaunchKernelGGL((crc32c_chunks_kernel<8, 11>), dim3(grid), dim3(256),
+                         0, stream, d, n_bytes, chunk_size, d_level_mats,
+                         d_out_raw, n_chunks);
+      break;
+    default:
+      return -22;
+  }
   return sy_check(hipGetLastError());
 }

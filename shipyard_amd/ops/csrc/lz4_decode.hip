@@ -1,4 +1,4 @@
-// LZ4 block decompression for MI355X (gfx950).
+// LZ4 block decompression for MI355X (gfx950) — v2.
 //
 // Role in the framework: the cascade-analogue image replicator and the
 // shard stager store container layers / data shards as sequences of
@@ -9,26 +9,26 @@
 // HBM without a CPU inflate pass.
 //
 // Design (CDNA4-first):
-//  * One wave (64 lanes) per block.  Lane 0 parses the sequence stream
-//    (token / LSIC lengths / offset) — parsing is inherently serial —
-//    and broadcasts via __shfl; all 64 lanes then copy literals and
-//    matches cooperatively.
-//  * Output is staged in LDS (64 KiB per wave), not global memory:
-//    match copies read bytes written by *other lanes* in previous
-//    rounds, and LDS ordering via s_waitcnt lgkmcnt(0) is cheap and
-//    wave-local, where global-memory ordering would need vmcnt drains
-//    through L2.  The decoded block then streams LDS -> HBM with
-//    coalesced 16 B stores — scattered byte writes never touch HBM.
-//  * Overlapping matches (offset < length) use the doubling schedule:
-//    round r may copy min(remaining, offset + done) bytes in parallel,
-//    so rounds grow geometrically instead of byte-serial.
-//  * LDS budget 64 KiB -> 2 concurrent blocks per CU (160 KiB LDS/CU);
-//    grid = n_blocks waves; 512 blocks in flight across 256 CUs.  This
-//    kernel is latency/parse bound per block, so throughput comes from
-//    block-level parallelism, which container layers have in abundance.
-//
-// Workgroup = 1 wave (64 threads) to keep the LDS tile per-wave private
-// (no __syncthreads needed across waves).
+//  * One wave (64 lanes) per block.  The sequence stream is parsed
+//    UNIFORMLY by all 64 lanes: every lane fetches the same header
+//    bytes, so the loads coalesce to a single broadcast L1 transaction
+//    per wave and no cross-lane shuffles are needed (v1 parsed on lane
+//    0 and broadcast 6 scalars per sequence via ds_bpermute — that
+//    chain dominated decode time).
+//  * Header bytes come from a 16 B register window aligned to the
+//    underlying buffer (one dwordx4 refill per ~16 header bytes)
+//    instead of per-byte global loads.
+//  * Output is staged in LDS (64 KiB per wave): match copies read
+//    bytes written by other lanes in previous rounds, and LDS ordering
+//    via s_waitcnt lgkmcnt(0) is cheap and wave-local.  The decoded
+//    block then streams LDS -> HBM with coalesced 16 B stores.
+//  * Overlapping matches use the doubling schedule: round r copies
+//    n = min(remaining, done+offset) bytes reading at stride
+//    -(done+offset), which is a multiple of `offset` by induction, so
+//    the periodic match pattern is preserved and every read lands in a
+//    completed round.  O(log(mlen/offset)) rounds.
+//  * LDS budget 64 KiB -> 2 concurrent blocks per CU; throughput comes
+//    from block-level parallelism (512 blocks in flight chip-wide).
 
 #include "common.h"
 
@@ -36,13 +36,38 @@ namespace {
 
 constexpr int kBlockRaw = 64 * 1024;  // max raw bytes per LZ4 block
 
-// status codes per block
 enum : uint32_t {
   SY_LZ4_OK = 0,
-  SY_LZ4_ERR_OFFSET = 1,    // match offset reaches before block start
-  SY_LZ4_ERR_OVERFLOW = 2,  // output exceeded declared raw size
-  SY_LZ4_ERR_TRUNC = 3,     // input ran out mid-sequence
-  SY_LZ4_ERR_MISMATCH = 4,  // decoded size != declared raw size
+  SY_LZ4_ERR_OFFSET = 1,
+  SY_LZ4_ERR_OVERFLOW = 2,
+  SY_LZ4_ERR_TRUNC = 3,
+  SY_LZ4_ERR_MISMATCH = 4,
+};
+
+// s_waitcnt immediate: wait lgkmcnt(0) only (vmcnt/expcnt unconstrained)
+constexpr int kWaitLgkm0 = 0xC07F;
+
+// Register byte-window over a global buffer.  `abs` positions are
+// relative to `base0`, which itself must be 16 B aligned (torch
+// allocations are 256 B aligned; callers pass the buffer start and
+// absolute offsets).  All lanes hold identical windows -> refill loads
+// broadcast.
+struct ByteWindow {
+  const uint8_t* base0;
+  uint64_t win_base = ~0ull;
+  uint4 win;
+
+  __device__ __forceinline__ uint8_t get(uint64_t apos) {
+    const uint64_t wb = apos & ~15ull;
+    if (wb != win_base) {
+      win_base = wb;
+      win = *reinterpret_cast<const uint4*>(base0 + wb);
+    }
+    const uint32_t word = (apos & 8ull)
+        ? ((apos & 4ull) ? win.w : win.z)
+        : ((apos & 4ull) ? win.y : win.x);
+    return (uint8_t)(word >> (uint32_t)((apos & 3ull) * 8u));
+  }
 };
 
 __global__ __launch_bounds__(SY_WAVE) void lz4_decode_kernel(
@@ -55,51 +80,42 @@ __global__ __launch_bounds__(SY_WAVE) void lz4_decode_kernel(
   const int lane = threadIdx.x;
 
   for (uint32_t blk = blockIdx.x; blk < n_blocks; blk += gridDim.x) {
-    const uint8_t* src = comp + in_off[blk];
+    // absolute base of this block within `comp` (comp itself 16B-aligned)
+    const uint64_t abase = in_off[blk];
+    const uint8_t* src = comp + abase;
     const uint32_t slen = in_len[blk];
     const uint32_t rawlen = out_len[blk];
     uint32_t st = SY_LZ4_OK;
 
-    // Lane 0 parses; sequence vars are broadcast each iteration.
-    uint32_t pos = 0;   // input cursor
+    ByteWindow w;
+    w.base0 = comp;
+    w.win_base = ~0ull;
+
+    // Uniform parse state (identical in every lane — no broadcasts).
+    uint32_t pos = 0;   // input cursor (relative to src)
     uint32_t dpos = 0;  // output cursor
     for (;;) {
-      uint32_t litlen = 0, mlen = 0, offset = 0, stop = 0;
-      if (lane == 0) {
-        if (pos >= slen) {
-          stop = 1;
-          if (dpos != rawlen) st = SY_LZ4_ERR_MISMATCH;
-        } else {
-          const uint32_t token = src[pos++];
-          litlen = token >> 4;
-          if (litlen == 15) {
-            uint8_t b;
-            do {
-              if (pos >= slen) { st = SY_LZ4_ERR_TRUNC; break; }
-              b = src[pos++];
-              litlen += b;
-            } while (b == 255);
-          }
-          if (st == SY_LZ4_OK) {
-            if (dpos + litlen > rawlen || pos + litlen > slen) {
-              st = SY_LZ4_ERR_OVERFLOW;
-            } else {
-              mlen = (token & 0xfu) + 4;
-              // a block's last sequence is literals-only
-              if (pos + litlen == slen) {
-                stop = 2;  // copy literals then stop
-              }
-            }
-          }
-        }
+      if (pos >= slen) {
+        if (dpos != rawlen) st = SY_LZ4_ERR_MISMATCH;
+        break;
       }
-      st = __shfl(st, 0);
-      if (st != SY_LZ4_OK) break;
-      stop = __shfl(stop, 0);
-      if (stop == 1) break;
-      litlen = __shfl(litlen, 0);
-      pos = __shfl(pos, 0);
-      dpos = __shfl(dpos, 0);
+      const uint32_t token = w.get(abase + pos);
+      ++pos;
+      uint32_t litlen = token >> 4;
+      if (litlen == 15) {
+        uint8_t b;
+        do {
+          if (pos >= slen) { st = SY_LZ4_ERR_TRUNC; break; }
+          b = w.get(abase + pos);
+          ++pos;
+          litlen += b;
+        } while (b == 255);
+        if (st != SY_LZ4_OK) break;
+      }
+      if (dpos + litlen > rawlen || pos + litlen > slen) {
+        st = SY_LZ4_ERR_OVERFLOW;
+        break;
+      }
 
       // ---- cooperative literal copy: global src -> LDS dst ----
       for (uint32_t i = lane; i < litlen; i += SY_WAVE) {
@@ -107,76 +123,61 @@ __global__ __launch_bounds__(SY_WAVE) void lz4_decode_kernel(
       }
       pos += litlen;
       dpos += litlen;
-      if (stop == 2) {
-        if (lane == 0 && dpos != rawlen) st = SY_LZ4_ERR_MISMATCH;
-        st = __shfl(st, 0);
+      if (pos == slen) {
+        // last sequence is literals-only
+        if (dpos != rawlen) st = SY_LZ4_ERR_MISMATCH;
         break;
       }
 
-      if (lane == 0) {
-        if (pos + 2 > slen) {
-          st = SY_LZ4_ERR_TRUNC;
-        } else {
-          offset = (uint32_t)src[pos] | ((uint32_t)src[pos + 1] << 8);
-          pos += 2;
-          uint32_t ml = mlen;  // token nibble + 4, merged above
-          if ((ml - 4) == 15) {
-            uint8_t b;
-            do {
-              if (pos >= slen) { st = SY_LZ4_ERR_TRUNC; break; }
-              b = src[pos++];
-              ml += b;
-            } while (b == 255);
-          }
-          mlen = ml;
-          if (st == SY_LZ4_OK) {
-            if (offset == 0 || offset > dpos) st = SY_LZ4_ERR_OFFSET;
-            else if (dpos + mlen > rawlen) st = SY_LZ4_ERR_OVERFLOW;
-          }
-        }
+      if (pos + 2 > slen) { st = SY_LZ4_ERR_TRUNC; break; }
+      const uint32_t offset =
+          (uint32_t)w.get(abase + pos) |
+          ((uint32_t)w.get(abase + pos + 1) << 8);
+      pos += 2;
+      uint32_t mlen = (token & 0xFu) + 4;
+      if ((token & 0xFu) == 15) {
+        uint8_t b;
+        do {
+          if (pos >= slen) { st = SY_LZ4_ERR_TRUNC; break; }
+          b = w.get(abase + pos);
+          ++pos;
+          mlen += b;
+        } while (b == 255);
+        if (st != SY_LZ4_OK) break;
       }
-      st = __shfl(st, 0);
-      if (st != SY_LZ4_OK) break;
-      offset = __shfl(offset, 0);
-      mlen = __shfl(mlen, 0);
-      pos = __shfl(pos, 0);
+      if (offset == 0 || offset > dpos) { st = SY_LZ4_ERR_OFFSET; break; }
+      if (dpos + mlen > rawlen) { st = SY_LZ4_ERR_OVERFLOW; break; }
 
-      // ---- cooperative match copy with doubling over overlap ----
-      // The decoded match is periodic with period `offset`.  Round r
-      // may copy up to dist = done + offset bytes by reading at stride
-      // -dist: dist is a multiple of offset by induction (offset, 2o,
-      // 4o, ...), so dst[j - dist] == dst[j - offset], and every read
-      // lands strictly below `done` (a completed round) — no
-      // intra-round hazard.  Round sizes double: O(log(mlen/offset))
-      // rounds instead of byte-serial.
+      // literal bytes just written must be visible before match reads
+      __builtin_amdgcn_s_waitcnt(kWaitLgkm0);
+
+      // ---- cooperative match copy, doubling over overlap ----
       uint32_t done = 0;
       while (done < mlen) {
-        const uint32_t dist = done + offset;
+        const uint32_t dist = done + offset;  // multiple of offset
         const uint32_t n = min(mlen - done, dist);
         for (uint32_t i = lane; i < n; i += SY_WAVE) {
           dst[dpos + done + i] = dst[dpos + done + i - dist];
         }
         done += n;
-        // order LDS writes before next round's cross-lane reads
-        __builtin_amdgcn_s_waitcnt(0 /* vmcnt=0 lgkmcnt=0 ... */);
+        __builtin_amdgcn_s_waitcnt(kWaitLgkm0);
       }
       dpos += mlen;
     }
 
     if (lane == 0) status[blk] = st;
     if (st == SY_LZ4_OK) {
+      // wait for the final literal copy's vm loads + lds writes
+      __builtin_amdgcn_s_waitcnt(0);
       // ---- stream LDS -> HBM, coalesced 16 B per lane ----
       uint8_t* g = out + out_off[blk];
-      uint32_t n16 = rawlen >> 4;
+      const uint32_t n16 = rawlen >> 4;
       const uint4* s4 = reinterpret_cast<const uint4*>(dst);
-      uint4* g4 = reinterpret_cast<uint4*>(g);  // out_off 16B-aligned (host)
+      uint4* g4 = reinterpret_cast<uint4*>(g);  // out_off 16B-aligned
       for (uint32_t i = lane; i < n16; i += SY_WAVE) g4[i] = s4[i];
       for (uint32_t i = (n16 << 4) + lane; i < rawlen; i += SY_WAVE)
         g[i] = dst[i];
     }
-    // next loop iteration reuses dst; waves are independent, but lanes
-    // must finish reading before any rewrite — wave lockstep plus the
-    // waitcnt above suffices (single-wave workgroup).
     __builtin_amdgcn_s_waitcnt(0);
   }
 }

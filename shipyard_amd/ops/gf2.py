@@ -81,16 +81,26 @@ def finish(raw: int, msg_len: int) -> int:
     return (raw ^ init_term) ^ 0xFFFFFFFF
 
 
-def level_matrices(chunk_size: int, threads: int = 256) -> List[int]:
-    """Flattened 8x32 words for crc32c.hip's combine tree."""
-    if chunk_size % (threads * 16):
-        raise ValueError("chunk_size must be a multiple of threads*16")
-    seg = chunk_size // threads
+def level_matrices(chunk_size: int, segments: int = 256) -> List[int]:
+    """Flattened [log2(segments)]x32 words for crc32c.hip's combine
+    tree (in-register chain combine + cross-lane tree).  `segments` =
+    256 threads x NCHAINS."""
+    if chunk_size % (segments * 16):
+        raise ValueError("chunk_size must be a multiple of segments*16")
+    seg = chunk_size // segments
     out: List[int] = []
-    levels = threads.bit_length() - 1  # 8 for 256
+    levels = segments.bit_length() - 1
     for k in range(levels):
         out.extend(zero_shift_operator(seg * (1 << k)))
     return out
+
+
+def pick_crc_chains(chunk_size: int) -> int:
+    """Largest interleave factor the chunk size admits (kernel ILP)."""
+    for n in (8, 4, 2, 1):
+        if chunk_size % (256 * n * 16) == 0:
+            return n
+    raise ValueError("chunk_size must be a multiple of 4096")
 
 
 # --- pure-python reference CRC32C (for CPU tests and tails) ---
