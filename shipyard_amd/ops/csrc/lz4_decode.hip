@@ -1,4 +1,4 @@
-// LZ4 block decompression for MI355X (gfx950) — v2.
+// LZ4 block decompression for MI355X (gfx950) — v3.
 //
 // Role in the framework: the cascade-analogue image replicator and the
 // shard stager store container layers / data shards as sequences of
@@ -8,33 +8,35 @@
 // a layer goes NVMe -> pinned host -> HBM (hipMemcpyAsync) -> decoded in
 // HBM without a CPU inflate pass.
 //
-// Design (CDNA4-first):
-//  * One wave (64 lanes) per block.  The sequence stream is parsed
-//    UNIFORMLY by all 64 lanes: every lane fetches the same header
-//    bytes, so the loads coalesce to a single broadcast L1 transaction
-//    per wave and no cross-lane shuffles are needed (v1 parsed on lane
-//    0 and broadcast 6 scalars per sequence via ds_bpermute — that
-//    chain dominated decode time).
-//  * Header bytes come from a 16 B register window aligned to the
-//    underlying buffer (one dwordx4 refill per ~16 header bytes)
-//    instead of per-byte global loads.
-//  * Output is staged in LDS (64 KiB per wave): match copies read
-//    bytes written by other lanes in previous rounds, and LDS ordering
-//    via s_waitcnt lgkmcnt(0) is cheap and wave-local.  The decoded
-//    block then streams LDS -> HBM with coalesced 16 B stores.
-//  * Overlapping matches use the doubling schedule: round r copies
-//    n = min(remaining, done+offset) bytes reading at stride
-//    -(done+offset), which is a multiple of `offset` by induction, so
-//    the periodic match pattern is preserved and every read lands in a
-//    completed round.  O(log(mlen/offset)) rounds.
-//  * LDS budget 64 KiB -> 2 concurrent blocks per CU; throughput comes
-//    from block-level parallelism (512 blocks in flight chip-wide).
+// Design history (measured on MI355X):
+//   v1: lane-0 parse + ds_bpermute broadcasts      -> 4.7 GB/s
+//   v2: uniform-lane parse + register byte window  -> 5.0 GB/s
+//   The bound was never the broadcasts: LZ4 decode is a serial
+//   dependency chain per block, and every literal-copy global load was
+//   a COLD HBM/L2 miss (~600 cycles) sitting on that chain.
+//   v3: stage the whole compressed block into LDS up front with one
+//   coalesced wave-wide copy (streams at full bandwidth, off the
+//   critical path), then parse + copy entirely LDS->LDS (~64-cycle
+//   dependent accesses).
+//
+// Geometry: one wave (64 lanes) per block; LDS = 64 KiB decoded output
+// + 66 KiB staged input (compressed blocks may slightly exceed raw size
+// for incompressible data) -> one workgroup per CU, 256 blocks in
+// flight chip-wide.  Throughput comes from block parallelism; the
+// framework's SYSHARD format stores incompressible blocks raw
+// ("stored"), which bypass this kernel entirely via device memcpy.
+//
+// Match copies use the doubling schedule: round r copies
+// n = min(remaining, done+offset) bytes at read stride -(done+offset)
+// (a multiple of `offset` by induction), so the periodic pattern is
+// preserved and every read lands in a completed round.
 
 #include "common.h"
 
 namespace {
 
-constexpr int kBlockRaw = 64 * 1024;  // max raw bytes per LZ4 block
+constexpr int kBlockRaw = 64 * 1024;          // max raw bytes per block
+constexpr int kSrcBuf = kBlockRaw + 1024;     // staged comp (+overhead)
 
 enum : uint32_t {
   SY_LZ4_OK = 0,
@@ -42,33 +44,11 @@ enum : uint32_t {
   SY_LZ4_ERR_OVERFLOW = 2,
   SY_LZ4_ERR_TRUNC = 3,
   SY_LZ4_ERR_MISMATCH = 4,
+  SY_LZ4_ERR_TOOBIG = 5,
 };
 
-// s_waitcnt immediate: wait lgkmcnt(0) only (vmcnt/expcnt unconstrained)
+// s_waitcnt immediate: lgkmcnt(0) only (vmcnt/expcnt unconstrained)
 constexpr int kWaitLgkm0 = 0xC07F;
-
-// Register byte-window over a global buffer.  `abs` positions are
-// relative to `base0`, which itself must be 16 B aligned (torch
-// allocations are 256 B aligned; callers pass the buffer start and
-// absolute offsets).  All lanes hold identical windows -> refill loads
-// broadcast.
-struct ByteWindow {
-  const uint8_t* base0;
-  uint64_t win_base = ~0ull;
-  uint4 win;
-
-  __device__ __forceinline__ uint8_t get(uint64_t apos) {
-    const uint64_t wb = apos & ~15ull;
-    if (wb != win_base) {
-      win_base = wb;
-      win = *reinterpret_cast<const uint4*>(base0 + wb);
-    }
-    const uint32_t word = (apos & 8ull)
-        ? ((apos & 4ull) ? win.w : win.z)
-        : ((apos & 4ull) ? win.y : win.x);
-    return (uint8_t)(word >> (uint32_t)((apos & 3ull) * 8u));
-  }
-};
 
 __global__ __launch_bounds__(SY_WAVE) void lz4_decode_kernel(
     const uint8_t* __restrict__ comp, const uint64_t* __restrict__ in_off,
@@ -76,98 +56,112 @@ __global__ __launch_bounds__(SY_WAVE) void lz4_decode_kernel(
     const uint64_t* __restrict__ out_off, const uint32_t* __restrict__ out_len,
     uint32_t* __restrict__ status, uint32_t n_blocks) {
   __shared__ uint8_t dst[kBlockRaw];
+  __shared__ uint8_t sbuf[kSrcBuf];
 
   const int lane = threadIdx.x;
 
   for (uint32_t blk = blockIdx.x; blk < n_blocks; blk += gridDim.x) {
-    // absolute base of this block within `comp` (comp itself 16B-aligned)
     const uint64_t abase = in_off[blk];
-    const uint8_t* src = comp + abase;
     const uint32_t slen = in_len[blk];
     const uint32_t rawlen = out_len[blk];
     uint32_t st = SY_LZ4_OK;
 
-    ByteWindow w;
-    w.base0 = comp;
-    w.win_base = ~0ull;
+    // -48: staging copies the 16B-aligned enclosing region (up to +15
+    // head, +15 tail rounding) — keep the last sbuf slot unwritten
+    if (slen > (uint32_t)kSrcBuf - 48 || rawlen > (uint32_t)kBlockRaw) {
+      if (lane == 0) status[blk] = SY_LZ4_ERR_TOOBIG;
+      continue;
+    }
 
-    // Uniform parse state (identical in every lane — no broadcasts).
-    uint32_t pos = 0;   // input cursor (relative to src)
-    uint32_t dpos = 0;  // output cursor
-    for (;;) {
-      if (pos >= slen) {
-        if (dpos != rawlen) st = SY_LZ4_ERR_MISMATCH;
-        break;
-      }
-      const uint32_t token = w.get(abase + pos);
-      ++pos;
-      uint32_t litlen = token >> 4;
-      if (litlen == 15) {
-        uint8_t b;
-        do {
-          if (pos >= slen) { st = SY_LZ4_ERR_TRUNC; break; }
-          b = w.get(abase + pos);
-          ++pos;
-          litlen += b;
-        } while (b == 255);
-        if (st != SY_LZ4_OK) break;
-      }
-      if (dpos + litlen > rawlen || pos + litlen > slen) {
-        st = SY_LZ4_ERR_OVERFLOW;
-        break;
-      }
+    // ---- stage compressed block into LDS (coalesced 16 B/lane) ----
+    // Copy the 16B-aligned enclosing region so loads stay aligned;
+    // the stream starts at `srcoff` inside sbuf.
+    {
+      const uint64_t astart = abase & ~15ull;
+      const uint32_t srcoff = (uint32_t)(abase - astart);
+      const uint32_t stage_bytes = srcoff + slen;
+      const uint4* g4 = reinterpret_cast<const uint4*>(comp + astart);
+      uint4* s4 = reinterpret_cast<uint4*>(sbuf);
+      const uint32_t n16 = (stage_bytes + 15) >> 4;
+      for (uint32_t i = lane; i < n16; i += SY_WAVE) s4[i] = g4[i];
+      // all lanes' vm loads -> lds writes must land before parsing
+      __builtin_amdgcn_s_waitcnt(0);
+      // parse below uses src = sbuf + srcoff
+      const uint8_t* src = sbuf + srcoff;
 
-      // ---- cooperative literal copy: global src -> LDS dst ----
-      for (uint32_t i = lane; i < litlen; i += SY_WAVE) {
-        dst[dpos + i] = src[pos + i];
-      }
-      pos += litlen;
-      dpos += litlen;
-      if (pos == slen) {
-        // last sequence is literals-only
-        if (dpos != rawlen) st = SY_LZ4_ERR_MISMATCH;
-        break;
-      }
-
-      if (pos + 2 > slen) { st = SY_LZ4_ERR_TRUNC; break; }
-      const uint32_t offset =
-          (uint32_t)w.get(abase + pos) |
-          ((uint32_t)w.get(abase + pos + 1) << 8);
-      pos += 2;
-      uint32_t mlen = (token & 0xFu) + 4;
-      if ((token & 0xFu) == 15) {
-        uint8_t b;
-        do {
-          if (pos >= slen) { st = SY_LZ4_ERR_TRUNC; break; }
-          b = w.get(abase + pos);
-          ++pos;
-          mlen += b;
-        } while (b == 255);
-        if (st != SY_LZ4_OK) break;
-      }
-      if (offset == 0 || offset > dpos) { st = SY_LZ4_ERR_OFFSET; break; }
-      if (dpos + mlen > rawlen) { st = SY_LZ4_ERR_OVERFLOW; break; }
-
-      // literal bytes just written must be visible before match reads
-      __builtin_amdgcn_s_waitcnt(kWaitLgkm0);
-
-      // ---- cooperative match copy, doubling over overlap ----
-      uint32_t done = 0;
-      while (done < mlen) {
-        const uint32_t dist = done + offset;  // multiple of offset
-        const uint32_t n = min(mlen - done, dist);
-        for (uint32_t i = lane; i < n; i += SY_WAVE) {
-          dst[dpos + done + i] = dst[dpos + done + i - dist];
+      // Uniform parse state (identical in every lane; LDS byte reads
+      // of the same address broadcast).
+      uint32_t pos = 0;
+      uint32_t dpos = 0;
+      for (;;) {
+        if (pos >= slen) {
+          if (dpos != rawlen) st = SY_LZ4_ERR_MISMATCH;
+          break;
         }
-        done += n;
+        const uint32_t token = src[pos++];
+        uint32_t litlen = token >> 4;
+        if (litlen == 15) {
+          uint8_t b;
+          do {
+            if (pos >= slen) { st = SY_LZ4_ERR_TRUNC; break; }
+            b = src[pos++];
+            litlen += b;
+          } while (b == 255);
+          if (st != SY_LZ4_OK) break;
+        }
+        if (dpos + litlen > rawlen || pos + litlen > slen) {
+          st = SY_LZ4_ERR_OVERFLOW;
+          break;
+        }
+
+        // ---- cooperative literal copy: LDS sbuf -> LDS dst ----
+        for (uint32_t i = lane; i < litlen; i += SY_WAVE) {
+          dst[dpos + i] = src[pos + i];
+        }
+        pos += litlen;
+        dpos += litlen;
+        if (pos == slen) {
+          if (dpos != rawlen) st = SY_LZ4_ERR_MISMATCH;
+          break;
+        }
+
+        if (pos + 2 > slen) { st = SY_LZ4_ERR_TRUNC; break; }
+        const uint32_t offset =
+            (uint32_t)src[pos] | ((uint32_t)src[pos + 1] << 8);
+        pos += 2;
+        uint32_t mlen = (token & 0xFu) + 4;
+        if ((token & 0xFu) == 15) {
+          uint8_t b;
+          do {
+            if (pos >= slen) { st = SY_LZ4_ERR_TRUNC; break; }
+            b = src[pos++];
+            mlen += b;
+          } while (b == 255);
+          if (st != SY_LZ4_OK) break;
+        }
+        if (offset == 0 || offset > dpos) { st = SY_LZ4_ERR_OFFSET; break; }
+        if (dpos + mlen > rawlen) { st = SY_LZ4_ERR_OVERFLOW; break; }
+
+        // literal writes must be visible before match reads
         __builtin_amdgcn_s_waitcnt(kWaitLgkm0);
+
+        // ---- cooperative match copy, doubling over overlap ----
+        uint32_t done = 0;
+        while (done < mlen) {
+          const uint32_t dist = done + offset;  // multiple of offset
+          const uint32_t n = min(mlen - done, dist);
+          for (uint32_t i = lane; i < n; i += SY_WAVE) {
+            dst[dpos + done + i] = dst[dpos + done + i - dist];
+          }
+          done += n;
+          __builtin_amdgcn_s_waitcnt(kWaitLgkm0);
+        }
+        dpos += mlen;
       }
-      dpos += mlen;
     }
 
     if (lane == 0) status[blk] = st;
     if (st == SY_LZ4_OK) {
-      // wait for the final literal copy's vm loads + lds writes
       __builtin_amdgcn_s_waitcnt(0);
       // ---- stream LDS -> HBM, coalesced 16 B per lane ----
       uint8_t* g = out + out_off[blk];

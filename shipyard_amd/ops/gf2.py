@@ -96,8 +96,16 @@ def level_matrices(chunk_size: int, segments: int = 256) -> List[int]:
 
 
 def pick_crc_chains(chunk_size: int) -> int:
-    """Largest interleave factor the chunk size admits (kernel ILP)."""
-    for n in (8, 4, 2, 1):
+    """Interleave factor for the CRC kernel.  2 chains double the
+    chain-latency ILP while keeping the per-CU line working set
+    (64 lanes x chains x 128 B) inside the 32 KiB L1; 8 chains measured
+    SLOWER (884 vs 1172 GB/s) from L1 thrash.  Overridable for sweeps
+    via SHIPYARD_CRC_CHAINS."""
+    import os
+
+    override = os.environ.get("SHIPYARD_CRC_CHAINS")
+    candidates = [int(override)] if override else [2, 1]
+    for n in candidates:
         if chunk_size % (256 * n * 16) == 0:
             return n
     raise ValueError("chunk_size must be a multiple of 4096")
