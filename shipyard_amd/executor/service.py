@@ -426,6 +426,74 @@ class LocalExecutor:
                 "UPDATE jobs SET state='terminated', completed_at=? "
                 "WHERE id=?", (time.time(), job_id))
 
+    def job_disable_requeue(self, job_id: str) -> None:
+        """`jobs disable --requeue` (reference convoy/batch.py:2102):
+        kill running tasks and return them to ready, then disable."""
+        for (jid, tid), h in list(self._handles.items()):
+            if jid == job_id:
+                h.kill()
+                del self._handles[(jid, tid)]
+                row = self.store.query_one(
+                    "SELECT slots_json FROM tasks WHERE job_id=? AND id=?",
+                    (jid, tid))
+                slots = json.loads(row["slots_json"] or "[]")
+                self._release_slots(self._job_pool(jid), slots)
+                self.store.execute(
+                    "UPDATE tasks SET state='ready', slots_json=NULL "
+                    "WHERE job_id=? AND id=?", (jid, tid))
+        self.job_disable(job_id)
+
+    def job_migrate(self, job_id: str, dest_pool: str) -> None:
+        """Job migration between pools (reference convoy/batch.py:
+        1855-1971 disable -> requeue -> patch pool)."""
+        if not self.store.query_one("SELECT id FROM pools WHERE id=?",
+                                    (dest_pool,)):
+            raise ExecutorError(f"no pool {dest_pool}")
+        self.job_disable_requeue(job_id)
+        self.store.execute("UPDATE jobs SET pool_id=? WHERE id=?",
+                           (dest_pool, job_id))
+        self.job_enable(job_id)
+        self.store.add_event(f"job:{job_id}", "migrated",
+                             {"pool": dest_pool})
+
+    def slot_offline(self, pool_id: str, slot_id: int) -> None:
+        """Manual remediation analogue of `pool nodes zap/del`
+        (reference convoy/fleet.py:3691-3879): take a slot out of
+        scheduling."""
+        self.store.execute(
+            "UPDATE slots SET state='offline' WHERE pool_id=? AND "
+            "slot_id=? AND state='idle'", (pool_id, slot_id))
+
+    def slot_online(self, pool_id: str, slot_id: int) -> None:
+        self.store.execute(
+            "UPDATE slots SET state='idle' WHERE pool_id=? AND slot_id=? "
+            "AND state='offline'", (pool_id, slot_id))
+
+    def clean_retained(self, now: Optional[float] = None) -> int:
+        """Delete task directories past their retention_time (the
+        reference's retention_time semantics; task records stay)."""
+        import shutil
+
+        now = now or time.time()
+        n = 0
+        rows = self.store.query(
+            "SELECT t.job_id, t.id, t.spec_json, t.end_time, j.pool_id "
+            "FROM tasks t JOIN jobs j ON t.job_id=j.id WHERE t.state IN "
+            "('completed','failed','cancelled') AND t.end_time IS NOT NULL")
+        for r in rows:
+            js = self._job_settings(r["job_id"])
+            ps = self._pool_settings(r["pool_id"])
+            ts = cfg.task_settings(json.loads(r["spec_json"]), js, ps)
+            if ts.retention_time is None:
+                continue
+            if now - r["end_time"] > ts.retention_time.total_seconds():
+                d = (self.pool_root(r["pool_id"]) / "jobs" / r["job_id"] /
+                     "tasks" / r["id"])
+                if d.exists():
+                    shutil.rmtree(d, ignore_errors=True)
+                    n += 1
+        return n
+
     def job_del(self, job_id: str) -> None:
         self.job_terminate(job_id)
         with self.store.transaction() as conn:

@@ -1,0 +1,102 @@
+"""Tests: job migrate / disable-requeue / slot remediation / retention
+cleanup / secrets store."""
+import time
+
+import pytest
+
+from shipyard_amd.config.secrets import SecretsStore, parse_secret_ids
+from shipyard_amd.executor import LocalExecutor
+
+
+@pytest.fixture()
+def ex(tmp_path):
+    e = LocalExecutor(tmp_path / "r", detect_gpus=False)
+    yield e
+    e.store.close()
+
+
+def mkpool(ex, pid, cpu=2):
+    ex.pool_add({"pool_specification": {
+        "id": pid, "gpus": {"dedicated": 0}, "cpu_slots": cpu,
+        "node_configuration": {"rocm": {"verify": False}}}})
+
+
+def test_job_migrate(ex):
+    mkpool(ex, "src")
+    mkpool(ex, "dst")
+    ex.jobs_add({"job_specifications": [{
+        "id": "jm", "tasks": [{"id": "t", "command": "echo migrated"}]}]},
+        "src")
+    ex.job_migrate("jm", "dst")
+    ex.run_until_idle(timeout=30)
+    t = ex.tasks_list("jm")[0]
+    assert t["state"] == "completed"
+    out = ex.task_file("dst", "jm", "t").read_text()
+    assert "migrated" in out
+
+
+def test_disable_requeue_kills_and_requeues(ex):
+    mkpool(ex, "p")
+    ex.jobs_add({"job_specifications": [{
+        "id": "jr", "tasks": [{"id": "s", "command": "sleep 30"}]}]}, "p")
+    for _ in range(200):
+        ex.schedule_once()
+        if ex.tasks_list("jr")[0]["state"] == "running":
+            break
+        time.sleep(0.02)
+    ex.job_disable_requeue("jr")
+    t = ex.tasks_list("jr")[0]
+    assert t["state"] == "ready"
+    jobs = {j["id"]: j["state"] for j in ex.jobs_list()}
+    assert jobs["jr"] == "disabled"
+    ex.job_terminate("jr")
+
+
+def test_slot_offline_blocks_scheduling(ex):
+    mkpool(ex, "p", cpu=1)
+    ex.slot_offline("p", 0)
+    ex.jobs_add({"job_specifications": [{
+        "id": "jo", "tasks": [{"id": "t", "command": "true"}]}]}, "p")
+    for _ in range(5):
+        ex.schedule_once()
+    assert ex.tasks_list("jo")[0]["state"] == "ready"
+    ex.slot_online("p", 0)
+    ex.run_until_idle(timeout=30)
+    assert ex.tasks_list("jo")[0]["state"] == "completed"
+
+
+def test_retention_cleanup(ex):
+    mkpool(ex, "p")
+    ex.jobs_add({"job_specifications": [{
+        "id": "jret",
+        "tasks": [{"id": "t", "command": "echo x",
+                   "retention_time": "00:00:01"}]}]}, "p")
+    ex.run_until_idle(timeout=30)
+    d = ex.pool_root("p") / "jobs" / "jret" / "tasks" / "t"
+    assert d.exists()
+    assert ex.clean_retained(now=time.time() + 10) == 1
+    assert not d.exists()
+
+
+def test_secrets_roundtrip(tmp_path):
+    st = SecretsStore(tmp_path / "sec.bin", passphrase="hunter2")
+    st.set("dockerhub-pw", "s3cret")
+    st.set("other", "x")
+    assert st.get("dockerhub-pw") == "s3cret"
+    assert st.list() == ["dockerhub-pw", "other"]
+    # wrong passphrase fails closed
+    bad = SecretsStore(tmp_path / "sec.bin", passphrase="wrong")
+    with pytest.raises(ValueError):
+        bad.get("dockerhub-pw")
+    assert st.delete("other")
+
+
+def test_parse_secret_ids(tmp_path):
+    st = SecretsStore(tmp_path / "sec.bin", passphrase="p")
+    st.set("reg-pw", "topsecret")
+    doc = {"credentials": {"registries": {"docker": {
+        "myreg": {"username": "u", "password_secret_id": "reg-pw"}}}}}
+    out = parse_secret_ids(doc, st)
+    reg = out["credentials"]["registries"]["docker"]["myreg"]
+    assert reg["password"] == "topsecret"
+    assert "password_secret_id" not in reg
