@@ -94,7 +94,7 @@ def bench_sha(size=1 << 30, page=4096):
             "cpu_ref": "hashlib (openssl) single thread"}
 
 
-def bench_lz4(total_raw=1 << 30, distinct=64, block_raw=64 * 1024):
+def bench_lz4(total_raw=1 << 30, distinct=64, block_raw=16 * 1024):
     # author `distinct` compressible blocks once on CPU, replicate the
     # block table to reach total_raw decoded bytes
     random.seed(7)
@@ -136,7 +136,8 @@ def bench_lz4(total_raw=1 << 30, distinct=64, block_raw=64 * 1024):
             ctypes.c_void_p(d_out_off.data_ptr()),
             ctypes.c_void_p(d_out_len.data_ptr()),
             ctypes.c_void_p(d_status.data_ptr()),
-            ctypes.c_uint32(n_blocks), ops._stream())
+            ctypes.c_uint32(n_blocks), ctypes.c_uint32(block_raw),
+            ops._stream())
 
     sec = timeit(run, warmup=1, iters=3)
     assert ops.lz4_all_ok(d_status)

@@ -149,7 +149,8 @@ class ShardStager:
                     d_payload, mk64([b.comp_off for _, b in lz4_blocks]),
                     mk32([b.comp_len for _, b in lz4_blocks]), out,
                     mk64([raw_offs[i] for i, _ in lz4_blocks]),
-                    mk32([b.raw_len for _, b in lz4_blocks]))
+                    mk32([b.raw_len for _, b in lz4_blocks]),
+                    raw_cap=idx.block_raw)
             self.stream.synchronize()
             if not ops.lz4_all_ok(status):
                 raise ValueError("GPU decode failed in stager")

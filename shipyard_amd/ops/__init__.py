@@ -57,7 +57,7 @@ def _load() -> ctypes.CDLL:
     lib.sy_lz4_decode_blocks.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_uint32,
-        ctypes.c_void_p,
+        ctypes.c_uint32, ctypes.c_void_p,
     ]
     lib.sy_sha256_pages.restype = ctypes.c_int
     lib.sy_sha256_pages.argtypes = [
@@ -135,12 +135,16 @@ def crc32c_file_digest(data, chunk_size: int = 1 << 20) -> int:
     return gf2.finish(acc, n)
 
 
-def lz4_decode_blocks(comp, in_off, in_len, out, out_off, out_len):
+def lz4_decode_blocks(comp, in_off, in_len, out, out_off, out_len,
+                      raw_cap: int = 64 * 1024):
     """Decode independent LZ4 blocks on the GPU.
 
     All tensors are CUDA: ``comp``/``out`` uint8, ``in_off``/``out_off``
     int64 (byte offsets; out offsets 16 B aligned), ``in_len``/``out_len``
-    uint32.  Returns a CUDA uint32 status tensor (0 == OK per block).
+    uint32.  ``raw_cap`` is the max raw block size (selects the LDS
+    geometry: smaller blocks -> more workgroups/CU -> more latency
+    hiding; the decoder is serial per block).  Returns a CUDA uint32
+    status tensor (0 == OK per block).
     """
     import torch
 
@@ -152,7 +156,7 @@ def lz4_decode_blocks(comp, in_off, in_len, out, out_off, out_len):
         ctypes.c_void_p(in_len.data_ptr()), ctypes.c_void_p(out.data_ptr()),
         ctypes.c_void_p(out_off.data_ptr()), ctypes.c_void_p(out_len.data_ptr()),
         ctypes.c_void_p(status.data_ptr()), ctypes.c_uint32(n_blocks),
-        _stream())
+        ctypes.c_uint32(raw_cap), _stream())
     _check(rc, "sy_lz4_decode_blocks")
     return status
 

@@ -35,8 +35,10 @@
 
 namespace {
 
-constexpr int kBlockRaw = 64 * 1024;          // max raw bytes per block
-constexpr int kSrcBuf = kBlockRaw + 1024;     // staged comp (+overhead)
+// RAWCAP instantiations: 64 KiB (1 workgroup/CU) for legacy/large
+// blocks; 16 KiB (4 workgroups/CU) is the throughput geometry — the
+// decoder is serial-latency-bound per block, so occupancy is the
+// lever.  SYSHARD defaults to 16 KiB blocks for this reason.
 
 enum : uint32_t {
   SY_LZ4_OK = 0,
@@ -50,12 +52,14 @@ enum : uint32_t {
 // s_waitcnt immediate: lgkmcnt(0) only (vmcnt/expcnt unconstrained)
 constexpr int kWaitLgkm0 = 0xC07F;
 
+template <int RAWCAP>
 __global__ __launch_bounds__(SY_WAVE) void lz4_decode_kernel(
     const uint8_t* __restrict__ comp, const uint64_t* __restrict__ in_off,
     const uint32_t* __restrict__ in_len, uint8_t* __restrict__ out,
     const uint64_t* __restrict__ out_off, const uint32_t* __restrict__ out_len,
     uint32_t* __restrict__ status, uint32_t n_blocks) {
-  __shared__ uint8_t dst[kBlockRaw];
+  constexpr int kSrcBuf = RAWCAP + 1024;  // comp may exceed raw slightly
+  __shared__ uint8_t dst[RAWCAP];
   __shared__ uint8_t sbuf[kSrcBuf];
 
   const int lane = threadIdx.x;
@@ -68,7 +72,7 @@ __global__ __launch_bounds__(SY_WAVE) void lz4_decode_kernel(
 
     // -48: staging copies the 16B-aligned enclosing region (up to +15
     // head, +15 tail rounding) — keep the last sbuf slot unwritten
-    if (slen > (uint32_t)kSrcBuf - 48 || rawlen > (uint32_t)kBlockRaw) {
+    if (slen > (uint32_t)kSrcBuf - 48 || rawlen > (uint32_t)RAWCAP) {
       if (lane == 0) status[blk] = SY_LZ4_ERR_TOOBIG;
       continue;
     }
@@ -183,12 +187,29 @@ SY_EXPORT int sy_lz4_decode_blocks(const void* d_comp, const uint64_t* d_in_off,
                                    const uint64_t* d_out_off,
                                    const uint32_t* d_out_len,
                                    uint32_t* d_status, uint32_t n_blocks,
-                                   hipStream_t stream) {
+                                   uint32_t raw_cap, hipStream_t stream) {
   if (n_blocks == 0) return 0;
-  uint32_t grid = n_blocks < 4096u ? n_blocks : 4096u;
-  hipLaunchKernelGGL(lz4_decode_kernel, dim3(grid), dim3(SY_WAVE), 0, stream,
-                     static_cast<const uint8_t*>(d_comp), d_in_off, d_in_len,
-                     static_cast<uint8_t*>(d_out), d_out_off, d_out_len,
-                     d_status, n_blocks);
+  uint32_t grid = n_blocks < 8192u ? n_blocks : 8192u;
+  const uint8_t* c = static_cast<const uint8_t*>(d_comp);
+  uint8_t* o = static_cast<uint8_t*>(d_out);
+  if (raw_cap <= 8 * 1024) {
+    hipLaunchKernelGGL((lz4_decode_kernel<8 * 1024>), dim3(grid),
+                       dim3(SY_WAVE), 0, stream, c, d_in_off, d_in_len, o,
+                       d_out_off, d_out_len, d_status, n_blocks);
+  } else if (raw_cap <= 16 * 1024) {
+    hipLaunchKernelGGL((lz4_decode_kernel<16 * 1024>), dim3(grid),
+                       dim3(SY_WAVE), 0, stream, c, d_in_off, d_in_len, o,
+                       d_out_off, d_out_len, d_status, n_blocks);
+  } else if (raw_cap <= 32 * 1024) {
+    hipLaunchKernelGGL((lz4_decode_kernel<32 * 1024>), dim3(grid),
+                       dim3(SY_WAVE), 0, stream, c, d_in_off, d_in_len, o,
+                       d_out_off, d_out_len, d_status, n_blocks);
+  } else if (raw_cap <= 64 * 1024) {
+    hipLaunchKernelGGL((lz4_decode_kernel<64 * 1024>), dim3(grid),
+                       dim3(SY_WAVE), 0, stream, c, d_in_off, d_in_len, o,
+                       d_out_off, d_out_len, d_status, n_blocks);
+  } else {
+    return -22;
+  }
   return sy_check(hipGetLastError());
 }

@@ -96,15 +96,16 @@ def level_matrices(chunk_size: int, segments: int = 256) -> List[int]:
 
 
 def pick_crc_chains(chunk_size: int) -> int:
-    """Interleave factor for the CRC kernel.  2 chains double the
-    chain-latency ILP while keeping the per-CU line working set
-    (64 lanes x chains x 128 B) inside the 32 KiB L1; 8 chains measured
-    SLOWER (884 vs 1172 GB/s) from L1 thrash.  Overridable for sweeps
-    via SHIPYARD_CRC_CHAINS."""
+    """Interleave factor for the CRC kernel.  Measured on MI355X:
+    1 chain = 1154 GB/s, 2 = 893, 4 = 873, 8 = 884 — wave-level
+    parallelism (8 waves/SIMD) already hides the chain latency, and
+    multi-chain strides thrash the 32 KiB L1 (64 lanes x chains x
+    128 B lines).  Default 1; overridable for sweeps via
+    SHIPYARD_CRC_CHAINS."""
     import os
 
     override = os.environ.get("SHIPYARD_CRC_CHAINS")
-    candidates = [int(override)] if override else [2, 1]
+    candidates = [int(override)] if override else [1]
     for n in candidates:
         if chunk_size % (256 * n * 16) == 0:
             return n
