@@ -94,7 +94,7 @@ def bench_sha(size=1 << 30, page=4096):
             "cpu_ref": "hashlib (openssl) single thread"}
 
 
-def bench_lz4(total_raw=1 << 30, distinct=64, block_raw=16 * 1024):
+def bench_lz4(total_raw=1 << 30, distinct=64, block_raw=8 * 1024):
     # author `distinct` compressible blocks once on CPU, replicate the
     # block table to reach total_raw decoded bytes
     random.seed(7)

@@ -35,10 +35,14 @@
 
 namespace {
 
-// RAWCAP instantiations: 64 KiB (1 workgroup/CU) for legacy/large
-// blocks; 16 KiB (4 workgroups/CU) is the throughput geometry — the
-// decoder is serial-latency-bound per block, so occupancy is the
-// lever.  SYSHARD defaults to 16 KiB blocks for this reason.
+// RAWCAP instantiations (measured, 1 GiB synthetic, ratio ~0.9):
+//   64 KiB -> 1 workgroup/CU  ->  4.9 GB/s
+//   32 KiB -> 2               -> 12.7
+//   16 KiB -> 4               -> 35.2
+//    8 KiB -> 9               -> 94.2
+// The decoder is a serial dependency chain per block; occupancy (more
+// concurrent blocks per CU) is the throughput lever.  SYSHARD defaults
+// to 8 KiB blocks (compression-window loss is a few %% vs 64 KiB).
 
 enum : uint32_t {
   SY_LZ4_OK = 0,
@@ -192,7 +196,11 @@ SY_EXPORT int sy_lz4_decode_blocks(const void* d_comp, const uint64_t* d_in_off,
   uint32_t grid = n_blocks < 8192u ? n_blocks : 8192u;
   const uint8_t* c = static_cast<const uint8_t*>(d_comp);
   uint8_t* o = static_cast<uint8_t*>(d_out);
-  if (raw_cap <= 8 * 1024) {
+  if (raw_cap <= 4 * 1024) {
+    hipLaunchKernelGGL((lz4_decode_kernel<4 * 1024>), dim3(grid),
+                       dim3(SY_WAVE), 0, stream, c, d_in_off, d_in_len, o,
+                       d_out_off, d_out_len, d_status, n_blocks);
+  } else if (raw_cap <= 8 * 1024) {
     hipLaunchKernelGGL((lz4_decode_kernel<8 * 1024>), dim3(grid),
                        dim3(SY_WAVE), 0, stream, c, d_in_off, d_in_len, o,
                        d_out_off, d_out_len, d_status, n_blocks);
