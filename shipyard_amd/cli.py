@@ -501,6 +501,210 @@ def fed_process(ctx, configdir, root, raw):
     ctx.emit({"processed": n})
 
 
+@jobs.command("migrate")
+@click.option("--jobid", required=True)
+@click.option("--poolid", required=True, help="destination pool")
+@_common
+@pass_ctx
+def jobs_migrate(ctx, jobid, poolid, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ctx.executor.job_migrate(jobid, poolid)
+    ctx.emit({"migrated": jobid, "pool": poolid})
+
+
+@jobs.command("requeue")
+@click.option("--jobid", required=True)
+@_common
+@pass_ctx
+def jobs_requeue(ctx, jobid, configdir, root, raw):
+    """Disable with requeue (reference `jobs disable --requeue`)."""
+    _apply(ctx, configdir, root, raw)
+    ctx.executor.job_disable_requeue(jobid)
+    ctx.emit({"requeued": jobid})
+
+
+@pool.group("nodes")
+def pool_nodes():
+    """Slot remediation (the `pool nodes` analogue)."""
+
+
+@pool_nodes.command("offline")
+@click.option("--poolid", required=True)
+@click.option("--slot", type=int, required=True)
+@_common
+@pass_ctx
+def nodes_offline(ctx, poolid, slot, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ctx.executor.slot_offline(poolid, slot)
+    ctx.emit(ctx.executor.pool_stats(poolid))
+
+
+@pool_nodes.command("online")
+@click.option("--poolid", required=True)
+@click.option("--slot", type=int, required=True)
+@_common
+@pass_ctx
+def nodes_online(ctx, poolid, slot, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ctx.executor.slot_online(poolid, slot)
+    ctx.emit(ctx.executor.pool_stats(poolid))
+
+
+# ---------------------------------------------------------------- fs
+@cli.group()
+def fs():
+    """Storage clusters (RemoteFS analogue)."""
+
+
+@fs.group("cluster")
+def fs_cluster():
+    pass
+
+
+@fs_cluster.command("add")
+@click.option("--cluster-id", required=True)
+@_common
+@pass_ctx
+def fs_cluster_add(ctx, cluster_id, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.data.remotefs import StorageClusterManager
+
+    mgr = StorageClusterManager(ctx.executor.store)
+    ctx.emit(mgr.create(cluster_id, ctx.conf(ConfigType.fs)))
+
+
+@fs_cluster.command("status")
+@click.option("--cluster-id", required=True)
+@_common
+@pass_ctx
+def fs_cluster_status(ctx, cluster_id, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.data.remotefs import StorageClusterManager
+
+    ctx.emit(StorageClusterManager(ctx.executor.store).status(cluster_id))
+
+
+@fs_cluster.command("del")
+@click.option("--cluster-id", required=True)
+@click.option("--keep-data/--no-keep-data", default=True)
+@_common
+@pass_ctx
+def fs_cluster_del(ctx, cluster_id, keep_data, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.data.remotefs import StorageClusterManager
+
+    StorageClusterManager(ctx.executor.store).delete(cluster_id,
+                                                    keep_data=keep_data)
+    ctx.emit({"deleted": cluster_id})
+
+
+@fs_cluster.command("expand")
+@click.option("--cluster-id", required=True)
+@_common
+@pass_ctx
+def fs_cluster_expand(ctx, cluster_id, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.data.remotefs import StorageClusterManager
+
+    mgr = StorageClusterManager(ctx.executor.store)
+    ctx.emit(mgr.expand(cluster_id, ctx.conf(ConfigType.fs)))
+
+
+# ---------------------------------------------------------------- slurm
+@cli.group()
+def slurm():
+    """Elastic Slurm adapter."""
+
+
+@slurm.command("generate")
+@click.option("--outdir", required=True)
+@_common
+@pass_ctx
+def slurm_generate(ctx, outdir, configdir, root, raw):
+    """Emit slurm.conf fragment + Resume/Suspend programs."""
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.slurm_elastic import generate_slurm_conf
+
+    ctx.emit(generate_slurm_conf(ctx.conf(ConfigType.slurm), outdir))
+
+
+@slurm.command("resume")
+@click.option("--hosts", required=True)
+@_common
+@pass_ctx
+def slurm_resume(ctx, hosts, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.slurm_elastic import SlurmAdapter
+
+    ad = SlurmAdapter(ctx.executor, ctx.conf(ConfigType.slurm))
+    ctx.emit({"resumed": ad.resume(hosts)})
+
+
+@slurm.command("suspend")
+@click.option("--hosts", required=True)
+@_common
+@pass_ctx
+def slurm_suspend(ctx, hosts, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.slurm_elastic import SlurmAdapter
+
+    ad = SlurmAdapter(ctx.executor, ctx.conf(ConfigType.slurm))
+    ctx.emit({"suspended": ad.suspend(hosts)})
+
+
+# ---------------------------------------------------------------- keyvault
+@cli.group()
+def keyvault():
+    """Secrets store (KeyVault analogue)."""
+
+
+def _secrets(ctx):
+    from shipyard_amd.config.secrets import SecretsStore
+
+    creds = ctx.conf(ConfigType.credentials)
+    ss = (creds.get("credentials", {}).get("secrets_store") or {})
+    path = ss.get("file") or str(Path(ctx.executor.root) / "secrets.bin")
+    return SecretsStore(path, passphrase_env=ss.get(
+        "passphrase_env", "SHIPYARD_SECRETS_PASSPHRASE"))
+
+
+@keyvault.command("set")
+@click.option("--name", required=True)
+@click.option("--value", required=True)
+@_common
+@pass_ctx
+def kv_set(ctx, name, value, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    _secrets(ctx).set(name, value)
+    ctx.emit({"set": name})
+
+
+@keyvault.command("get")
+@click.option("--name", required=True)
+@_common
+@pass_ctx
+def kv_get(ctx, name, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    click.echo(_secrets(ctx).get(name))
+
+
+@keyvault.command("list")
+@_common
+@pass_ctx
+def kv_list(ctx, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ctx.emit(_secrets(ctx).list())
+
+
+@keyvault.command("del")
+@click.option("--name", required=True)
+@_common
+@pass_ctx
+def kv_del(ctx, name, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ctx.emit({"deleted": _secrets(ctx).delete(name)})
+
+
 # ---------------------------------------------------------------- daemon
 @cli.command("daemon")
 @click.option("--idle-exit", is_flag=True,
