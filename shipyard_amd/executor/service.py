@@ -495,6 +495,15 @@ class LocalExecutor:
         return n
 
     def job_del(self, job_id: str) -> None:
+        import shutil
+
+        try:
+            pool_id = self._job_pool(job_id)
+            scratch = self.pool_root(pool_id) / "scratch" / job_id
+            if scratch.exists():
+                shutil.rmtree(scratch, ignore_errors=True)
+        except Exception:
+            pass
         self.job_terminate(job_id)
         with self.store.transaction() as conn:
             conn.execute("DELETE FROM tasks WHERE job_id=?", (job_id,))
@@ -895,6 +904,13 @@ class LocalExecutor:
                                      list(js.input_data) + list(ts.input_data))
             env["SHIPYARD_TASK_INPUT_DIR"] = str(
                 self._task_wd(ps.id, jid, tid))
+        if ps.per_job_auto_scratch:
+            # per-job distributed-scratch analogue (reference
+            # shipyard_auto_scratch.sh / BeeOND): one shared fast dir
+            # per job, torn down with the job
+            scratch = self.pool_root(ps.id) / "scratch" / jid
+            scratch.mkdir(parents=True, exist_ok=True)
+            env["SHIPYARD_AUTO_SCRATCH_DIR"] = str(scratch)
         spec = LaunchSpec(
             pool_id=ps.id,
             job_id=jid,
