@@ -213,6 +213,7 @@ def _base_env(spec: LaunchSpec, paths: TaskPaths,
     pp = env.get("PYTHONPATH", "")
     if pkg_root not in pp.split(os.pathsep):
         env["PYTHONPATH"] = (pkg_root + (os.pathsep + pp if pp else ""))
+    env["SHIPYARD_REPO_ROOT"] = pkg_root
     return env
 
 
