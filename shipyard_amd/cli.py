@@ -705,6 +705,37 @@ def kv_del(ctx, name, configdir, root, raw):
     ctx.emit({"deleted": _secrets(ctx).delete(name)})
 
 
+# ---------------------------------------------------------------- misc
+@cli.group()
+def misc():
+    """Miscellaneous: tensorboard on a task's logs (reference
+    convoy/misc.py:62 tunnel_tensorboard, local edition)."""
+
+
+@misc.command("tensorboard")
+@click.option("--jobid", required=True)
+@click.option("--taskid", required=True)
+@click.option("--logdir", default="wd", help="subdir of the task dir")
+@click.option("--port", type=int, default=6006)
+@_common
+@pass_ctx
+def misc_tensorboard(ctx, jobid, taskid, logdir, port, configdir, root,
+                     raw):  # pragma: no cover - interactive
+    _apply(ctx, configdir, root, raw)
+    import shutil as _shutil
+    import subprocess as _sp
+
+    pool_id = ctx.executor._job_pool(jobid)
+    path = (ctx.executor.pool_root(pool_id) / "jobs" / jobid / "tasks" /
+            taskid / logdir)
+    if _shutil.which("tensorboard") is None:
+        raise click.ClickException(
+            f"tensorboard not installed; logdir is {path}")
+    click.echo(f"serving tensorboard on :{port} for {path}")
+    _sp.run(["tensorboard", "--logdir", str(path), "--port", str(port),
+             "--bind_all"])
+
+
 # ---------------------------------------------------------------- daemon
 @cli.command("daemon")
 @click.option("--idle-exit", is_flag=True,
