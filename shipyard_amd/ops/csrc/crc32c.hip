@@ -163,25 +163,27 @@ __global__ __launch_bounds__(kThreads) void crc32c_chunks_kernel(
       ++level;
     }
 
-    lane_crc[t] = crc[0];
-    __syncthreads();
-
-    // ---- log-tree combine across lanes ----
+    // ---- in-wave shfl tree (6 levels, no barriers) ----
+    // adjacent segments live on adjacent lanes, so level k combines
+    // lane pairs at distance 2^k; every lane computes (cheap), only
+    // the aligned lane's value is meaningful.  This replaces the v1
+    // LDS tree whose 16 __syncthreads per chunk dominated at small
+    // chunk sizes (measured: 1 MiB chunks 1.1 TB/s vs 4 MiB 2.05 —
+    // the inner loop itself sustains ~2 TB/s).
+    uint32_t acc = crc[0];
 #pragma unroll
-    for (int k = 0; k < 8; ++k) {
-      const int stride = 1 << k;
-      const int idx = t * (stride << 1);
-      uint32_t merged = 0;
-      const bool active = idx + stride < kThreads;
-      if (active) {
-        merged = gf2_matvec(mats[level + k], lane_crc[idx]) ^
-                 lane_crc[idx + stride];
-      }
-      __syncthreads();
-      if (active) lane_crc[idx] = merged;
-      __syncthreads();
+    for (int k = 0; k < 6; ++k) {
+      const uint32_t other = __shfl_down(acc, 1 << k);
+      acc = gf2_matvec(mats[level + k], acc) ^ other;
     }
-    if (t == 0) out_raw[chunk] = lane_crc[0];
+    // ---- cross-wave combine (kThreads/64 = 4 partials, 2 levels) ----
+    if ((t & 63) == 0) lane_crc[t >> 6] = acc;
+    __syncthreads();
+    if (t == 0) {
+      uint32_t a01 = gf2_matvec(mats[level + 6], lane_crc[0]) ^ lane_crc[1];
+      uint32_t a23 = gf2_matvec(mats[level + 6], lane_crc[2]) ^ lane_crc[3];
+      out_raw[chunk] = gf2_matvec(mats[level + 7], a01) ^ a23;
+    }
     __syncthreads();
   }
 }
