@@ -546,6 +546,62 @@ class LocalExecutor:
         self._complete_auto_jobs()
         return n
 
+    def start_scheduler(self, poll: float = 0.02,
+                        autoscale: bool = True) -> None:
+        """Run the scheduling loop in a background thread (the library
+        equivalent of `shipyard daemon`).  Includes autoscale
+        evaluation; stop with stop_scheduler()."""
+        import threading
+
+        if getattr(self, "_sched_thread", None):
+            return
+        self._sched_stop = threading.Event()
+
+        def loop():
+            from shipyard_amd.executor.autoscale import AutoscaleController
+
+            controllers = {}
+            while not self._sched_stop.is_set():
+                try:
+                    self.schedule_once()
+                    if autoscale:
+                        now = time.time()
+                        for p in self.pool_list():
+                            pid = p["id"]
+                            if pid not in controllers:
+                                controllers[pid] = AutoscaleController(
+                                    self, pid,
+                                    self._pool_settings(pid).autoscale)
+                            controllers[pid].maybe_evaluate(now)
+                except Exception as exc:  # keep the loop alive
+                    logger.error("scheduler loop error: %s", exc)
+                self._sched_stop.wait(poll)
+
+        self._sched_thread = threading.Thread(target=loop, daemon=True,
+                                              name="shipyard-scheduler")
+        self._sched_thread.start()
+
+    def stop_scheduler(self) -> None:
+        t = getattr(self, "_sched_thread", None)
+        if t:
+            self._sched_stop.set()
+            t.join(timeout=10)
+            self._sched_thread = None
+
+    def wait_for_job(self, job_id: str, timeout: float = 300.0,
+                     poll: float = 0.05) -> None:
+        """Query-only wait (safe while the scheduler thread runs)."""
+        deadline = time.monotonic() + timeout
+        while True:
+            row = self.store.query_one(
+                "SELECT COUNT(*) n FROM tasks WHERE job_id=? AND state IN "
+                "('pending','ready','running')", (job_id,))
+            if row["n"] == 0:
+                return
+            if time.monotonic() > deadline:
+                raise TimeoutError(f"job {job_id} did not finish")
+            time.sleep(poll)
+
     def run_until_idle(self, timeout: Optional[float] = None,
                        poll: float = 0.02) -> None:
         deadline = None if timeout is None else time.monotonic() + timeout

@@ -100,3 +100,18 @@ def test_parse_secret_ids(tmp_path):
     reg = out["credentials"]["registries"]["docker"]["myreg"]
     assert reg["password"] == "topsecret"
     assert "password_secret_id" not in reg
+
+
+def test_background_scheduler_thread(tmp_path):
+    ex = LocalExecutor(tmp_path / "bg", detect_gpus=False)
+    mkpool(ex, "bgp")
+    ex.start_scheduler(poll=0.01)
+    try:
+        ex.jobs_add({"job_specifications": [{
+            "id": "bgj",
+            "tasks": [{"id": "t", "command": "echo bg-ok"}]}]}, "bgp")
+        ex.wait_for_job("bgj", timeout=30)
+        assert ex.tasks_list("bgj")[0]["state"] == "completed"
+    finally:
+        ex.stop_scheduler()
+        ex.store.close()
