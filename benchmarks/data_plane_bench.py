@@ -38,7 +38,7 @@ def timeit(fn, warmup=2, iters=5):
     return (time.perf_counter() - t0) / iters
 
 
-def bench_crc(size=1 << 30, chunk=1 << 20):
+def bench_crc(size=1 << 30, chunk=256 * 1024):
     data = torch.randint(0, 256, (size,), dtype=torch.uint8, device="cuda")
     n_chains = gf2.pick_crc_chains(chunk)
     mats = torch.tensor(gf2.level_matrices(chunk, 256 * n_chains),

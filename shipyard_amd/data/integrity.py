@@ -17,7 +17,7 @@ from typing import List, Optional
 
 from shipyard_amd.ops import gf2
 
-DEFAULT_CHUNK = 1 << 20  # 1 MiB
+DEFAULT_CHUNK = 256 * 1024  # 256 KiB: peak GPU CRC geometry
 
 
 @dataclass

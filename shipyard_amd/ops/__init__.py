@@ -79,7 +79,7 @@ def _check(rc: int, what: str) -> None:
         raise RuntimeError(f"{what} failed with code {rc}")
 
 
-def crc32c_chunks(data, chunk_size: int = 1 << 20, finish: bool = True):
+def crc32c_chunks(data, chunk_size: int = 256 * 1024, finish: bool = True):
     """CRC32C of each ``chunk_size`` slice of a uint8 CUDA tensor.
 
     Returns a CPU torch.uint32 tensor of per-chunk CRCs (standard
@@ -120,7 +120,7 @@ def crc32c_chunks(data, chunk_size: int = 1 << 20, finish: bool = True):
     return fin
 
 
-def crc32c_file_digest(data, chunk_size: int = 1 << 20) -> int:
+def crc32c_file_digest(data, chunk_size: int = 256 * 1024) -> int:
     """Whole-buffer CRC32C via GPU chunk CRCs + host GF(2) combine."""
     import torch
 

@@ -200,17 +200,16 @@ SY_EXPORT int sy_crc32c_chunks(const void* d_data, uint64_t n_bytes,
   const uint32_t segs = 256 * n_chains;
   if (chunk_size == 0 || chunk_size % (segs * 16) != 0) return -22;
   // Grid cap trades TLP against L1 working set: each resident wave
-  // walks 64 lanes x 128 B lines = 8 KiB of L1; 1 workgroup/CU
-  // (4 waves = 32 KiB) exactly fits the vector L1, and grid-striding
-  // the remaining chunks keeps lines warm.  Measured on MI355X at
-  // 64 KiB chunks: cap 8192 -> 1.82 TB/s, cap 256 (1 WG/CU, L1-fit)
-  // is the candidate peak (cf. 4 MiB chunks at natural 1 WG/CU:
-  // 2.03 TB/s).  Overridable for sweeps via SY_CRC_GRID.
+  // walks 64 lanes x 128 B lines = 8 KiB of L1 per wave.  Measured on
+  // MI355X (1 GiB, 64K/256K chunks): cap 128 -> 0.9/1.2 TB/s,
+  // 256 -> 1.8/2.3, 512 -> 2.4/2.7 (peak; 2 WG/CU = 8 waves, 64 KiB
+  // lines), 1024 -> 2.1/1.9, uncapped -> 1.8/1.4 (L1 thrash).
+  // Overridable for sweeps via SY_CRC_GRID.
   static uint32_t grid_cap = 0;
   if (grid_cap == 0) {
     const char* e = getenv("SY_CRC_GRID");
-    grid_cap = e ? (uint32_t)atoi(e) : 256;
-    if (grid_cap == 0) grid_cap = 256;
+    grid_cap = e ? (uint32_t)atoi(e) : 512;
+    if (grid_cap == 0) grid_cap = 512;
   }
   uint32_t grid = (uint32_t)(n_chunks < grid_cap ? n_chunks : grid_cap);
   const uint8_t* d = static_cast<const uint8_t*>(d_data);
