@@ -71,6 +71,78 @@ class GangComm:
         if self.is_distributed:
             self._dist.all_reduce(tensor)
 
+    def all_reduce_bucketed_(self, tensors, bucket_bytes: int = 64 << 20
+                             ) -> None:
+        """All-reduce a list of tensors in flat buckets.
+
+        xGMI is per-link bound (7 p2p links x ~153 GB/s), so buckets
+        should be large enough to amortize ring latency but small
+        enough to overlap with producers; 64 MiB default.  Buckets are
+        flattened/coalesced then scattered back (the DDP gradient
+        pattern, without requiring torch DDP)."""
+        if not self.is_distributed:
+            return
+        import torch
+
+        bucket, size = [], 0
+        for t in list(tensors) + [None]:
+            flush = t is None or (size + t.numel() * t.element_size()
+                                  > bucket_bytes and bucket)
+            if flush and bucket:
+                flat = torch.cat([b.reshape(-1) for b in bucket])
+                self._dist.all_reduce(flat)
+                off = 0
+                for b in bucket:
+                    n = b.numel()
+                    b.copy_(flat[off:off + n].view_as(b))
+                    off += n
+                bucket, size = [], 0
+            if t is not None:
+                bucket.append(t)
+                size += t.numel() * t.element_size()
+
+    def all_gather(self, tensor):
+        """Gather each rank's tensor -> list of world tensors."""
+        import torch
+
+        if not self.is_distributed:
+            return [tensor]
+        out = [torch.empty_like(tensor) for _ in range(self.world)]
+        self._dist.all_gather(out, tensor)
+        return out
+
+    def reduce_scatter_(self, out, shards) -> None:
+        """Reduce a per-rank list of shards, scattering shard i to
+        rank i (ZeRO/FSDP pattern)."""
+        if not self.is_distributed:
+            out.copy_(shards[self.rank])
+            return
+        self._dist.reduce_scatter(out, list(shards))
+
+    def all_to_all(self, shards):
+        """Exchange shard i with rank i (EP dispatch pattern).
+
+        RCCL supports all_to_all natively; gloo does not, so the CPU
+        test path emulates it with an all_gather of the stacked shards
+        (correct, just not bandwidth-optimal — fine for gloo tests)."""
+        import torch
+
+        if not self.is_distributed:
+            return list(shards)
+        if self.backend == "nccl":
+            out = [torch.empty_like(s) for s in shards]
+            self._dist.all_to_all(out, list(shards))
+            return out
+        stacked = torch.stack(list(shards))  # [world, ...]
+        gathered = [torch.empty_like(stacked) for _ in range(self.world)]
+        self._dist.all_gather(gathered, stacked)
+        return [gathered[src][self.rank].clone()
+                for src in range(self.world)]
+
+    def broadcast_(self, tensor, src: int = 0) -> None:
+        if self.is_distributed:
+            self._dist.broadcast(tensor, src=src)
+
     def barrier(self) -> None:
         if self.is_distributed:
             if self.backend == "nccl":
