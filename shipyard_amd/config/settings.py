@@ -235,6 +235,8 @@ class TaskSettings:
     default_working_dir: str
     task_factory: Optional[dict]
     labels: List[str]
+    rocprof: bool = False
+    rocprof_options: Tuple[str, ...] = ()
 
 
 @dataclasses.dataclass(frozen=True)
@@ -430,6 +432,9 @@ def task_settings(taskspec: Dict[str, Any], job: JobSettings,
                                          job.default_working_dir),
         task_factory=taskspec.get("task_factory"),
         labels=list(taskspec.get("labels") or []),
+        rocprof=_get(taskspec, "rocprof", "enabled", default=False),
+        rocprof_options=tuple(_get(taskspec, "rocprof", "options",
+                                   default=[])),
     )
 
 

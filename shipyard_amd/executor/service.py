@@ -967,6 +967,22 @@ class LocalExecutor:
             scratch = self.pool_root(ps.id) / "scratch" / jid
             scratch.mkdir(parents=True, exist_ok=True)
             env["SHIPYARD_AUTO_SCRATCH_DIR"] = str(scratch)
+        wrapper: List[str] = []
+        if ts.rocprof:
+            # per-task kernel tracing (the cascade-perf analogue for the
+            # compute plane); rocprofv3 needs a writable TMPDIR
+            import shutil as _sh
+
+            if _sh.which("rocprofv3"):
+                prof_dir = (self.pool_root(ps.id) / "jobs" / jid / "tasks" /
+                            tid / "prof")
+                prof_dir.mkdir(parents=True, exist_ok=True)
+                opts = list(ts.rocprof_options) or ["--kernel-trace",
+                                                    "--stats"]
+                wrapper = ["rocprofv3", *opts, "-d", str(prof_dir), "--"]
+                env.setdefault("TMPDIR", "/tmp")
+            else:
+                logger.warning("rocprof requested but rocprofv3 missing")
         spec = LaunchSpec(
             pool_id=ps.id,
             job_id=jid,
@@ -988,6 +1004,7 @@ class LocalExecutor:
             pre_execution_command=mi.pre_execution_command if mi else None,
             max_wall_time_s=(ts.max_wall_time.total_seconds()
                              if ts.max_wall_time else None),
+            wrapper=wrapper,
         )
         return launch(spec, self.pool_root(ps.id))
 

@@ -78,6 +78,9 @@ class LaunchSpec:
     master_port: Optional[int] = None
     pre_execution_command: Optional[str] = None
     max_wall_time_s: Optional[float] = None
+    # command-vector prefix, e.g. ["rocprofv3", ..., "--"] for per-task
+    # kernel tracing (vector level: no shell quoting hazards)
+    wrapper: List[str] = field(default_factory=list)
 
 
 @dataclass
@@ -288,7 +291,7 @@ def launch(spec: LaunchSpec, pool_root: Path) -> TaskHandle:
 def _runtime_cmd(spec: LaunchSpec, command: str, paths: TaskPaths,
                  env: Dict[str, str]) -> List[str]:
     if spec.runtime == "process":
-        return rt.process_run_command(command)
+        return list(spec.wrapper) + rt.process_run_command(command)
     if spec.runtime == "docker":
         if not rt.runtime_available("docker"):
             raise RuntimeError("docker runtime requested but not installed")
