@@ -56,54 +56,6 @@ enum : uint32_t {
 // s_waitcnt immediate: lgkmcnt(0) only (vmcnt/expcnt unconstrained)
 constexpr int kWaitLgkm0 = 0xC07F;
 
-// Unaligned 32-bit read from an LDS byte array via funnel shift: the
-// two covering words are read and bytes [a, a+4) extracted.  Reading
-// up to 7 bytes past `a` is safe in LDS (no faults; extra bits are
-// shifted out) — buffers carry slack for this.
-__device__ __forceinline__ uint32_t lds_read_u32_unaligned(
-    const uint8_t* base, uint32_t a) {
-  const uint32_t* w = reinterpret_cast<const uint32_t*>(
-      __builtin_assume_aligned(base, 4));
-  const uint32_t lo = w[a >> 2];
-  const uint32_t sh = (a & 3u) * 8u;
-  if (sh == 0) return lo;
-  const uint32_t hi = w[(a >> 2) + 1];
-  return (lo >> sh) | (hi << (32 - sh));
-}
-
-// Wave-cooperative copy of n bytes from one LDS byte array to another
-// (dst region 4-byte-alignable; src arbitrary).  Head bytes until dst
-// is word-aligned, then 4 B per lane per round, then tail.  For long
-// literal/match runs this is ~4x the byte path (ds ops move 4 B/lane
-// instead of 1).  Caller handles ordering (lgkm waits).
-__device__ __forceinline__ void lds_copy_bytes(
-    uint8_t* __restrict__ dstb, uint32_t doff,
-    const uint8_t* __restrict__ srcb, uint32_t soff, uint32_t n,
-    int lane) {
-  // Threshold 128: typical LZ4 runs are 5-30 B where one byte-path
-  // wave iteration beats the head/word/tail fragmentation (measured:
-  // threshold 16 cost 15-20% overall); the wide path pays only on
-  // long literal runs (stored-ish content).
-  if (n < 128) {
-    for (uint32_t i = lane; i < n; i += SY_WAVE) {
-      dstb[doff + i] = srcb[soff + i];
-    }
-    return;
-  }
-  const uint32_t head = ((4u - (doff & 3u)) & 3u);
-  for (uint32_t i = lane; i < head; i += SY_WAVE) {
-    dstb[doff + i] = srcb[soff + i];
-  }
-  const uint32_t n4 = (n - head) >> 2;
-  uint32_t* d32 = reinterpret_cast<uint32_t*>(dstb + doff + head);
-  for (uint32_t i = lane; i < n4; i += SY_WAVE) {
-    d32[i] = lds_read_u32_unaligned(srcb, soff + head + i * 4);
-  }
-  for (uint32_t i = head + (n4 << 2) + lane; i < n; i += SY_WAVE) {
-    dstb[doff + i] = srcb[soff + i];
-  }
-}
-
 template <int RAWCAP>
 __global__ __launch_bounds__(SY_WAVE) void lz4_decode_kernel(
     const uint8_t* __restrict__ comp, const uint64_t* __restrict__ in_off,
@@ -171,7 +123,12 @@ __global__ __launch_bounds__(SY_WAVE) void lz4_decode_kernel(
         }
 
         // ---- cooperative literal copy: LDS sbuf -> LDS dst ----
-        lds_copy_bytes(dst, dpos, sbuf, srcoff + pos, litlen, lane);
+        // (byte-granular; funnel-shift wide copies measured 15-20%
+        // SLOWER overall — typical runs are 5-30 B and the head/word/
+        // tail fragmentation costs more than 4 B/lane saves)
+        for (uint32_t i = lane; i < litlen; i += SY_WAVE) {
+          dst[dpos + i] = src[pos + i];
+        }
         pos += litlen;
         dpos += litlen;
         if (pos == slen) {
@@ -200,15 +157,13 @@ __global__ __launch_bounds__(SY_WAVE) void lz4_decode_kernel(
         __builtin_amdgcn_s_waitcnt(kWaitLgkm0);
 
         // ---- cooperative match copy, doubling over overlap ----
-        // Wide reads may touch up to 7 bytes past the source range;
-        // those bytes are shifted out of the extracted word, so a
-        // concurrent same-round write to them is harmless.
         uint32_t done = 0;
         while (done < mlen) {
           const uint32_t dist = done + offset;  // multiple of offset
           const uint32_t n = min(mlen - done, dist);
-          lds_copy_bytes(dst, dpos + done, dst, dpos + done - dist, n,
-                         lane);
+          for (uint32_t i = lane; i < n; i += SY_WAVE) {
+            dst[dpos + done + i] = dst[dpos + done + i - dist];
+          }
           done += n;
           __builtin_amdgcn_s_waitcnt(kWaitLgkm0);
         }
