@@ -7,7 +7,7 @@ cascade/graph.py timelines) over the store's perf table.
 from __future__ import annotations
 
 import json
-from typing import List, Optional
+from typing import List
 
 from shipyard_amd.executor.store import Store
 

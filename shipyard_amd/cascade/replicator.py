@@ -25,7 +25,7 @@ import os
 import time
 from concurrent.futures import ThreadPoolExecutor
 from pathlib import Path
-from typing import Callable, Dict, List, Optional, Sequence
+from typing import Callable, List, Optional, Sequence
 
 from shipyard_amd import utils
 from shipyard_amd.data import shardfmt

@@ -10,7 +10,6 @@ from __future__ import annotations
 
 import json
 import os
-import sys
 import time
 from pathlib import Path
 
@@ -748,7 +747,6 @@ def daemon(ctx, idle_exit, interval, configdir, root, raw):
     recurrences (the local stand-in for the Azure Batch service)."""
     _apply(ctx, configdir, root, raw)
     ex = ctx.executor
-    from shipyard_amd.config import settings as cfg
     from shipyard_amd.executor.autoscale import AutoscaleController
 
     controllers = {}

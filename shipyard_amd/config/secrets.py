@@ -13,7 +13,6 @@ at-rest credentials on shared nodes; it is not a KMS.
 """
 from __future__ import annotations
 
-import base64
 import hashlib
 import hmac
 import json

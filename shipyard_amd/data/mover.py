@@ -20,7 +20,6 @@ from __future__ import annotations
 
 import concurrent.futures as cf
 import fnmatch
-import os
 import time
 from dataclasses import dataclass
 from pathlib import Path

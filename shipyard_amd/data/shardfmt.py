@@ -26,7 +26,7 @@ import io
 import struct
 from dataclasses import dataclass
 from pathlib import Path
-from typing import List, Optional, Tuple
+from typing import List
 
 from shipyard_amd.data import lz4py
 from shipyard_amd.ops import gf2

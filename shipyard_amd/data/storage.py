@@ -12,7 +12,7 @@ from __future__ import annotations
 import fnmatch
 import shutil
 from pathlib import Path
-from typing import Iterator, List, Optional, Tuple
+from typing import Iterator, List, Optional
 
 from shipyard_amd.data import integrity, shardfmt
 

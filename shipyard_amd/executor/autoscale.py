@@ -16,7 +16,7 @@ from __future__ import annotations
 
 import datetime
 from dataclasses import dataclass
-from typing import Optional, Tuple
+from typing import Optional
 
 from shipyard_amd.config.settings import (AutoscaleScenarioSettings,
                                           AutoscaleSettings)

@@ -16,7 +16,7 @@ from __future__ import annotations
 
 import ctypes
 from pathlib import Path
-from typing import Optional, Tuple
+from typing import Optional
 
 from . import gf2
 

@@ -24,7 +24,7 @@ import subprocess
 import time
 from dataclasses import dataclass, field
 from pathlib import Path
-from typing import Dict, List, Optional, Sequence
+from typing import Dict, List, Optional
 
 from shipyard_amd import utils
 from shipyard_amd.runner import runtime as rt
