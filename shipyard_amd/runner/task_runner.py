@@ -104,6 +104,30 @@ class TaskHandle:
         self.timed_out = False
         self._first_fail: Optional[int] = None
 
+    def _maybe_inject_fault(self) -> None:
+        """Fault injection for resilience tests (the reference has
+        none — SURVEY.md §5).  SHIPYARD_FAULT_INJECT in the task env:
+        ``kill_rank:<rank>:after:<seconds>`` SIGKILLs that rank once
+        elapsed, simulating a died/wedged rank."""
+        spec = self.spec.env.get("SHIPYARD_FAULT_INJECT")
+        if not spec or getattr(self, "_fault_done", False):
+            return
+        try:
+            parts = str(spec).split(":")
+            if parts[0] != "kill_rank":
+                return
+            rank, after = int(parts[1]), float(parts[3])
+        except (IndexError, ValueError):
+            return
+        if time.monotonic() - self.start_time >= after:
+            self._fault_done = True
+            for r in self.ranks:
+                if r.rank == rank and r.proc.poll() is None:
+                    try:
+                        os.killpg(r.proc.pid, signal.SIGKILL)
+                    except (ProcessLookupError, PermissionError):
+                        pass
+
     def poll(self) -> Optional[int]:
         """None while running; aggregate exit code when done.
 
@@ -111,6 +135,7 @@ class TaskHandle:
         non-zero exit the remaining ranks are torn down; the reported
         exit code is the ORIGINATING rank's, not the SIGTERM codes of
         the torn-down peers."""
+        self._maybe_inject_fault()
         codes = [r.proc.poll() for r in self.ranks]
         if any(c is not None and c != 0 for c in codes):
             if self._first_fail is None:

@@ -115,3 +115,30 @@ def test_background_scheduler_thread(tmp_path):
     finally:
         ex.stop_scheduler()
         ex.store.close()
+
+
+def test_fault_injection_kills_gang_rank(tmp_path):
+    """Chaos: injected rank death tears down the gang (resilience the
+    reference cannot test without real Azure)."""
+    ex = LocalExecutor(tmp_path / "fi", detect_gpus=False)
+    ex.pool_add({"pool_specification": {
+        "id": "fip", "gpus": {"dedicated": 0}, "cpu_slots": 3,
+        "node_configuration": {"rocm": {"verify": False}},
+        "inter_node_communication_enabled": True}})
+    ex.jobs_add({"job_specifications": [{
+        "id": "fij",
+        "tasks": [{"id": "g", "command": "sleep 30",
+                   "max_task_retries": 0,
+                   "environment_variables": {
+                       "SHIPYARD_FAULT_INJECT": "kill_rank:1:after:0.3"},
+                   "multi_instance": {"num_instances": 3,
+                                      "gang": {"backend": "gloo",
+                                               "gpus_per_rank": 0}}}]}]},
+        "fip")
+    t0 = time.time()
+    ex.run_until_idle(timeout=60)
+    assert time.time() - t0 < 20
+    t = ex.tasks_list("fij")[0]
+    assert t["state"] == "failed"
+    assert t["exit_code"] == -9  # SIGKILL of the injected rank
+    ex.store.close()
