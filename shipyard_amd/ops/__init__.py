@@ -64,6 +64,12 @@ def _load() -> ctypes.CDLL:
         ctypes.c_void_p, ctypes.c_uint64, ctypes.c_uint32, ctypes.c_void_p,
         ctypes.c_uint64, ctypes.c_void_p,
     ]
+    lib.sy_stage_file.restype = ctypes.c_int
+    lib.sy_stage_file.argtypes = [
+        ctypes.c_char_p, ctypes.c_void_p, ctypes.c_uint64, ctypes.c_uint64,
+        ctypes.c_uint64, ctypes.c_int, ctypes.c_void_p,
+        ctypes.POINTER(ctypes.c_double),
+    ]
     _lib = lib
     return lib
 
@@ -184,6 +190,30 @@ def sha256_pages(data, page_size: int = 4096):
         ctypes.c_uint64(n_pages), _stream())
     _check(rc, "sy_sha256_pages")
     return out
+
+
+def stage_file_native(path, dst, file_off: int = 0,
+                      n_bytes: int = None, staging_mb: int = 64,
+                      use_direct: bool = True) -> float:
+    """Native C++ staging pipeline: pread (O_DIRECT when possible) ->
+    pinned ring -> hipMemcpyAsync into the uint8 CUDA tensor ``dst``.
+    Returns pipeline seconds."""
+    import os as _os
+
+    lib = _load()
+    if n_bytes is None:
+        n_bytes = _os.path.getsize(path) - file_off
+    assert dst.numel() >= n_bytes
+    secs = ctypes.c_double(0.0)
+    rc = lib.sy_stage_file(
+        str(path).encode(), ctypes.c_void_p(dst.data_ptr()),
+        ctypes.c_uint64(file_off), ctypes.c_uint64(n_bytes),
+        ctypes.c_uint64(staging_mb << 20), ctypes.c_int(int(use_direct)),
+        _stream(), ctypes.byref(secs))
+    if rc != 0:
+        raise RuntimeError(f"sy_stage_file failed rc={rc} "
+                           f"({_os.strerror(-rc) if rc < 0 else 'hip'})")
+    return secs.value
 
 
 def require_native() -> None:
