@@ -39,20 +39,41 @@ class StageResult:
 
 class ShardStager:
     def __init__(self, device=None, staging_mb: int = 64,
-                 verify: bool = True):
+                 verify: bool = True, native: bool = True):
+        """``native=True`` uses the C++ pipeline (sy_stage_file:
+        O_DIRECT pread -> pinned ring -> hipMemcpyAsync) for the file
+        upload; False keeps the pure-python double buffer."""
         import torch
 
         self.torch = torch
         self.device = device or torch.device("cuda",
                                              torch.cuda.current_device())
+        self.staging_mb = staging_mb
         self.staging_bytes = staging_mb << 20
         self.verify = verify
+        self.native = native
         self.stream = torch.cuda.Stream(device=self.device)
-        self._pinned = [
-            torch.empty(self.staging_bytes, dtype=torch.uint8,
-                        pin_memory=True)
-            for _ in range(2)
-        ]
+        self._pinned = None
+        if not native:
+            self._pinned = [
+                torch.empty(self.staging_bytes, dtype=torch.uint8,
+                            pin_memory=True)
+                for _ in range(2)
+            ]
+
+    def _upload_native(self, path, file_off: int, total: int):
+        import torch
+
+        from shipyard_amd import ops
+
+        dev_buf = torch.empty(max(total, 1), dtype=torch.uint8,
+                              device=self.device)
+        if total:
+            with torch.cuda.stream(self.stream):
+                ops.stage_file_native(path, dev_buf, file_off=file_off,
+                                      n_bytes=total,
+                                      staging_mb=self.staging_mb)
+        return dev_buf
 
     def _upload(self, f, total: int):
         """Double-buffered read->H2D of `total` bytes from open file
@@ -94,8 +115,11 @@ class ShardStager:
             head = f.read(shardfmt.HEADER.size)
             is_shard = head[:8] == shardfmt.MAGIC
             if not is_shard:
-                f.seek(0)
-                out = self._upload(f, size)
+                if self.native:
+                    out = self._upload_native(p, 0, size)
+                else:
+                    f.seek(0)
+                    out = self._upload(f, size)
                 sec = time.perf_counter() - t0
                 return out, StageResult(str(p), size, size, sec,
                                         decoded=False, verified=False)
@@ -103,7 +127,10 @@ class ShardStager:
                 shardfmt.HEADER.unpack(head)
             table = f.read(shardfmt.ENTRY.size * n_blocks)
             payload_total = size - shardfmt.HEADER.size - len(table)
-            d_payload = self._upload(f, payload_total)
+            if self.native:
+                d_payload = self._upload_native(p, f.tell(), payload_total)
+            else:
+                d_payload = self._upload(f, payload_total)
 
         # parse table on host (tiny), decode + verify in HBM
         idx = shardfmt.ShardIndex(

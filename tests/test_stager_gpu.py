@@ -13,13 +13,15 @@ from shipyard_amd.data import shardfmt  # noqa: E402
 from shipyard_amd.data.stager import ShardStager  # noqa: E402
 
 
-def test_stage_shard_file(tmp_path):
+@pytest.mark.parametrize("native", [True, False],
+                         ids=["native-cxx", "python"])
+def test_stage_shard_file(tmp_path, native):
     random.seed(5)
     data = (bytes(random.choices(b"abcdefgh", k=500_000)) +
             os.urandom(300_000))
     src = tmp_path / "shard.syshard"
     src.write_bytes(shardfmt.pack(data))
-    st = ShardStager(staging_mb=1, verify=True)  # force multiple windows
+    st = ShardStager(staging_mb=1, verify=True, native=native)
     tensor, res = st.stage_file(src)
     torch.cuda.synchronize()
     assert res.decoded and res.verified
