@@ -94,7 +94,7 @@ def bench_sha(size=1 << 30, page=4096):
             "cpu_ref": "hashlib (openssl) single thread"}
 
 
-def bench_lz4(total_raw=1 << 30, distinct=64, block_raw=8 * 1024):
+def bench_lz4(total_raw=1 << 30, distinct=64, block_raw=8 * 1024, pc=False):
     # author `distinct` compressible blocks once on CPU, replicate the
     # block table to reach total_raw decoded bytes
     random.seed(7)
@@ -127,8 +127,10 @@ def bench_lz4(total_raw=1 << 30, distinct=64, block_raw=8 * 1024):
 
     lib = ops._load()
 
+    fn = lib.sy_lz4_decode_blocks_pc if pc else lib.sy_lz4_decode_blocks
+
     def run():
-        lib.sy_lz4_decode_blocks(
+        fn(
             ctypes.c_void_p(d_comp.data_ptr()),
             ctypes.c_void_p(d_in_off.data_ptr()),
             ctypes.c_void_p(d_in_len.data_ptr()),

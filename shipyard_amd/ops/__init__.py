@@ -64,6 +64,8 @@ def _load() -> ctypes.CDLL:
         ctypes.c_void_p, ctypes.c_uint64, ctypes.c_uint32, ctypes.c_void_p,
         ctypes.c_uint64, ctypes.c_void_p,
     ]
+    lib.sy_lz4_decode_blocks_pc.restype = ctypes.c_int
+    lib.sy_lz4_decode_blocks_pc.argtypes = lib.sy_lz4_decode_blocks.argtypes
     lib.sy_stage_file.restype = ctypes.c_int
     lib.sy_stage_file.argtypes = [
         ctypes.c_char_p, ctypes.c_void_p, ctypes.c_uint64, ctypes.c_uint64,
@@ -142,7 +144,7 @@ def crc32c_file_digest(data, chunk_size: int = 256 * 1024) -> int:
 
 
 def lz4_decode_blocks(comp, in_off, in_len, out, out_off, out_len,
-                      raw_cap: int = 64 * 1024):
+                      raw_cap: int = 64 * 1024, pc: bool = None):
     """Decode independent LZ4 blocks on the GPU.
 
     All tensors are CUDA: ``comp``/``out`` uint8, ``in_off``/``out_off``
@@ -152,12 +154,17 @@ def lz4_decode_blocks(comp, in_off, in_len, out, out_off, out_len,
     hiding; the decoder is serial per block).  Returns a CUDA uint32
     status tensor (0 == OK per block).
     """
+    import os as _os
+
     import torch
 
     lib = _load()
+    if pc is None:
+        pc = _os.environ.get("SHIPYARD_LZ4_PC", "0") == "1"
+    fn = lib.sy_lz4_decode_blocks_pc if pc else lib.sy_lz4_decode_blocks
     n_blocks = in_off.numel()
     status = torch.empty(n_blocks, dtype=torch.uint32, device=comp.device)
-    rc = lib.sy_lz4_decode_blocks(
+    rc = fn(
         ctypes.c_void_p(comp.data_ptr()), ctypes.c_void_p(in_off.data_ptr()),
         ctypes.c_void_p(in_len.data_ptr()), ctypes.c_void_p(out.data_ptr()),
         ctypes.c_void_p(out_off.data_ptr()), ctypes.c_void_p(out_len.data_ptr()),

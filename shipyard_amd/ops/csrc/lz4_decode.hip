@@ -228,3 +228,253 @@ SY_EXPORT int sy_lz4_decode_blocks(const void* d_comp, const uint64_t* d_in_off,
   }
   return sy_check(hipGetLastError());
 }
+
+namespace {
+
+// ---------------------------------------------------------------------
+// Producer/consumer variant: one 128-thread workgroup (2 waves) per
+// block.  Wave 0 parses the sequence stream into an LDS record ring;
+// wave 1 executes literal/match copies.  The two serial dependency
+// chains (parse: dependent LDS byte reads; copy: LDS round trips +
+// lgkm waits) overlap instead of interleaving, targeting ~2x per
+// block at unchanged LDS footprint (occupancy is LDS-bound).
+// All spins are bounded: a protocol bug yields SY_LZ4_ERR_DEADLOCK,
+// never a wedged GPU.
+// ---------------------------------------------------------------------
+
+constexpr uint32_t SY_LZ4_ERR_DEADLOCK = 6;
+constexpr int kRingSz = 64;  // records; power of two
+constexpr uint32_t kSpinLimit = 1u << 27;
+
+struct SeqRec {
+  uint32_t lit_src;   // comp offset of literal bytes (rel. to src)
+  uint32_t lit_len;
+  uint32_t offset;    // 0 == literals-only terminator
+  uint32_t mlen;
+};
+
+template <int RAWCAP>
+__global__ __launch_bounds__(2 * SY_WAVE) void lz4_decode_pc_kernel(
+    const uint8_t* __restrict__ comp, const uint64_t* __restrict__ in_off,
+    const uint32_t* __restrict__ in_len, uint8_t* __restrict__ out,
+    const uint64_t* __restrict__ out_off, const uint32_t* __restrict__ out_len,
+    uint32_t* __restrict__ status, uint32_t n_blocks) {
+  constexpr int kSrcBuf = RAWCAP + 1024;
+  __shared__ uint8_t dst[RAWCAP];
+  __shared__ uint8_t sbuf[kSrcBuf];
+  __shared__ SeqRec ring[kRingSz];
+  __shared__ uint32_t ctl[4];  // [0]=produced [1]=consumed [2]=status [3]=nseq
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  volatile uint32_t* vctl = ctl;
+
+  for (uint32_t blk = blockIdx.x; blk < n_blocks; blk += gridDim.x) {
+    const uint64_t abase = in_off[blk];
+    const uint32_t slen = in_len[blk];
+    const uint32_t rawlen = out_len[blk];
+
+    if (slen > (uint32_t)kSrcBuf - 48 || rawlen > (uint32_t)RAWCAP) {
+      if (tid == 0) status[blk] = SY_LZ4_ERR_TOOBIG;
+      __syncthreads();
+      continue;
+    }
+
+    // ---- stage compressed block into LDS (both waves, 16 B/lane) ----
+    const uint64_t astart = abase & ~15ull;
+    const uint32_t srcoff = (uint32_t)(abase - astart);
+    const uint32_t stage_bytes = srcoff + slen;
+    {
+      const uint4* g4 = reinterpret_cast<const uint4*>(comp + astart);
+      uint4* s4 = reinterpret_cast<uint4*>(sbuf);
+      const uint32_t n16 = (stage_bytes + 15) >> 4;
+      for (uint32_t i = tid; i < n16; i += 2 * SY_WAVE) s4[i] = g4[i];
+    }
+    if (tid == 0) {
+      ctl[0] = 0; ctl[1] = 0; ctl[2] = SY_LZ4_OK; ctl[3] = 0;
+    }
+    __syncthreads();
+    const uint8_t* src = sbuf + srcoff;
+
+    if (wave == 0) {
+      // ---------------- producer: parse into the ring ----------------
+      uint32_t pos = 0;
+      uint32_t produced = 0;
+      uint32_t st = SY_LZ4_OK;
+      uint32_t dtotal = 0;  // decoded bytes accounted (bounds checks)
+      for (;;) {
+        if (pos >= slen) {
+          if (dtotal != rawlen) st = SY_LZ4_ERR_MISMATCH;
+          break;
+        }
+        const uint32_t token = src[pos++];
+        uint32_t litlen = token >> 4;
+        if (litlen == 15) {
+          uint8_t b;
+          do {
+            if (pos >= slen) { st = SY_LZ4_ERR_TRUNC; break; }
+            b = src[pos++];
+            litlen += b;
+          } while (b == 255);
+          if (st != SY_LZ4_OK) break;
+        }
+        if (dtotal + litlen > rawlen || pos + litlen > slen) {
+          st = SY_LZ4_ERR_OVERFLOW;
+          break;
+        }
+        const uint32_t lit_src = pos;
+        pos += litlen;
+        dtotal += litlen;
+        uint32_t offset = 0, mlen = 0;
+        if (pos != slen) {
+          if (pos + 2 > slen) { st = SY_LZ4_ERR_TRUNC; break; }
+          offset = (uint32_t)src[pos] | ((uint32_t)src[pos + 1] << 8);
+          pos += 2;
+          mlen = (token & 0xFu) + 4;
+          if ((token & 0xFu) == 15) {
+            uint8_t b;
+            do {
+              if (pos >= slen) { st = SY_LZ4_ERR_TRUNC; break; }
+              b = src[pos++];
+              mlen += b;
+            } while (b == 255);
+            if (st != SY_LZ4_OK) break;
+          }
+          if (offset == 0 || offset > dtotal) {
+            st = SY_LZ4_ERR_OFFSET;
+            break;
+          }
+          if (dtotal + mlen > rawlen) { st = SY_LZ4_ERR_OVERFLOW; break; }
+          dtotal += mlen;
+        }
+        // flow control: wait for ring space (lane 0 polls, broadcast)
+        uint32_t spins = 0;
+        uint32_t consumed;
+        do {
+          consumed = vctl[1];
+          if (++spins > kSpinLimit) { st = SY_LZ4_ERR_DEADLOCK; break; }
+        } while (produced - consumed >= (uint32_t)kRingSz);
+        if (st != SY_LZ4_OK) break;
+        if (lane == 0) {
+          SeqRec r;
+          r.lit_src = lit_src;
+          r.lit_len = litlen;
+          r.offset = offset;
+          r.mlen = mlen;
+          ring[produced & (kRingSz - 1)] = r;
+          __builtin_amdgcn_s_waitcnt(kWaitLgkm0);  // record before counter
+          ctl[0] = produced + 1;
+          __builtin_amdgcn_s_waitcnt(kWaitLgkm0);
+        }
+        ++produced;
+        if (pos == slen && (dtotal == rawlen)) {
+          // final literals-only sequence had offset==0 terminator
+          if (offset == 0) break;
+        }
+        if (pos == slen) {
+          if (dtotal != rawlen) st = SY_LZ4_ERR_MISMATCH;
+          break;
+        }
+      }
+      if (lane == 0) {
+        if (st != SY_LZ4_OK) vctl[2] = st;
+        vctl[3] = produced | 0x80000000u;  // parse done + count
+        __builtin_amdgcn_s_waitcnt(kWaitLgkm0);
+      }
+    } else {
+      // ---------------- consumer: execute copies ---------------------
+      uint32_t consumed = 0;
+      uint32_t dpos = 0;
+      uint32_t st = SY_LZ4_OK;
+      for (;;) {
+        // wait for a record or parse-done
+        uint32_t spins = 0;
+        uint32_t produced, done_word;
+        for (;;) {
+          produced = vctl[0];
+          done_word = vctl[3];
+          if (produced > consumed) break;
+          if (done_word & 0x80000000u) break;
+          if (vctl[2] != SY_LZ4_OK) break;
+          if (++spins > kSpinLimit) break;
+        }
+        if (produced <= consumed) {
+          if (spins > kSpinLimit) st = SY_LZ4_ERR_DEADLOCK;
+          break;  // parse done (or error): no more records
+        }
+        SeqRec r = ring[consumed & (kRingSz - 1)];
+        // literal copy: sbuf -> dst
+        for (uint32_t i = lane; i < r.lit_len; i += SY_WAVE) {
+          dst[dpos + i] = src[r.lit_src + i];
+        }
+        dpos += r.lit_len;
+        if (r.offset) {
+          __builtin_amdgcn_s_waitcnt(kWaitLgkm0);
+          uint32_t done = 0;
+          while (done < r.mlen) {
+            const uint32_t dist = done + r.offset;
+            const uint32_t n = min(r.mlen - done, dist);
+            for (uint32_t i = lane; i < n; i += SY_WAVE) {
+              dst[dpos + done + i] = dst[dpos + done + i - dist];
+            }
+            done += n;
+            __builtin_amdgcn_s_waitcnt(kWaitLgkm0);
+          }
+          dpos += r.mlen;
+        }
+        ++consumed;
+        if (lane == 0) {
+          vctl[1] = consumed;
+        }
+      }
+      if (lane == 0 && st != SY_LZ4_OK && vctl[2] == SY_LZ4_OK) {
+        vctl[2] = st;
+      }
+    }
+    __syncthreads();
+
+    const uint32_t st = ctl[2];
+    if (tid == 0) status[blk] = st;
+    if (st == SY_LZ4_OK) {
+      // ---- stream LDS -> HBM (both waves) ----
+      uint8_t* g = out + out_off[blk];
+      const uint32_t n16 = rawlen >> 4;
+      const uint4* s4 = reinterpret_cast<const uint4*>(dst);
+      uint4* g4 = reinterpret_cast<uint4*>(g);
+      for (uint32_t i = tid; i < n16; i += 2 * SY_WAVE) g4[i] = s4[i];
+      for (uint32_t i = (n16 << 4) + tid; i < rawlen; i += 2 * SY_WAVE)
+        g[i] = dst[i];
+    }
+    __syncthreads();
+  }
+}
+
+}  // namespace
+
+SY_EXPORT int sy_lz4_decode_blocks_pc(
+    const void* d_comp, const uint64_t* d_in_off, const uint32_t* d_in_len,
+    void* d_out, const uint64_t* d_out_off, const uint32_t* d_out_len,
+    uint32_t* d_status, uint32_t n_blocks, uint32_t raw_cap,
+    hipStream_t stream) {
+  if (n_blocks == 0) return 0;
+  uint32_t grid = n_blocks < 4096u ? n_blocks : 4096u;
+  const uint8_t* c = static_cast<const uint8_t*>(d_comp);
+  uint8_t* o = static_cast<uint8_t*>(d_out);
+  if (raw_cap <= 8 * 1024) {
+    hipLaunchKernelGGL((lz4_decode_pc_kernel<8 * 1024>), dim3(grid),
+                       dim3(2 * SY_WAVE), 0, stream, c, d_in_off, d_in_len,
+                       o, d_out_off, d_out_len, d_status, n_blocks);
+  } else if (raw_cap <= 16 * 1024) {
+    hipLaunchKernelGGL((lz4_decode_pc_kernel<16 * 1024>), dim3(grid),
+                       dim3(2 * SY_WAVE), 0, stream, c, d_in_off, d_in_len,
+                       o, d_out_off, d_out_len, d_status, n_blocks);
+  } else if (raw_cap <= 64 * 1024) {
+    hipLaunchKernelGGL((lz4_decode_pc_kernel<64 * 1024>), dim3(grid),
+                       dim3(2 * SY_WAVE), 0, stream, c, d_in_off, d_in_len,
+                       o, d_out_off, d_out_len, d_status, n_blocks);
+  } else {
+    return -22;
+  }
+  return sy_check(hipGetLastError());
+}
