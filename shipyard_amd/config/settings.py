@@ -502,3 +502,78 @@ def storage_account_settings(creds: Dict[str, Any],
         raise KeyError(f"storage account settings not found: {name}")
     return StorageAccountSettings(name=name, root=sa["root"],
                                   create=sa.get("create", True))
+
+
+# --------------------------------------------------------------------
+# monitoring / federation / slurm (reference settings.py:4907-5550)
+# --------------------------------------------------------------------
+@dataclasses.dataclass(frozen=True)
+class MonitoringSettings:
+    exporter_port: int
+    exporter_interval_s: float
+    collectors: Tuple[str, ...]
+    file_sd_directory: Optional[str]
+    scrape_interval: str
+    grafana_dashboard_output: Optional[str]
+
+
+def monitoring_settings(conf: Dict[str, Any]) -> MonitoringSettings:
+    m = conf.get("monitoring") or {}
+    return MonitoringSettings(
+        exporter_port=_get(m, "exporter", "port", default=9400),
+        exporter_interval_s=_get(m, "exporter", "interval_seconds",
+                                 default=1.0),
+        collectors=tuple(_get(m, "exporter", "collectors",
+                              default=["gpu", "executor"])),
+        file_sd_directory=_get(m, "prometheus", "file_sd_directory"),
+        scrape_interval=_get(m, "prometheus", "scrape_interval",
+                             default="5s"),
+        grafana_dashboard_output=_get(m, "grafana", "dashboard_output"),
+    )
+
+
+@dataclasses.dataclass(frozen=True)
+class FederationSettings:
+    federations: Dict[str, dict]
+    poll_federations_s: float
+    poll_actions_s: float
+
+
+def federation_settings(conf: Dict[str, Any]) -> FederationSettings:
+    f = conf.get("federation") or {}
+    return FederationSettings(
+        federations=dict(f.get("federations") or {}),
+        poll_federations_s=_get(f, "proxy_options", "polling_interval",
+                                "federations", default=5.0),
+        poll_actions_s=_get(f, "proxy_options", "polling_interval",
+                            "actions", default=1.0),
+    )
+
+
+@dataclasses.dataclass(frozen=True)
+class SlurmPartitionSettings:
+    name: str
+    batch_pools: Dict[str, dict]
+    default: bool
+    max_runtime_limit: Optional[datetime.timedelta]
+
+
+@dataclasses.dataclass(frozen=True)
+class SlurmSettings:
+    cluster_id: str
+    partitions: Tuple[SlurmPartitionSettings, ...]
+
+
+def slurm_settings(conf: Dict[str, Any]) -> SlurmSettings:
+    s = conf.get("slurm") or {}
+    parts = []
+    for name, p in (s.get("elastic_partitions") or {}).items():
+        parts.append(SlurmPartitionSettings(
+            name=name,
+            batch_pools=dict(p.get("batch_pools") or {}),
+            default=p.get("default", False),
+            max_runtime_limit=utils.parse_timedelta(
+                p.get("max_runtime_limit")),
+        ))
+    return SlurmSettings(cluster_id=s.get("cluster_id", "shipyard"),
+                         partitions=tuple(parts))

@@ -119,3 +119,17 @@ def test_timedelta_and_size_types():
     doc["job_specifications"][0]["shm_size"] = "xx"
     with pytest.raises(SchemaViolation):
         validate_config(ConfigType.jobs, doc)
+
+
+def test_typed_accessors_for_aux_families():
+    from shipyard_amd.config.settings import (federation_settings,
+                                              monitoring_settings,
+                                              slurm_settings)
+
+    m = monitoring_settings(yaml.safe_load(VALID[ConfigType.monitor]))
+    assert m.exporter_port == 9400 and "gpu" in m.collectors
+    f = federation_settings(yaml.safe_load(VALID[ConfigType.federation]))
+    assert f.federations["f1"]["pools"] == ["a", "b"]
+    s = slurm_settings(yaml.safe_load(VALID[ConfigType.slurm]))
+    assert s.cluster_id == "sy"
+    assert s.partitions[0].batch_pools["p"]["max_compute_nodes"] == 4
