@@ -363,12 +363,9 @@ __global__ __launch_bounds__(2 * SY_WAVE) void lz4_decode_pc_kernel(
           r.offset = offset;
           r.mlen = mlen;
           ring[produced & (kRingSz - 1)] = r;
-          // batched publish: a counter round trip per record measured
-          // slower than no overlap at all; publish every 8
-          if (((produced + 1) & 7u) == 0) {
-            __builtin_amdgcn_s_waitcnt(kWaitLgkm0);  // records first
-            ctl[0] = produced + 1;
-          }
+          __builtin_amdgcn_s_waitcnt(kWaitLgkm0);  // record before counter
+          ctl[0] = produced + 1;
+          __builtin_amdgcn_s_waitcnt(kWaitLgkm0);
         }
         ++produced;
         if (pos == slen && (dtotal == rawlen)) {
@@ -382,9 +379,7 @@ __global__ __launch_bounds__(2 * SY_WAVE) void lz4_decode_pc_kernel(
       }
       if (lane == 0) {
         if (st != SY_LZ4_OK) vctl[2] = st;
-        __builtin_amdgcn_s_waitcnt(kWaitLgkm0);  // drain ring writes
-        ctl[0] = produced;                       // final publish
-        vctl[3] = produced | 0x80000000u;        // parse done + count
+        vctl[3] = produced | 0x80000000u;  // parse done + count
         __builtin_amdgcn_s_waitcnt(kWaitLgkm0);
       }
     } else {
@@ -429,7 +424,7 @@ __global__ __launch_bounds__(2 * SY_WAVE) void lz4_decode_pc_kernel(
           dpos += r.mlen;
         }
         ++consumed;
-        if (lane == 0 && (consumed & 7u) == 0) {
+        if (lane == 0) {
           vctl[1] = consumed;
         }
       }
