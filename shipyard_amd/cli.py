@@ -220,11 +220,19 @@ def jobs():
 @click.option("--poolid")
 @click.option("--tail", help="stream this file of the last task")
 @click.option("--wait", is_flag=True, help="run scheduler until idle")
+@click.option("--recreate", is_flag=True,
+              help="delete an existing job of the same id first "
+                   "(reference `jobs add --recreate`)")
 @_common
 @pass_ctx
-def jobs_add(ctx, poolid, tail, wait, configdir, root, raw):
+def jobs_add(ctx, poolid, tail, wait, recreate, configdir, root, raw):
     _apply(ctx, configdir, root, raw)
     pid = poolid or ctx.conf(ConfigType.pool)["pool_specification"]["id"]
+    if recreate:
+        for spec in ctx.conf(ConfigType.jobs)["job_specifications"]:
+            if ctx.executor.store.query_one(
+                    "SELECT id FROM jobs WHERE id=?", (spec["id"],)):
+                ctx.executor.job_del(spec["id"])
     added = ctx.executor.jobs_add(ctx.conf(ConfigType.jobs), pid)
     if wait or tail:
         ctx.executor.run_until_idle()
@@ -416,6 +424,32 @@ def diag_events(ctx, prefix, configdir, root, raw):
     from shipyard_amd.cascade import perf
 
     click.echo(perf.dump(ctx.executor.store, prefix))
+
+
+@diag.command("du")
+@_common
+@pass_ctx
+def diag_du(ctx, configdir, root, raw):
+    """Disk usage of the executor root by area."""
+    _apply(ctx, configdir, root, raw)
+    import os as _os
+
+    out = {}
+    base = Path(ctx.executor.root)
+    for sub in ("objects", "pools", "store.db"):
+        p = base / sub
+        if p.is_file():
+            out[sub] = p.stat().st_size
+        elif p.is_dir():
+            total = 0
+            for dirpath, _dirs, files in _os.walk(p):
+                for f in files:
+                    try:
+                        total += _os.path.getsize(_os.path.join(dirpath, f))
+                    except OSError:
+                        pass
+            out[sub] = total
+    ctx.emit(out)
 
 
 @diag.command("timeline")

@@ -121,3 +121,19 @@ def test_storage_and_ingress(configdir, tmp_path):
     assert r.exit_code == 0, r.output
     r = run(["storage", "list", "--prefix", "staged"], configdir, tmp_path)
     assert "x.dat" in r.output
+
+
+def test_jobs_add_recreate(configdir, tmp_path):
+    run(["pool", "add"], configdir, tmp_path)
+    r = run(["jobs", "add", "--wait"], configdir, tmp_path)
+    assert r.exit_code == 0
+    # same job id again without --recreate fails...
+    r = run(["jobs", "add"], configdir, tmp_path, catch=True)
+    assert r.exit_code != 0
+    # ...and succeeds with --recreate
+    r = run(["jobs", "add", "--recreate", "--wait"], configdir, tmp_path)
+    assert r.exit_code == 0, r.output
+
+    r = run(["diag", "du"], configdir, tmp_path)
+    assert r.exit_code == 0
+    assert "pools" in r.output

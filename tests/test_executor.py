@@ -284,3 +284,17 @@ def test_gpu_task_rejected_on_cpu_pool(ex):
         ex.schedule_once()
     t = ex.tasks_list("jgpu")[0]
     assert t["state"] == "ready"
+
+
+def test_docker_task_without_docker_fails_cleanly(ex):
+    import shutil
+
+    if shutil.which("docker"):
+        pytest.skip("docker present on this host")
+    make_pool(ex)
+    ex.jobs_add(job("jdock", [{"id": "d", "command": "echo hi",
+                               "docker_image": "busybox",
+                               "max_task_retries": 0}]), "p1")
+    ex.run_until_idle(timeout=30)
+    t = ex.tasks_list("jdock")[0]
+    assert t["state"] == "failed" and t["exit_code"] == -1
