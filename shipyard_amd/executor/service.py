@@ -415,13 +415,25 @@ class LocalExecutor:
             "state='disabled'", (job_id,))
 
     def job_terminate(self, job_id: str, wait: bool = True) -> None:
+        # kill + POP handles (leaving them would make a later collect
+        # pass look up rows this method or job_del may have removed)
         for (jid, tid), h in list(self._handles.items()):
             if jid == job_id:
                 h.kill()
+                del self._handles[(jid, tid)]
+                row = self.store.query_one(
+                    "SELECT slots_json FROM tasks WHERE job_id=? AND id=?",
+                    (jid, tid))
+                if row:
+                    self._release_slots(
+                        self._job_pool(jid),
+                        json.loads(row["slots_json"] or "[]"))
         with self.store.transaction() as conn:
             conn.execute(
-                "UPDATE tasks SET state='cancelled' WHERE job_id=? AND "
-                "state IN ('pending','ready','blocked')", (job_id,))
+                "UPDATE tasks SET state='cancelled', end_time=? WHERE "
+                "job_id=? AND state IN "
+                "('pending','ready','blocked','running')",
+                (time.time(), job_id))
             conn.execute(
                 "UPDATE jobs SET state='terminated', completed_at=? "
                 "WHERE id=?", (time.time(), job_id))

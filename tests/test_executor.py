@@ -298,3 +298,23 @@ def test_docker_task_without_docker_fails_cleanly(ex):
     ex.run_until_idle(timeout=30)
     t = ex.tasks_list("jdock")[0]
     assert t["state"] == "failed" and t["exit_code"] == -1
+
+
+def test_terminate_then_schedule_does_not_crash(ex):
+    """Regression: killed handles must be popped at terminate, else a
+    later collect pass dereferences deleted task rows."""
+    make_pool(ex)
+    ex.jobs_add(job("jtk", [{"id": "s", "command": "sleep 30"}]), "p1")
+    for _ in range(200):
+        ex.schedule_once()
+        if ex.tasks_list("jtk")[0]["state"] == "running":
+            break
+        time.sleep(0.02)
+    ex.job_terminate("jtk")
+    ex.job_del("jtk")
+    for _ in range(5):
+        ex.schedule_once()  # must not raise
+    # the slot must be reusable
+    ex.jobs_add(job("jtk2", [{"id": "t", "command": "true"}]), "p1")
+    ex.run_until_idle(timeout=30)
+    assert ex.tasks_list("jtk2")[0]["state"] == "completed"
