@@ -644,6 +644,8 @@ class LocalExecutor:
                      h: TaskHandle) -> None:
         row = self.store.query_one(
             "SELECT * FROM tasks WHERE job_id=? AND id=?", (jid, tid))
+        if row is None:  # job deleted between poll and finish
+            return
         spec = json.loads(row["spec_json"])
         slots = json.loads(row["slots_json"] or "[]")
         with self.store.transaction() as conn:
