@@ -354,6 +354,8 @@ __global__ __launch_bounds__(2 * SY_WAVE) void lz4_decode_pc_kernel(
         do {
           consumed = vctl[1];
           if (++spins > kSpinLimit) { st = SY_LZ4_ERR_DEADLOCK; break; }
+          if (produced - consumed >= (uint32_t)kRingSz)
+            __builtin_amdgcn_s_sleep(2);  // back off the LDS pipe
         } while (produced - consumed >= (uint32_t)kRingSz);
         if (st != SY_LZ4_OK) break;
         if (lane == 0) {
@@ -398,6 +400,7 @@ __global__ __launch_bounds__(2 * SY_WAVE) void lz4_decode_pc_kernel(
           if (done_word & 0x80000000u) break;
           if (vctl[2] != SY_LZ4_OK) break;
           if (++spins > kSpinLimit) break;
+          __builtin_amdgcn_s_sleep(1);  // back off the LDS pipe
         }
         if (produced <= consumed) {
           if (spins > kSpinLimit) st = SY_LZ4_ERR_DEADLOCK;
