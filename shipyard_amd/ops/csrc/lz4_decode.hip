@@ -153,7 +153,12 @@ __global__ __launch_bounds__(SY_WAVE) void lz4_decode_kernel(
         if (offset == 0 || offset > dpos) { st = SY_LZ4_ERR_OFFSET; break; }
         if (dpos + mlen > rawlen) { st = SY_LZ4_ERR_OVERFLOW; break; }
 
-        // literal writes must be visible before match reads
+        // all prior LDS writes (this sequence's literals, earlier
+        // matches) must be visible before match reads; between rounds
+        // the same wait orders round i -> i+1.  The LAST round needs
+        // no wait: the next sequence's literal copy only WRITES (to
+        // disjoint, higher dst addresses), and its own pre-match wait
+        // re-orders before any read.
         __builtin_amdgcn_s_waitcnt(kWaitLgkm0);
 
         // ---- cooperative match copy, doubling over overlap ----
@@ -165,7 +170,7 @@ __global__ __launch_bounds__(SY_WAVE) void lz4_decode_kernel(
             dst[dpos + done + i] = dst[dpos + done + i - dist];
           }
           done += n;
-          __builtin_amdgcn_s_waitcnt(kWaitLgkm0);
+          if (done < mlen) __builtin_amdgcn_s_waitcnt(kWaitLgkm0);
         }
         dpos += mlen;
       }
