@@ -185,6 +185,25 @@ def pool_stats(ctx, poolid, configdir, root, raw):
     ctx.emit(ctx.executor.pool_stats(pid))
 
 
+@pool.command("autoscale-evaluate")
+@click.option("--poolid")
+@_common
+@pass_ctx
+def pool_autoscale_evaluate(ctx, poolid, configdir, root, raw):
+    """One-shot autoscale evaluation (reference `pool autoscale
+    lastexec`/formula evaluation)."""
+    _apply(ctx, configdir, root, raw)
+    import time as _time
+
+    from shipyard_amd.executor.autoscale import AutoscaleController
+
+    pid = poolid or ctx.conf(ConfigType.pool)["pool_specification"]["id"]
+    ctl = AutoscaleController(ctx.executor, pid,
+                              ctx.executor._pool_settings(pid).autoscale)
+    dec = ctl.maybe_evaluate(_time.time() + 10**9)
+    ctx.emit(dec.__dict__ if dec else {"autoscale": "disabled"})
+
+
 @pool.group("images")
 def pool_images():
     """Global resource (image) ops on a pool."""
@@ -306,6 +325,17 @@ def jobs_stats(ctx, jobid, configdir, root, raw):
 @jobs.group("tasks")
 def jobs_tasks():
     """Task-level operations."""
+
+
+@jobs_tasks.command("term")
+@click.option("--jobid", required=True)
+@click.option("--taskid", required=True)
+@_common
+@pass_ctx
+def tasks_term(ctx, jobid, taskid, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ctx.executor.task_terminate(jobid, taskid)
+    ctx.emit({"terminated": f"{jobid}/{taskid}"})
 
 
 @jobs_tasks.command("list")

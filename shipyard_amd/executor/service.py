@@ -438,6 +438,25 @@ class LocalExecutor:
                 "UPDATE jobs SET state='terminated', completed_at=? "
                 "WHERE id=?", (time.time(), job_id))
 
+    def task_terminate(self, job_id: str, task_id: str) -> None:
+        """Terminate one task (reference `jobs tasks term`): kill if
+        running, cancel if queued; no retry."""
+        h = self._handles.pop((job_id, task_id), None)
+        if h:
+            h.kill()
+            row = self.store.query_one(
+                "SELECT slots_json FROM tasks WHERE job_id=? AND id=?",
+                (job_id, task_id))
+            if row:
+                self._release_slots(self._job_pool(job_id),
+                                    json.loads(row["slots_json"] or "[]"))
+        self.store.execute(
+            "UPDATE tasks SET state='cancelled', end_time=? WHERE "
+            "job_id=? AND id=? AND state IN "
+            "('pending','ready','blocked','running')",
+            (time.time(), job_id, task_id))
+        self.store.add_event(f"task:{job_id}/{task_id}", "terminated")
+
     def job_disable_requeue(self, job_id: str) -> None:
         """`jobs disable --requeue` (reference convoy/batch.py:2102):
         kill running tasks and return them to ready, then disable."""

@@ -142,3 +142,22 @@ def test_fault_injection_kills_gang_rank(tmp_path):
     assert t["state"] == "failed"
     assert t["exit_code"] == -9  # SIGKILL of the injected rank
     ex.store.close()
+
+
+def test_task_terminate_single(tmp_path):
+    ex = LocalExecutor(tmp_path / "tt", detect_gpus=False)
+    mkpool(ex, "tp")
+    ex.jobs_add({"job_specifications": [{
+        "id": "tj",
+        "tasks": [{"id": "slow", "command": "sleep 30"},
+                  {"id": "fast", "command": "true"}]}]}, "tp")
+    for _ in range(200):
+        ex.schedule_once()
+        if ex.tasks_list("tj")[0]["state"] == "running":
+            break
+        time.sleep(0.02)
+    ex.task_terminate("tj", "slow")
+    ex.run_until_idle(timeout=30)
+    states = {t["id"]: t["state"] for t in ex.tasks_list("tj")}
+    assert states["slow"] == "cancelled" and states["fast"] == "completed"
+    ex.store.close()
