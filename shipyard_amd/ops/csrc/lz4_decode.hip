@@ -32,6 +32,7 @@
 // preserved and every read lands in a completed round.
 
 #include "common.h"
+#include <cstdlib>
 
 namespace {
 
@@ -201,11 +202,15 @@ SY_EXPORT int sy_lz4_decode_blocks(const void* d_comp, const uint64_t* d_in_off,
                                    uint32_t* d_status, uint32_t n_blocks,
                                    uint32_t raw_cap, hipStream_t stream) {
   if (n_blocks == 0) return 0;
-    // grid cap 4096: at the high-occupancy small-RAWCAP geometries a
-  // larger grid measured 20-35% slower on real content (more resident
-  // strides fighting over L2 block locality); 4096 was the best of
-  // 4096/8192 at 4-8 KiB blocks.
-  uint32_t grid = n_blocks < 4096u ? n_blocks : 4096u;
+    // Grid cap (SY_LZ4_GRID overrides; cold-clock noise tainted early
+  // 4096-vs-8192 readings — re-sweep with DVFS-warmed methodology).
+  static uint32_t grid_cap = 0;
+  if (grid_cap == 0) {
+    const char* e = getenv("SY_LZ4_GRID");
+    grid_cap = e ? (uint32_t)atoi(e) : 4096;
+    if (grid_cap == 0) grid_cap = 4096;
+  }
+  uint32_t grid = n_blocks < grid_cap ? n_blocks : grid_cap;
   const uint8_t* c = static_cast<const uint8_t*>(d_comp);
   uint8_t* o = static_cast<uint8_t*>(d_out);
   if (raw_cap <= 4 * 1024) {
