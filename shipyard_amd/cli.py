@@ -322,6 +322,28 @@ def jobs_stats(ctx, jobid, configdir, root, raw):
     ctx.emit(ctx.executor.job_stats(jobid))
 
 
+@jobs.group("schedules")
+def jobs_schedules():
+    """Job schedules (recurrences)."""
+
+
+@jobs_schedules.command("list")
+@_common
+@pass_ctx
+def schedules_list(ctx, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ctx.emit(ctx.executor.schedules_list())
+
+
+@jobs_schedules.command("del")
+@click.option("--scheduleid", required=True)
+@_common
+@pass_ctx
+def schedules_del(ctx, scheduleid, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ctx.emit({"deleted": ctx.executor.schedule_del(scheduleid)})
+
+
 @jobs.group("tasks")
 def jobs_tasks():
     """Task-level operations."""
@@ -823,6 +845,7 @@ def daemon(ctx, idle_exit, interval, configdir, root, raw):
     logger.info("daemon running (idle_exit=%s)", idle_exit)
     while True:
         ex.schedule_once()
+        ex.process_schedules()
         now = time.time()
         for p in ex.pool_list():
             pid = p["id"]

@@ -312,6 +312,15 @@ class LocalExecutor:
         added = []
         for jobspec in jobs_conf["job_specifications"]:
             js = cfg.job_settings(jobspec)
+            if js.recurrence is not None:
+                # job schedule (reference convoy/batch.py:5390 JobSchedule):
+                # register; instances materialize from process_schedules()
+                self.store.kv_set(f"schedule:{js.id}", json.dumps(
+                    {"pool": pool_id, "jobspec": jobspec}))
+                self.store.add_event(f"jobschedule:{js.id}", "registered",
+                                     {"pool": pool_id})
+                added.append(js.id)
+                continue
             if self.store.query_one("SELECT id FROM jobs WHERE id=?",
                                     (js.id,)):
                 raise ExecutorError(f"job {js.id} exists")
@@ -577,6 +586,50 @@ class LocalExecutor:
         self._complete_auto_jobs()
         return n
 
+    # -- job schedules (recurrences) ---------------------------------
+    def process_schedules(self, now: Optional[float] = None) -> List[str]:
+        """Spawn due recurrence instances (the cargo job-manager
+        analogue).  Called by the daemon/scheduler thread, not by
+        run_until_idle (a live schedule never goes idle)."""
+        from shipyard_amd.executor.recurrence import JobScheduleRunner
+
+        if not hasattr(self, "_schedule_runners"):
+            self._schedule_runners = {}
+        spawned = []
+        rows = self.store.query(
+            "SELECT key, value FROM kv WHERE key LIKE 'schedule:%'")
+        live = set()
+        for r in rows:
+            sid = r["key"][len("schedule:"):]
+            live.add(sid)
+            if sid not in self._schedule_runners:
+                rec = json.loads(r["value"])
+                self._schedule_runners[sid] = JobScheduleRunner(
+                    self, rec["pool"], rec["jobspec"])
+            inst = self._schedule_runners[sid].maybe_spawn(now)
+            if inst:
+                spawned.append(inst)
+        for sid in list(self._schedule_runners):
+            if sid not in live:
+                del self._schedule_runners[sid]
+        return spawned
+
+    def schedule_del(self, schedule_id: str) -> bool:
+        n = self.store.execute("DELETE FROM kv WHERE key=?",
+                               (f"schedule:{schedule_id}",)).rowcount
+        if hasattr(self, "_schedule_runners"):
+            self._schedule_runners.pop(schedule_id, None)
+        return bool(n)
+
+    def schedules_list(self) -> List[dict]:
+        out = []
+        for r in self.store.query(
+                "SELECT key, value FROM kv WHERE key LIKE 'schedule:%'"):
+            rec = json.loads(r["value"])
+            out.append({"id": r["key"][len("schedule:"):],
+                        "pool": rec["pool"]})
+        return out
+
     def start_scheduler(self, poll: float = 0.02,
                         autoscale: bool = True) -> None:
         """Run the scheduling loop in a background thread (the library
@@ -595,6 +648,7 @@ class LocalExecutor:
             while not self._sched_stop.is_set():
                 try:
                     self.schedule_once()
+                    self.process_schedules()
                     if autoscale:
                         now = time.time()
                         for p in self.pool_list():

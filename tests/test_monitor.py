@@ -74,3 +74,26 @@ def test_recurrence_spawns_instances(tmp_path):
     states = [t["state"] for t in ex.tasks_list("rec-000")]
     assert states == ["completed"]
     ex.store.close()
+
+
+def test_recurrence_through_jobs_add(tmp_path):
+    """jobs_add registers a schedule (no immediate job); the scheduler
+    loop materializes instances (reference JobSchedule semantics)."""
+    ex = LocalExecutor(tmp_path / "rs", detect_gpus=False)
+    ex.pool_add({"pool_specification": {
+        "id": "prs", "gpus": {"dedicated": 0}, "cpu_slots": 1,
+        "node_configuration": {"rocm": {"verify": False}}}})
+    added = ex.jobs_add({"job_specifications": [{
+        "id": "schd", "tasks": [{"id": "t", "command": "true"}],
+        "recurrence": {"schedule": {"recurrence_interval": "00:00:01"}},
+    }]}, "prs")
+    assert added == ["schd"]
+    assert ex.jobs_list() == []  # no immediate job
+    assert ex.schedules_list()[0]["id"] == "schd"
+    spawned = ex.process_schedules()
+    assert spawned == ["schd-000"]
+    ex.run_until_idle(timeout=30)
+    assert ex.tasks_list("schd-000")[0]["state"] == "completed"
+    assert ex.schedule_del("schd")
+    assert ex.schedules_list() == []
+    ex.store.close()
