@@ -205,13 +205,14 @@ SY_EXPORT int sy_lz4_decode_blocks(const void* d_comp, const uint64_t* d_in_off,
     // Grid cap (SY_LZ4_GRID overrides).  DVFS-warmed sweep on real
   // content at 4 KiB blocks: cap 2048 -> 47 GB/s, 4096 -> 74,
   // 6144 -> 82, 8192 -> 90.5 (and 8 KiB blocks scale the same) —
-  // monotone in resident+queued workgroups; an earlier contrary
-  // reading was cold-clock noise.  Default 16384.
+  // monotone in resident+queued workgroups (8192 -> 197, 16384 -> 216,
+  // 32768 -> 229 GB/s synthetic 4 KiB) — hardware dispatch of queued
+  // workgroups beats grid-striding; effectively uncapped by default.
   static uint32_t grid_cap = 0;
   if (grid_cap == 0) {
     const char* e = getenv("SY_LZ4_GRID");
-    grid_cap = e ? (uint32_t)atoi(e) : 16384;
-    if (grid_cap == 0) grid_cap = 16384;
+    grid_cap = e ? (uint32_t)atoi(e) : (1u << 22);
+    if (grid_cap == 0) grid_cap = 1u << 22;
   }
   uint32_t grid = n_blocks < grid_cap ? n_blocks : grid_cap;
   const uint8_t* c = static_cast<const uint8_t*>(d_comp);
