@@ -27,10 +27,16 @@ from shipyard_amd.data import lz4py
 from shipyard_amd.ops import gf2
 
 
-def timeit(fn, warmup=2, iters=5):
+def timeit(fn, warmup=2, iters=5, warm_seconds=0.7):
+    """DVFS-warmed timing: spin the op for ~warm_seconds before the
+    timed window (cold-clock bursts under-read by up to 2x)."""
     for _ in range(warmup):
         fn()
     torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    while time.perf_counter() - t0 < warm_seconds:
+        fn()
+        torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(iters):
         fn()
