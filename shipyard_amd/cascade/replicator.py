@@ -85,6 +85,13 @@ def image_digest(name: str) -> str:
     return hashlib.sha1(name.encode()).hexdigest()[:16]
 
 
+def _check_image_name(name: str) -> None:
+    """Image names become cache subdirectories — forbid traversal."""
+    if (not name or name.startswith(("/", "."))
+            or ".." in name.split("/")):
+        raise ValueError(f"invalid image name: {name!r}")
+
+
 class Replicator:
     def __init__(self, store: ObjectStore, cache_dir,
                  concurrency: int = 4,
@@ -99,6 +106,7 @@ class Replicator:
     # -- authoring: pack a directory tree as an image's layer set -----
     def pack_image(self, name: str, source_dir,
                    block_raw: int = shardfmt.DEFAULT_BLOCK_RAW) -> dict:
+        _check_image_name(name)
         src = Path(source_dir)
         layers = []
         for p in sorted(src.rglob("*")):
@@ -123,6 +131,7 @@ class Replicator:
         use_gpu None = auto (GPU when available): decode+verify runs
         through shipyard_amd.ops; CPU fallback is the lz4py reference.
         """
+        _check_image_name(name)
         digest = image_digest(name)
         dest = self.cache_dir / name
         done_marker = dest / ".complete"

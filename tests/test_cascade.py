@@ -83,3 +83,13 @@ def test_distribute_and_perf_table(tmp_path):
     spans = cperf.timeline(ex.store)
     assert any(s.startswith("image:img-a") for s in spans)
     ex.store.close()
+
+
+def test_image_name_traversal_rejected(tmp_path):
+    import pytest
+
+    store = ObjectStore(tmp_path / "s")
+    rep = Replicator(store, tmp_path / "cache")
+    for bad in ("../evil", "/abs", "a/../../b", ""):
+        with pytest.raises(ValueError):
+            rep.stage_image(bad)
