@@ -192,7 +192,8 @@ def egress_from_object_store(store: ObjectStore, remote_path: str,
                              include: Sequence[str] = (),
                              exclude: Sequence[str] = (),
                              workers: int = DEFAULT_WORKERS,
-                             verify: bool = True) -> TransferResult:
+                             verify: bool = True,
+                             unpack: bool = True) -> TransferResult:
     names = list(store.list(remote_path, include=list(include) or None,
                             exclude=list(exclude) or None))
     dest = Path(dest_path)
@@ -200,9 +201,9 @@ def egress_from_object_store(store: ObjectStore, remote_path: str,
     total = 0
 
     def one(rel):
-        data = store.download_bytes(rel, verify=verify)
+        data = store.download_bytes(rel, verify=verify, unpack=unpack)
         out_rel = rel[len(remote_path):].lstrip("/")
-        if out_rel.endswith(".syshard"):
+        if unpack and out_rel.endswith(".syshard"):
             out_rel = out_rel[:-len(".syshard")]
         dst = dest / out_rel
         dst.parent.mkdir(parents=True, exist_ok=True)

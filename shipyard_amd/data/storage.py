@@ -58,12 +58,14 @@ class ObjectStore:
         return dst
 
     def download_bytes(self, remote_path: str,
-                       verify: bool = True) -> bytes:
+                       verify: bool = True, unpack: bool = True) -> bytes:
         p = self._path(remote_path)
         if not p.exists() and p.with_suffix(p.suffix + ".syshard").exists():
             p = p.with_suffix(p.suffix + ".syshard")
         data = p.read_bytes()
         if p.suffix == ".syshard" or data[:8] == shardfmt.MAGIC:
+            if not unpack:
+                return data  # raw SYSHARD: consumer decodes (GPU stager)
             return shardfmt.unpack_cpu(data, verify=verify)
         if verify:
             m = integrity.read_manifest(p)

@@ -968,7 +968,8 @@ class LocalExecutor:
                     store, spec["remote_path"], d,
                     include=spec.get("include") or (),
                     exclude=spec.get("exclude") or (),
-                    verify=spec.get("verify", True))
+                    verify=spec.get("verify", True),
+                    unpack=spec.get("decode", True))
             elif ds.kind == "local_batch":
                 src_job, src_task = spec["job_id"], spec["task_id"]
                 src_pool = self._job_pool(src_job)

@@ -1,0 +1,54 @@
+"""End-to-end BASELINE config #4 on one GPU: SYSHARD shards staged
+through the HIP data mover, then a DDP step gang task through the
+executor (world=1 on the single-GPU box; the same path fans to 8)."""
+import os
+import sys
+
+import pytest
+
+torch = pytest.importorskip("torch")
+
+pytestmark = pytest.mark.gpu
+
+from shipyard_amd.data import shardfmt  # noqa: E402
+from shipyard_amd.executor import LocalExecutor  # noqa: E402
+
+
+def test_staged_ddp_step(tmp_path):
+    assert torch.cuda.is_available()
+    ex = LocalExecutor(tmp_path / "root")
+    ex.pool_add({"pool_specification": {
+        "id": "ddp", "gpus": {"dedicated": 1},
+        "inter_node_communication_enabled": True}})
+    # author two synthetic shards into the object store
+    for i in range(2):
+        data = os.urandom(256 * 1024) + bytes(b"w" * 128 * 1024)
+        ex.stores["default"].upload_bytes(
+            f"shards/ddp/s{i}.syshard", shardfmt.pack(data))
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    ex.jobs_add({"job_specifications": [{
+        "id": "ddpjob",
+        "tasks": [{
+            "id": "step",
+            "command": f"{sys.executable} {repo}/benchmarks/ddp_step.py "
+                       "--steps 3 --warmup 1 --hidden 512 --layers 2",
+            "max_task_retries": 0,
+            "input_data": {"local_storage": [
+                {"remote_path": "shards/ddp", "decode": False,
+                 "verify": True}]},
+            "multi_instance": {
+                "num_instances": 1,
+                "gang": {"backend": "rccl", "gpus_per_rank": 1}},
+        }],
+    }]}, "ddp")
+    ex.run_until_idle(timeout=300)
+    t = ex.tasks_list("ddpjob")[0]
+    base = ex.pool_root("ddp") / "jobs" / "ddpjob" / "tasks" / "step"
+    err = (base / "rank000" / "stderr.txt").read_text()
+    assert t["state"] == "completed", err[-800:]
+    out = (base / "rank000" / "stdout.txt").read_text()
+    assert '"workload": "ddp-step"' in out
+    assert '"staged_shard_bytes"' in out
+    # the mover materialized + verified the shards into the task dir
+    assert '"staged_shard_bytes": 786432' in out
+    ex.store.close()
