@@ -44,9 +44,10 @@ def test_staged_ddp_step(tmp_path):
     ex.run_until_idle(timeout=300)
     t = ex.tasks_list("ddpjob")[0]
     base = ex.pool_root("ddp") / "jobs" / "ddpjob" / "tasks" / "step"
-    err = (base / "rank000" / "stderr.txt").read_text()
+    # a 1-rank gang uses the single-task layout (rank dirs appear at n>1)
+    err = (base / "stderr.txt").read_text()
     assert t["state"] == "completed", err[-800:]
-    out = (base / "rank000" / "stdout.txt").read_text()
+    out = (base / "stdout.txt").read_text()
     assert '"workload": "ddp-step"' in out
     assert '"staged_shard_bytes"' in out
     # the mover materialized + verified the shards into the task dir
