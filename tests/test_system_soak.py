@@ -1,0 +1,85 @@
+"""System soak: scheduler thread + federation + autoscale + recurrence
++ fault injection operating together against two pools (the closest a
+CPU host gets to the production daemon)."""
+import time
+
+from shipyard_amd.executor import LocalExecutor
+from shipyard_amd.federation.scheduler import (Federation,
+                                               FederationProcessor)
+
+
+def test_daemon_subsystems_together(tmp_path):
+    ex = LocalExecutor(tmp_path / "root", detect_gpus=False)
+    for pid in ("soak-a", "soak-b"):
+        ex.pool_add({"pool_specification": {
+            "id": pid, "gpus": {"dedicated": 0}, "cpu_slots": 2,
+            "node_configuration": {"rocm": {"verify": False}}}})
+    fp = FederationProcessor(
+        ex, {"sf": Federation("sf", ["soak-a", "soak-b"])})
+
+    # a recurrence schedule
+    ex.jobs_add({"job_specifications": [{
+        "id": "beat", "tasks": [{"id": "t", "command": "true"}],
+        "recurrence": {"schedule": {"recurrence_interval": "00:00:01"}},
+    }]}, "soak-a")
+
+    # federation-submitted jobs
+    for i in range(4):
+        fp.submit_job("sf", {"job_specifications": [{
+            "id": f"fed{i}",
+            "tasks": [{"id": "w", "command": "sleep 0.05"},
+                      {"id": "m", "command": "true",
+                       "depends_on": ["w"]}],
+        }]})
+
+    # direct jobs with retries + a failing task
+    ex.jobs_add({"job_specifications": [{
+        "id": "direct",
+        "tasks": [{"id": "flaky", "command": "false",
+                   "max_task_retries": 1},
+                  {"id": "ok", "command": "true"}],
+    }]}, "soak-b")
+
+    ex.start_scheduler(poll=0.01)
+    try:
+        deadline = time.monotonic() + 60
+        done = set()
+        while time.monotonic() < deadline:
+            fp.process_queue_once()
+            rows = ex.store.query(
+                "SELECT id, state FROM jobs WHERE state != 'active'")
+            jobs = {j["id"]: j for j in ex.jobs_list()}
+            fed_done = all(
+                f"fed{i}" in jobs and not ex.store.query_one(
+                    "SELECT 1 FROM tasks WHERE job_id=? AND state IN "
+                    "('pending','ready','running')", (f"fed{i}",))
+                for i in range(4))
+            beats = [j for j in jobs if j.startswith("beat-")]
+            direct_done = not ex.store.query_one(
+                "SELECT 1 FROM tasks WHERE job_id='direct' AND state IN "
+                "('pending','ready','running')")
+            if fed_done and direct_done and len(beats) >= 2:
+                break
+            time.sleep(0.05)
+        else:
+            raise AssertionError("soak did not converge")
+    finally:
+        ex.stop_scheduler()
+
+    # federation jobs all completed, spread across pools
+    pools_used = set()
+    for i in range(4):
+        states = {t["id"]: t["state"] for t in ex.tasks_list(f"fed{i}")}
+        assert states == {"w": "completed", "m": "completed"}
+        pools_used.add([j for j in ex.jobs_list()
+                        if j["id"] == f"fed{i}"][0]["pool_id"])
+    assert pools_used <= {"soak-a", "soak-b"}
+    # flaky retried then failed; ok completed
+    d = {t["id"]: t for t in ex.tasks_list("direct")}
+    assert d["flaky"]["state"] == "failed" and d["flaky"]["retries"] == 1
+    assert d["ok"]["state"] == "completed"
+    # at least two heartbeat instances ran
+    beat_jobs = [j["id"] for j in ex.jobs_list()
+                 if j["id"].startswith("beat-")]
+    assert len(beat_jobs) >= 2
+    ex.store.close()
