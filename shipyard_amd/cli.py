@@ -252,7 +252,9 @@ def jobs_add(ctx, poolid, tail, wait, recreate, configdir, root, raw):
             if ctx.executor.store.query_one(
                     "SELECT id FROM jobs WHERE id=?", (spec["id"],)):
                 ctx.executor.job_del(spec["id"])
-    added = ctx.executor.jobs_add(ctx.conf(ConfigType.jobs), pid)
+    added = ctx.executor.jobs_add(
+        ctx.conf(ConfigType.jobs), pid,
+        pool_conf=ctx.conf(ConfigType.pool, required=False))
     if wait or tail:
         ctx.executor.run_until_idle()
     out = {"jobs": added}

@@ -307,11 +307,32 @@ class LocalExecutor:
     # ----------------------------------------------------------------
     # jobs (reference convoy/batch.py:5056 add_jobs)
     # ----------------------------------------------------------------
-    def jobs_add(self, jobs_conf: Dict[str, Any], pool_id: str) -> List[str]:
-        ps = self._pool_settings(pool_id)
+    def jobs_add(self, jobs_conf: Dict[str, Any], pool_id: str,
+                 pool_conf: Optional[Dict[str, Any]] = None) -> List[str]:
         added = []
         for jobspec in jobs_conf["job_specifications"]:
             js = cfg.job_settings(jobspec)
+            job_pool = pool_id
+            if jobspec.get("auto_pool") is not None:
+                # auto-pool (reference fleet.py:2904 + batch.py autopool):
+                # a dedicated pool per job, deleted with the job unless
+                # keep_alive
+                if pool_conf is None:
+                    raise ExecutorError(
+                        "auto_pool requires the pool specification")
+                import copy as _copy
+
+                ap_conf = _copy.deepcopy(pool_conf)
+                ap_id = f"{js.id}-autopool"
+                ap_conf["pool_specification"]["id"] = ap_id
+                if not self.store.query_one(
+                        "SELECT id FROM pools WHERE id=?", (ap_id,)):
+                    self.pool_add(ap_conf)
+                job_pool = ap_id
+                keep = bool(jobspec["auto_pool"].get("keep_alive", False))
+                self.store.kv_set(f"autopool:{js.id}",
+                                  json.dumps({"pool": ap_id, "keep": keep}))
+            ps = self._pool_settings(job_pool)
             if js.recurrence is not None:
                 # job schedule (reference convoy/batch.py:5390 JobSchedule):
                 # register; instances materialize from process_schedules()
@@ -327,13 +348,31 @@ class LocalExecutor:
             self.store.execute(
                 "INSERT INTO jobs (id, pool_id, spec_json, state, priority,"
                 " auto_complete, created_at) VALUES (?,?,?,?,?,?,?)",
-                (js.id, pool_id, json.dumps(jobspec), "active", js.priority,
+                (js.id, job_pool, json.dumps(jobspec), "active", js.priority,
                  int(js.auto_complete), time.time()))
             self._add_tasks_for_job(js, jobspec, ps)
             added.append(js.id)
             self.store.add_event(f"job:{js.id}", "submitted",
-                                 {"pool": pool_id})
+                                 {"pool": job_pool})
         return added
+
+    def _reap_auto_pool(self, job_id: str) -> None:
+        raw = self.store.kv_get(f"autopool:{job_id}")
+        if not raw:
+            return
+        rec = json.loads(raw)
+        self.store.execute("DELETE FROM kv WHERE key=?",
+                           (f"autopool:{job_id}",))
+        if rec.get("keep"):
+            return
+        # remove slots + pool row only — the job's records and task
+        # files outlive the auto-pool (reference autopool semantics)
+        with self.store.transaction() as conn:
+            conn.execute("DELETE FROM slots WHERE pool_id=?",
+                         (rec["pool"],))
+            conn.execute("DELETE FROM pools WHERE id=?", (rec["pool"],))
+        self.store.add_event(f"pool:{rec['pool']}", "autopool-reaped",
+                             {"job": job_id})
 
     def _autogen_id(self, js: cfg.JobSettings, seq: int) -> str:
         return f"{js.autogen_task_id_prefix}" \
@@ -545,6 +584,7 @@ class LocalExecutor:
         except Exception:
             pass
         self.job_terminate(job_id)
+        self._reap_auto_pool(job_id)
         with self.store.transaction() as conn:
             conn.execute("DELETE FROM tasks WHERE job_id=?", (job_id,))
             conn.execute("DELETE FROM task_deps WHERE job_id=?", (job_id,))
@@ -1110,6 +1150,7 @@ class LocalExecutor:
                 self.store.execute(
                     "UPDATE jobs SET state='completed', completed_at=? "
                     "WHERE id=?", (time.time(), jid))
+                self._reap_auto_pool(jid)
 
     # ----------------------------------------------------------------
     # files (reference convoy/batch.py:3243 stream_file_and_wait)

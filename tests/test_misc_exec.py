@@ -161,3 +161,26 @@ def test_task_terminate_single(tmp_path):
     states = {t["id"]: t["state"] for t in ex.tasks_list("tj")}
     assert states["slow"] == "cancelled" and states["fast"] == "completed"
     ex.store.close()
+
+
+def test_auto_pool_lifecycle(tmp_path):
+    """auto_pool jobs get a dedicated pool, reaped at completion."""
+    ex = LocalExecutor(tmp_path / "ap", detect_gpus=False)
+    mkpool(ex, "template-unused")  # a normal pool must not be touched
+    pool_conf = {"pool_specification": {
+        "id": "tmpl", "gpus": {"dedicated": 0}, "cpu_slots": 1,
+        "node_configuration": {"rocm": {"verify": False}}}}
+    ex.jobs_add({"job_specifications": [{
+        "id": "apjob", "auto_complete": True,
+        "auto_pool": {"keep_alive": False, "pool_lifetime": "job"},
+        "tasks": [{"id": "t", "command": "echo auto-pool-ok"}],
+    }]}, pool_id=None, pool_conf=pool_conf)
+    pools = {p["id"] for p in ex.pool_list()}
+    assert "apjob-autopool" in pools
+    ex.run_until_idle(timeout=30)
+    jobs = {j["id"]: j["state"] for j in ex.jobs_list()}
+    assert jobs["apjob"] == "completed"
+    pools = {p["id"] for p in ex.pool_list()}
+    assert "apjob-autopool" not in pools       # reaped
+    assert "template-unused" in pools          # untouched
+    ex.store.close()
