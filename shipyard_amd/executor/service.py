@@ -410,7 +410,17 @@ class LocalExecutor:
 
         for taskspec in js.tasks:
             if taskspec.get("task_factory"):
-                expanded = task_factory.generate_tasks(taskspec)
+                # file factories enumerate the object store: resolve the
+                # account named in the spec (default store otherwise)
+                froot = None
+                fspec = (taskspec["task_factory"].get("file") or {})
+                account = (fspec.get("local_storage") or {}).get(
+                    "storage_account_settings", "default")
+                store = self.stores.get(account)
+                if store is not None:
+                    froot = store.root
+                expanded = task_factory.generate_tasks(
+                    taskspec, storage_root=froot)
             else:
                 expanded = [taskspec]
             for spec1 in expanded:

@@ -150,8 +150,10 @@ def generate_tasks(taskspec: Dict[str, Any],
         for value in _random_iter(tf["random"]):
             out.append(_format_command(taskspec, value))
     elif "file" in tf:
-        root = storage_root or Path(".")
-        for kw in _file_iter(tf["file"], root):
+        if storage_root is None:
+            raise TaskFactoryError(
+                "file task_factory requires an object store root")
+        for kw in _file_iter(tf["file"], Path(storage_root)):
             out.append(_format_command(taskspec, kw))
     elif "repeat" in tf:
         for _ in range(tf["repeat"]):
