@@ -1014,12 +1014,16 @@ class LocalExecutor:
                                              "default")]
                 d = Path(utils.expand_env(spec.get("local_path") or
                                           str(dest)))
-                mover.egress_from_object_store(
+                res = mover.egress_from_object_store(
                     store, spec["remote_path"], d,
                     include=spec.get("include") or (),
                     exclude=spec.get("exclude") or (),
                     verify=spec.get("verify", True),
                     unpack=spec.get("decode", True))
+                self.store.add_perf(
+                    f"mover:{jid}/{tid}", "xfer-end",
+                    {"direction": "in", "bytes": res.bytes,
+                     "seconds": res.seconds, "files": res.files})
             elif ds.kind == "local_batch":
                 src_job, src_task = spec["job_id"], spec["task_id"]
                 src_pool = self._job_pool(src_job)

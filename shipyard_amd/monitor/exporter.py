@@ -88,6 +88,31 @@ def collect_gpu_metrics() -> List[Dict[str, float]]:
     return []
 
 
+def collect_perf_metrics(store) -> Dict[str, float]:
+    """Cascade/data-mover counters from the perf table (the
+    table_perf-analogue, reference cascade/perf.py)."""
+    import json as _json
+
+    out: Dict[str, float] = {
+        "cascade_pulls": 0.0, "cascade_pull_seconds": 0.0,
+        "cascade_raw_bytes": 0.0, "cascade_comp_bytes": 0.0,
+        "mover_transfers": 0.0, "mover_bytes": 0.0,
+        "mover_seconds": 0.0,
+    }
+    for r in store.query("SELECT source, event, payload FROM perf"):
+        payload = _json.loads(r["payload"]) if r["payload"] else {}
+        if r["event"] == "pull-end":
+            out["cascade_pulls"] += 1
+            out["cascade_pull_seconds"] += payload.get("seconds", 0.0)
+            out["cascade_raw_bytes"] += payload.get("raw_bytes", 0)
+            out["cascade_comp_bytes"] += payload.get("comp_bytes", 0)
+        elif r["event"] == "xfer-end":
+            out["mover_transfers"] += 1
+            out["mover_bytes"] += payload.get("bytes", 0)
+            out["mover_seconds"] += payload.get("seconds", 0.0)
+    return out
+
+
 def collect_executor_metrics(store) -> Dict[str, float]:
     out: Dict[str, float] = {}
     for r in store.query(
@@ -138,6 +163,8 @@ class Exporter:
                 self._g_gpu.labels(gpu=str(i), name=k).set(v)
         if self.store is not None:
             for k, v in collect_executor_metrics(self.store).items():
+                self._g_ex.labels(name=k).set(v)
+            for k, v in collect_perf_metrics(self.store).items():
                 self._g_ex.labels(name=k).set(v)
         return generate_latest(self._registry)
 

@@ -97,3 +97,25 @@ def test_recurrence_through_jobs_add(tmp_path):
     assert ex.schedule_del("schd")
     assert ex.schedules_list() == []
     ex.store.close()
+
+
+def test_perf_collectors_in_scrape(tmp_path):
+    ex = LocalExecutor(tmp_path / "pm", detect_gpus=False)
+    ex.pool_add({"pool_specification": {
+        "id": "pm", "gpus": {"dedicated": 0}, "cpu_slots": 1,
+        "node_configuration": {"rocm": {"verify": False}}}})
+    ex.stores["default"].upload_bytes("seed/a.bin", b"x" * 1000)
+    ex.jobs_add({"job_specifications": [{
+        "id": "pj", "tasks": [{
+            "id": "t", "command": "true",
+            "input_data": {"local_storage": [{"remote_path": "seed"}]},
+        }]}]}, "pm")
+    ex.run_until_idle(timeout=30)
+    body = Exporter(store=ex.store).scrape().decode()
+    assert 'name="mover_transfers"' in body
+    assert 'name="mover_bytes"' in body
+    from shipyard_amd.monitor.exporter import collect_perf_metrics
+
+    m = collect_perf_metrics(ex.store)
+    assert m["mover_transfers"] == 1 and m["mover_bytes"] == 1000
+    ex.store.close()
