@@ -421,6 +421,44 @@ def data_files():
     """Task file access."""
 
 
+@data_files.command("list")
+@click.option("--jobid", required=True)
+@click.option("--taskid", required=True)
+@_common
+@pass_ctx
+def files_list(ctx, jobid, taskid, configdir, root, raw):
+    """List a task's files (reference `data files list`)."""
+    _apply(ctx, configdir, root, raw)
+    pool_id = ctx.executor._job_pool(jobid)
+    base = (ctx.executor.pool_root(pool_id) / "jobs" / jobid / "tasks" /
+            taskid)
+    out = []
+    if base.exists():
+        for p in sorted(base.rglob("*")):
+            if p.is_file():
+                out.append({"path": str(p.relative_to(base)),
+                            "bytes": p.stat().st_size})
+    ctx.emit(out)
+
+
+@data_files.command("getall")
+@click.option("--jobid", required=True)
+@click.option("--taskid", required=True)
+@click.option("--dest", required=True)
+@_common
+@pass_ctx
+def files_getall(ctx, jobid, taskid, dest, configdir, root, raw):
+    """Copy a task's directory out (reference `data files task getall`)."""
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.data import mover as _mover
+
+    pool_id = ctx.executor._job_pool(jobid)
+    base = (ctx.executor.pool_root(pool_id) / "jobs" / jobid / "tasks" /
+            taskid)
+    res = _mover.ingress_directory(base, dest)
+    ctx.emit({"files": res.files, "bytes": res.bytes})
+
+
 @data_files.command("stream")
 @click.option("--filespec", required=True,
               help="jobid,taskid[,filename]")

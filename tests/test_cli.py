@@ -137,3 +137,16 @@ def test_jobs_add_recreate(configdir, tmp_path):
     r = run(["diag", "du"], configdir, tmp_path)
     assert r.exit_code == 0
     assert "pools" in r.output
+
+
+def test_files_list_and_getall(configdir, tmp_path):
+    run(["pool", "add"], configdir, tmp_path)
+    run(["jobs", "add", "--wait"], configdir, tmp_path)
+    r = run(["data", "files", "list", "--jobid", "clijob",
+             "--taskid", "t1"], configdir, tmp_path)
+    assert r.exit_code == 0 and "stdout.txt" in r.output
+    dest = tmp_path / "fetched"
+    r = run(["data", "files", "getall", "--jobid", "clijob",
+             "--taskid", "t1", "--dest", str(dest)], configdir, tmp_path)
+    assert r.exit_code == 0
+    assert (dest / "stdout.txt").exists()
