@@ -73,16 +73,22 @@ def collect_gpu_metrics() -> List[Dict[str, float]]:
                 if not card.startswith("card"):
                     continue
                 m = {}
-                for key, dst in (
-                        ("GPU use (%)", "gfx_activity_pct"),
-                        ("GPU Memory Allocated (VRAM%)", "vram_used_pct"),
-                        ("Average Graphics Package Power (W)", "power_w"),
-                        ("Temperature (Sensor junction) (C)", "temp_c")):
-                    if key in vals:
-                        try:
-                            m[dst] = float(vals[key])
-                        except (TypeError, ValueError):
-                            pass
+                exact = {
+                    "GPU use (%)": "gfx_activity_pct",
+                    "GPU Memory Allocated (VRAM%)": "vram_used_pct",
+                    "Temperature (Sensor junction) (C)": "temp_c",
+                }
+                for key, val in vals.items():
+                    dst = exact.get(key)
+                    # power key name varies across rocm-smi versions
+                    if dst is None and "Power (W)" in key:
+                        dst = "power_w"
+                    if dst is None:
+                        continue
+                    try:
+                        m[dst] = float(val)
+                    except (TypeError, ValueError):
+                        pass
                 out.append(m)
             return out
     return []
