@@ -50,7 +50,7 @@ __global__ __launch_bounds__(kThreads) void crc32c_chunks_kernel(
     const uint8_t* __restrict__ data, uint64_t n_bytes, uint32_t chunk_size,
     const uint32_t* __restrict__ level_mats,  // [NLEVELS][32]
     uint32_t* __restrict__ out_raw, uint64_t n_chunks) {
-  __shared__ uint32_t tab[8][256];
+  __shared__ uint32_t tab[16][256];
   __shared__ uint32_t lane_crc[kThreads];
   __shared__ uint32_t mats[NLEVELS][32];
 
@@ -67,7 +67,7 @@ __global__ __launch_bounds__(kThreads) void crc32c_chunks_kernel(
     mats[i / 32][i % 32] = level_mats[i];
   }
   __syncthreads();
-  for (int k = 1; k < 8; ++k) {
+  for (int k = 1; k < 16; ++k) {
     uint32_t prev = tab[k - 1][t];
     tab[k][t] = (prev >> 8) ^ tab[0][prev & 0xffu];
     __syncthreads();
@@ -127,17 +127,19 @@ __global__ __launch_bounds__(kThreads) void crc32c_chunks_kernel(
       for (int c = 0; c < NCHAINS; ++c) {
         if (!act[c]) continue;
         any = true;
-        uint32_t a = v[c].x ^ crc[c], b = v[c].y;
-        crc[c] = tab[7][a & 0xffu] ^ tab[6][(a >> 8) & 0xffu] ^
-                 tab[5][(a >> 16) & 0xffu] ^ tab[4][a >> 24] ^
-                 tab[3][b & 0xffu] ^ tab[2][(b >> 8) & 0xffu] ^
-                 tab[1][(b >> 16) & 0xffu] ^ tab[0][b >> 24];
-        a = v[c].z ^ crc[c];
-        b = v[c].w;
-        crc[c] = tab[7][a & 0xffu] ^ tab[6][(a >> 8) & 0xffu] ^
-                 tab[5][(a >> 16) & 0xffu] ^ tab[4][a >> 24] ^
-                 tab[3][b & 0xffu] ^ tab[2][(b >> 8) & 0xffu] ^
-                 tab[1][(b >> 16) & 0xffu] ^ tab[0][b >> 24];
+        // slice-by-16: ONE dependent lookup group per 16 B (the
+        // previous slice-by-8 x2 put two serial LDS round trips on
+        // the chain; measured model: chain latency x occupancy bound)
+        const uint32_t a = v[c].x ^ crc[c];
+        const uint32_t b = v[c].y, d = v[c].z, e = v[c].w;
+        crc[c] = tab[15][a & 0xffu] ^ tab[14][(a >> 8) & 0xffu] ^
+                 tab[13][(a >> 16) & 0xffu] ^ tab[12][a >> 24] ^
+                 tab[11][b & 0xffu] ^ tab[10][(b >> 8) & 0xffu] ^
+                 tab[9][(b >> 16) & 0xffu] ^ tab[8][b >> 24] ^
+                 tab[7][d & 0xffu] ^ tab[6][(d >> 8) & 0xffu] ^
+                 tab[5][(d >> 16) & 0xffu] ^ tab[4][d >> 24] ^
+                 tab[3][e & 0xffu] ^ tab[2][(e >> 8) & 0xffu] ^
+                 tab[1][(e >> 16) & 0xffu] ^ tab[0][e >> 24];
         p[c] += 16;
         rem[c] -= 16;
       }
