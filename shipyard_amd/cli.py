@@ -462,14 +462,21 @@ def files_getall(ctx, jobid, taskid, dest, configdir, root, raw):
 @data_files.command("stream")
 @click.option("--filespec", required=True,
               help="jobid,taskid[,filename]")
+@click.option("--follow/--no-follow", default=False,
+              help="print incrementally while the task runs")
 @_common
 @pass_ctx
-def files_stream(ctx, filespec, configdir, root, raw):
+def files_stream(ctx, filespec, follow, configdir, root, raw):
     _apply(ctx, configdir, root, raw)
+    import sys as _sys
+
     parts = filespec.split(",")
     jid, tid = parts[0], parts[1]
     name = parts[2] if len(parts) > 2 else "stdout.txt"
-    click.echo(ctx.executor.stream_task_file(jid, tid, name))
+    sink = _sys.stdout.write if follow else None
+    content = ctx.executor.stream_task_file(jid, tid, name, sink=sink)
+    if not follow:
+        click.echo(content)
 
 
 # ---------------------------------------------------------------- storage

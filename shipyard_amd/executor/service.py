@@ -1176,18 +1176,42 @@ class LocalExecutor:
 
     def stream_task_file(self, job_id: str, task_id: str,
                          name: str = "stdout.txt",
-                         timeout: float = 60.0) -> str:
+                         timeout: float = 60.0,
+                         sink=None) -> str:
+        """Stream a task file until the task finishes (reference
+        convoy/batch.py:3243 stream_file_and_wait_for_task).  When
+        ``sink`` is given (e.g. sys.stdout.write) content is delivered
+        INCREMENTALLY as the file grows (live tail); the full content
+        is returned either way."""
         pool_id = self._job_pool(job_id)
         path = self.task_file(pool_id, job_id, task_id, name)
         deadline = time.monotonic() + timeout
+        pos = 0
+        chunks: List[str] = []
+
+        def drain() -> None:
+            nonlocal pos
+            if not path.exists():
+                return
+            with open(path, "r") as f:
+                f.seek(pos)
+                new = f.read()
+            if new:
+                pos += len(new)
+                chunks.append(new)
+                if sink is not None:
+                    sink(new)
+
         while True:
             self.schedule_once()
+            drain()
             row = self.store.query_one(
                 "SELECT state FROM tasks WHERE job_id=? AND id=?",
                 (job_id, task_id))
             if row and row["state"] in ("completed", "failed", "cancelled"):
+                drain()
                 break
             if time.monotonic() > deadline:
                 raise TimeoutError("task did not finish")
             time.sleep(0.05)
-        return path.read_text() if path.exists() else ""
+        return "".join(chunks)

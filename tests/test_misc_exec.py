@@ -184,3 +184,19 @@ def test_auto_pool_lifecycle(tmp_path):
     assert "apjob-autopool" not in pools       # reaped
     assert "template-unused" in pools          # untouched
     ex.store.close()
+
+
+def test_live_tail_streams_incrementally(tmp_path):
+    ex = LocalExecutor(tmp_path / "lt", detect_gpus=False)
+    mkpool(ex, "lp")
+    ex.jobs_add({"job_specifications": [{
+        "id": "lj", "tasks": [{
+            "id": "t",
+            "command": "echo first; sleep 0.4; echo second"}]}]}, "lp")
+    seen = []
+    out = ex.stream_task_file("lj", "t", sink=seen.append, timeout=30)
+    assert "first" in out and "second" in out
+    # incremental: 'first' arrived in an earlier chunk than 'second'
+    assert len(seen) >= 2
+    assert "first" in seen[0]
+    ex.store.close()
