@@ -66,8 +66,11 @@ class CliContext:
         return self._executor
 
     def emit(self, obj) -> None:
+        """--raw = compact single-line JSON (machine), default pretty
+        (the reference's --raw convention, shipyard.py:634-647)."""
         if self.raw:
-            click.echo(json.dumps(obj, indent=2, default=str))
+            click.echo(json.dumps(obj, separators=(",", ":"),
+                                  default=str))
         else:
             click.echo(json.dumps(obj, indent=2, default=str))
 
