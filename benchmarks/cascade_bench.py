@@ -29,6 +29,16 @@ def main(layer_mb: int = 8, n_layers: int = 2) -> None:
         # vary layers slightly so they are distinct objects
         (src / f"layer{i}.bin").write_bytes(bytes([i]) * 4096 + base)
 
+    # exclude one-time CUDA context creation from the staging time
+    try:
+        import torch
+
+        if torch.cuda.is_available():
+            torch.zeros(1, device="cuda")
+            torch.cuda.synchronize()
+    except Exception:
+        pass
+
     store = ObjectStore(td / "store")
     events = []
     rep = Replicator(store, td / "cache",
