@@ -86,8 +86,12 @@ class GangComm:
 
         bucket, size = [], 0
         for t in list(tensors) + [None]:
-            flush = t is None or (size + t.numel() * t.element_size()
-                                  > bucket_bytes and bucket)
+            # flush on size overflow AND on dtype change (torch.cat
+            # cannot mix dtypes; mixed fp32/bf16 grads are common)
+            flush = t is None or (bucket and (
+                size + t.numel() * t.element_size() > bucket_bytes
+                or t.dtype != bucket[0].dtype
+                or t.device != bucket[0].device))
             if flush and bucket:
                 flat = torch.cat([b.reshape(-1) for b in bucket])
                 self._dist.all_reduce(flat)

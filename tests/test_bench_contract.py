@@ -1,0 +1,52 @@
+"""Driver-contract guard: `python bench.py` emits exactly one JSON line
+with every field the round harness depends on."""
+import json
+import subprocess
+import sys
+from pathlib import Path
+
+REPO = Path(__file__).parents[1]
+
+REQUIRED = {"metric", "value", "unit", "n_gpus", "steps", "warmup",
+            "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+            "dtype", "data", "config"}
+
+
+def test_bench_json_contract():
+    out = subprocess.run(
+        [sys.executable, str(REPO / "bench.py"), "--steps", "2",
+         "--warmup", "1", "--payload-mb", "4"],
+        capture_output=True, text=True, timeout=300, cwd=str(REPO))
+    assert out.returncode == 0, out.stderr[-500:]
+    lines = [ln for ln in out.stdout.splitlines() if ln.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    doc = json.loads(lines[0])
+    assert REQUIRED <= set(doc)
+    assert doc["metric"] == "rccl_allreduce_bus_GBps"
+    assert doc["n_gpus"] == 1 and doc["steps"] == 2 and doc["warmup"] == 1
+    assert doc["higher_is_better"] is True
+    assert doc["scaling"] == "weak"
+    assert doc["dtype"] == "bf16" and doc["data"] == "synthetic"
+    assert isinstance(doc["value"], (int, float)) and doc["value"] > 0
+    cfg = doc["config"]
+    for key in ("model", "global_batch", "seq_len", "parallelism"):
+        assert key in cfg
+    # p50 measured through the executor
+    assert cfg["submit_launch_p50_ms"] is None or \
+        cfg["submit_launch_p50_ms"] > 0
+
+
+def test_bench_world2_gloo_contract():
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29533", str(REPO / "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1",
+         "--payload-mb", "2"],
+        capture_output=True, text=True, timeout=300, cwd=str(REPO))
+    assert out.returncode == 0, out.stderr[-800:]
+    lines = [ln for ln in out.stdout.splitlines() if ln.startswith("{")]
+    assert len(lines) == 1, out.stdout  # rank 0 only
+    doc = json.loads(lines[0])
+    assert doc["n_gpus"] == 2
+    assert doc["config"]["parallelism"] == "gang2"
