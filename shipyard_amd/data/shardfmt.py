@@ -122,6 +122,23 @@ def unpack_cpu(buf: bytes, verify: bool = True) -> bytes:
     return data
 
 
+
+def _stored_contiguous(idx) -> bool:
+    """True when every block is stored AND the payload layout equals the
+    raw layout (comp_off == cumulative raw offset) — then the payload IS
+    the decoded data and no per-block copies are needed.  This is the
+    common case for incompressible shards packed with compress=False
+    (block sizes that are 16 B multiples keep offsets aligned)."""
+    acc = 0
+    for b in idx.blocks:
+        if not b.stored or b.comp_off != acc:
+            return False
+        acc += b.raw_len + (-b.raw_len) % 16
+        if b.raw_len % 16 and b is not idx.blocks[-1]:
+            return False
+    return True
+
+
 def unpack_gpu(buf: bytes, device=None, verify: bool = True):
     """GPU reader: upload payload once, LZ4-decode all blocks on the
     MI355X, CRC32C-verify the decoded bytes, return a uint8 CUDA tensor.
@@ -142,6 +159,16 @@ def unpack_gpu(buf: bytes, device=None, verify: bool = True):
     out = torch.empty(max(idx.raw_size, 1), dtype=torch.uint8, device=dev)
     if not idx.blocks:
         return out[:0]
+
+    if _stored_contiguous(idx):
+        out = d_comp[:idx.raw_size]
+        if verify:
+            crcs = ops.crc32c_chunks(out.contiguous(),
+                                     chunk_size=idx.block_raw)
+            want = [b.crc32c for b in idx.blocks]
+            if [int(x) for x in crcs.tolist()] != want:
+                raise ValueError("GPU CRC mismatch in stored shard")
+        return out
 
     lz4_blocks = [(i, b) for i, b in enumerate(idx.blocks) if not b.stored]
     stored_blocks = [(i, b) for i, b in enumerate(idx.blocks) if b.stored]

@@ -151,6 +151,20 @@ class ShardStager:
 
         from shipyard_amd import ops
 
+        from shipyard_amd.data.shardfmt import _stored_contiguous
+
+        if _stored_contiguous(idx):
+            out = d_payload[:idx.raw_size]
+            if self.verify and idx.blocks:
+                with torch.cuda.stream(self.stream):
+                    crcs = ops.crc32c_chunks(out.contiguous(),
+                                             chunk_size=idx.block_raw)
+                self.stream.synchronize()
+                want = [b.crc32c for b in idx.blocks]
+                if [int(x) for x in crcs.tolist()] != want:
+                    raise ValueError("CRC mismatch in staged shard")
+            return out
+
         out = torch.empty(max(idx.raw_size, 1), dtype=torch.uint8,
                           device=self.device)
         raw_offs = []
