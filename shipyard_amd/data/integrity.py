@@ -45,16 +45,15 @@ class Manifest:
 
 def compute_cpu(data: bytes, chunk_size: int = DEFAULT_CHUNK,
                 with_sha_root: bool = True) -> Manifest:
-    crcs = []
-    sha = hashlib.sha256()
-    for off in range(0, len(data), chunk_size):
-        chunk = data[off:off + chunk_size]
-        crcs.append(gf2.crc32c(chunk))
-        if with_sha_root:
-            sha.update(hashlib.sha256(chunk).digest())
+    crcs = gf2.crc32c_chunks_numpy(data, chunk_size)
+    root = None
+    if with_sha_root:
+        sha = hashlib.sha256()
+        for off in range(0, len(data), chunk_size):
+            sha.update(hashlib.sha256(data[off:off + chunk_size]).digest())
+        root = sha.hexdigest()
     return Manifest(length=len(data), chunk_size=chunk_size,
-                    chunk_crc32c=crcs,
-                    sha256_root=sha.hexdigest() if with_sha_root else None)
+                    chunk_crc32c=crcs, sha256_root=root)
 
 
 def compute_gpu(data_tensor, chunk_size: int = DEFAULT_CHUNK,

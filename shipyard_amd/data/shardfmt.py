@@ -71,12 +71,13 @@ def pack(data: bytes, block_raw: int = DEFAULT_BLOCK_RAW,
     blocks: List[BlockEntry] = []
     payload = io.BytesIO()
     off = 0
-    for boff in range(0, len(data), block_raw) or [0]:
+    crcs = gf2.crc32c_chunks_numpy(data, block_raw)
+    for bi, boff in enumerate(range(0, len(data), block_raw) or [0]):
         raw = data[boff:boff + block_raw]
         comp = lz4py.compress_block(raw) if compress else raw
         if not compress or len(comp) >= len(raw):
             comp = raw  # stored
-        crc = gf2.crc32c(raw)
+        crc = crcs[bi] if bi < len(crcs) else gf2.crc32c(raw)
         blocks.append(BlockEntry(off, len(comp), len(raw), crc))
         payload.write(comp)
         pad = _align16(len(comp)) - len(comp)
