@@ -554,6 +554,19 @@ def diag_du(ctx, configdir, root, raw):
     ctx.emit(out)
 
 
+@diag.command("prune")
+@click.option("--older-than-hours", type=float, default=168.0)
+@_common
+@pass_ctx
+def diag_prune(ctx, older_than_hours, configdir, root, raw):
+    """Trim old events/perf rows."""
+    _apply(ctx, configdir, root, raw)
+    import time as _time
+
+    n = ctx.executor.store.prune(_time.time() - older_than_hours * 3600)
+    ctx.emit({"pruned": n})
+
+
 @diag.command("timeline")
 @_common
 @pass_ctx

@@ -157,6 +157,18 @@ class Store:
             (ts or time.time(), source, event,
              json.dumps(payload) if payload else None))
 
+    def prune(self, before_ts: float) -> int:
+        """Trim events/perf rows older than ``before_ts`` (the tables
+        grow per task/transfer; reference's tables have TTL semantics
+        via storage lifecycle).  Returns rows removed."""
+        with self._lock:
+            c1 = self._conn.execute(
+                "DELETE FROM events WHERE ts < ?", (before_ts,)).rowcount
+            c2 = self._conn.execute(
+                "DELETE FROM perf WHERE ts < ?", (before_ts,)).rowcount
+            self._conn.commit()
+        return c1 + c2
+
     # -- kv ----------------------------------------------------------
     def kv_set(self, key: str, value: str) -> None:
         self.execute(
