@@ -35,7 +35,30 @@ class CliContext:
         if self._bundle is None:
             self._bundle = ConfigBundle.from_dir(self.configdir or ".",
                                                  **self.overrides)
+            self._deref_secrets()
         return self._bundle
+
+    def _deref_secrets(self) -> None:
+        """Replace *_secret_id values from the secrets store when one is
+        configured AND its passphrase is present (reference
+        keyvault.py:196 parse_secret_ids semantics)."""
+        import os as _os
+
+        creds = self._bundle.get(ConfigType.credentials)
+        if not creds:
+            return
+        ss = (creds.get("credentials", {}).get("secrets_store") or {})
+        if not ss.get("file"):
+            return
+        env = ss.get("passphrase_env", "SHIPYARD_SECRETS_PASSPHRASE")
+        if not _os.environ.get(env):
+            return  # no passphrase: leave references untouched
+        from shipyard_amd.config.secrets import (SecretsStore,
+                                                 parse_secret_ids)
+
+        store = SecretsStore(ss["file"], passphrase_env=env)
+        for ctype, doc in list(self._bundle.docs.items()):
+            self._bundle.docs[ctype] = parse_secret_ids(doc, store)
 
     def conf(self, ctype: ConfigType, required: bool = True):
         doc = self.bundle.get(ctype)

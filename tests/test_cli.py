@@ -150,3 +150,26 @@ def test_files_list_and_getall(configdir, tmp_path):
              "--taskid", "t1", "--dest", str(dest)], configdir, tmp_path)
     assert r.exit_code == 0
     assert (dest / "stdout.txt").exists()
+
+
+def test_secret_id_dereference_on_load(configdir, tmp_path, monkeypatch):
+    from shipyard_amd.config.secrets import SecretsStore
+
+    sec = tmp_path / "sec.bin"
+    SecretsStore(sec, passphrase="pw").set("reg-pw", "plain-secret")
+    creds = yaml.safe_load((configdir / "credentials.yaml").read_text())
+    creds["credentials"]["secrets_store"] = {"file": str(sec)}
+    creds["credentials"]["registries"] = {"docker": {
+        "r.example.com": {"username": "u",
+                          "password_secret_id": "reg-pw"}}}
+    (configdir / "credentials.yaml").write_text(yaml.safe_dump(creds))
+    monkeypatch.setenv("SHIPYARD_SECRETS_PASSPHRASE", "pw")
+    from shipyard_amd.cli import CliContext
+    from shipyard_amd.config import ConfigType
+
+    ctx = CliContext()
+    ctx.configdir = str(configdir)
+    doc = ctx.bundle.get(ConfigType.credentials)
+    reg = doc["credentials"]["registries"]["docker"]["r.example.com"]
+    assert reg["password"] == "plain-secret"
+    assert "password_secret_id" not in reg
