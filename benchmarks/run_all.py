@@ -1,0 +1,61 @@
+#!/usr/bin/env python3
+"""Run the full benchmark suite on one MI355X and emit a JSON-lines
+report (the round's consolidated evidence; copy into profiles/)."""
+from __future__ import annotations
+
+import json
+import subprocess
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
+
+REPO = Path(__file__).resolve().parents[1]
+
+
+def section(name):
+    print(json.dumps({"section": name}), flush=True)
+
+
+def main():
+    import torch
+
+    assert torch.cuda.is_available()
+    section("data_plane")
+    from benchmarks.data_plane_bench import bench_crc, bench_lz4, bench_sha
+
+    for fn in (bench_crc, bench_sha, bench_lz4):
+        r = fn()
+        r = {k: (round(v, 3) if isinstance(v, float) else v)
+             for k, v in r.items()}
+        print(json.dumps(r), flush=True)
+
+    section("lz4_realistic")
+    from benchmarks.lz4_realistic import bench_block_size, corpus
+
+    data = corpus()
+    for br in (4096, 8192):
+        print(json.dumps(bench_block_size(data, br, 256 << 20)),
+              flush=True)
+
+    section("stager")
+    subprocess.run([sys.executable,
+                    str(REPO / "benchmarks" / "stager_bench.py"), "1.0"])
+
+    section("cascade")
+    subprocess.run([sys.executable,
+                    str(REPO / "benchmarks" / "cascade_bench.py")])
+
+    section("allreduce_bench_py")
+    subprocess.run([sys.executable, str(REPO / "bench.py"),
+                    "--steps", "50", "--warmup", "10"])
+
+    section("rccl_binary")
+    subprocess.run([str(REPO / "shipyard_amd" / "comm" /
+                        "rccl_allreduce_bench"),
+                    "--min", "1048576", "--max", "268435456",
+                    "--iters", "20"])
+
+
+if __name__ == "__main__":
+    main()

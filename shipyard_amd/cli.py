@@ -235,6 +235,27 @@ def pool_images():
     """Global resource (image) ops on a pool."""
 
 
+@pool_images.command("list")
+@click.option("--poolid")
+@_common
+@pass_ctx
+def pool_images_list(ctx, poolid, configdir, root, raw):
+    """List images cached in the pool (reference `pool images list`)."""
+    _apply(ctx, configdir, root, raw)
+    pid = poolid or ctx.conf(ConfigType.pool)["pool_specification"]["id"]
+    cache = ctx.executor.pool_root(pid) / "images"
+    out = []
+    if cache.exists():
+        for d in sorted(cache.iterdir()):
+            if not d.is_dir() or d.name.startswith("."):
+                continue
+            size = sum(f.stat().st_size for f in d.rglob("*")
+                       if f.is_file())
+            out.append({"name": d.name, "bytes": size,
+                        "complete": (d / ".complete").exists()})
+    ctx.emit(out)
+
+
 @pool_images.command("update")
 @click.option("--poolid")
 @_common
