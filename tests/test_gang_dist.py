@@ -118,3 +118,33 @@ def test_rccl_gang_allreduce_1gpu(tmp_path):
            "stdout.txt").read_text()
     assert "rccl-1gpu ok" in out
     ex.store.close()
+
+
+@pytest.mark.gpu
+def test_gpu_slot_cycling(tmp_path):
+    """8 sequential GPU tasks through 1 slot: allocation/release cycling
+    on real hardware."""
+    import sys as _sys
+
+    import torch
+
+    assert torch.cuda.is_available()
+    ex = LocalExecutor(tmp_path / "root")
+    ex.pool_add({"pool_specification": {
+        "id": "cyc", "gpus": {"dedicated": 1}}})
+    script = tmp_path / "g.py"
+    script.write_text(
+        "import os, torch\n"
+        "assert os.environ['HIP_VISIBLE_DEVICES'] == '0'\n"
+        "x = torch.ones(1024, device='cuda')\n"
+        "assert float(x.sum().item()) == 1024.0\n"
+        "print('cycle ok')\n")
+    ex.jobs_add({"job_specifications": [{
+        "id": "cj",
+        "tasks": [{"id": f"t{i}", "command": f"{_sys.executable} {script}",
+                   "gpus": 1, "max_task_retries": 0} for i in range(8)],
+    }]}, "cyc")
+    ex.run_until_idle(timeout=600)
+    tasks = ex.tasks_list("cj")
+    assert all(t["state"] == "completed" for t in tasks), tasks
+    ex.store.close()
