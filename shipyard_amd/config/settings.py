@@ -232,6 +232,8 @@ class TaskSettings:
     multi_instance: Optional[MultiInstanceSettings]
     remove_container_after_exit: bool
     additional_docker_run_options: List[str]
+    additional_singularity_options: List[str]
+    singularity_cmd: str
     default_working_dir: str
     task_factory: Optional[dict]
     labels: List[str]
@@ -428,6 +430,10 @@ def task_settings(taskspec: Dict[str, Any], job: JobSettings,
             "remove_container_after_exit", True),
         additional_docker_run_options=list(
             taskspec.get("additional_docker_run_options") or []),
+        additional_singularity_options=list(
+            taskspec.get("additional_singularity_options") or []),
+        singularity_cmd=_get(taskspec, "singularity_execution", "cmd",
+                             default="exec"),
         default_working_dir=taskspec.get("default_working_dir",
                                          job.default_working_dir),
         task_factory=taskspec.get("task_factory"),

@@ -1138,6 +1138,8 @@ class LocalExecutor:
             shm_size=ts.shm_size,
             volumes=volumes,
             docker_options=ts.additional_docker_run_options,
+            singularity_options=ts.additional_singularity_options,
+            singularity_cmd=ts.singularity_cmd,
             remove_container=ts.remove_container_after_exit,
             num_instances=ranks if mi else 1,
             gang_backend=mi.gang.backend if mi else "rccl",

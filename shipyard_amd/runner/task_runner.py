@@ -70,6 +70,8 @@ class LaunchSpec:
     shm_size: Optional[int] = None
     volumes: List[str] = field(default_factory=list)
     docker_options: List[str] = field(default_factory=list)
+    singularity_options: List[str] = field(default_factory=list)
+    singularity_cmd: str = "exec"
     remove_container: bool = True
     # gang
     num_instances: int = 1
@@ -335,7 +337,9 @@ def _runtime_cmd(spec: LaunchSpec, command: str, paths: TaskPaths,
                 "singularity runtime requested but not installed")
         return rt.singularity_run_command(
             image=spec.image or "", command=command,
-            device_ids=spec.device_ids, volumes=spec.volumes)
+            device_ids=spec.device_ids, volumes=spec.volumes,
+            exec_cmd=spec.singularity_cmd,
+            extra_options=spec.singularity_options)
     raise ValueError(f"unknown runtime {spec.runtime}")
 
 

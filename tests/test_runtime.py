@@ -47,3 +47,12 @@ def test_gpu_env_hides_gpus_for_cpu_tasks():
     assert env["HIP_VISIBLE_DEVICES"] == "3,4"
     assert env["ROCR_VISIBLE_DEVICES"] == "3,4"
     assert env["HSA_ENABLE_IPC_MODE_LEGACY"] == "0"
+
+
+def test_singularity_exec_and_options_passthrough():
+    cmd = rt.singularity_run_command(
+        "img.sif", "python run.py", device_ids=[0], exec_cmd="run",
+        extra_options=["--contain"])
+    assert cmd[:2] == ["singularity", "run"]
+    assert "--contain" in cmd and "--rocm" in cmd
+    assert cmd[-3:] == ["img.sif", "python", "run.py"]
