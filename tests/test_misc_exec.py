@@ -192,11 +192,13 @@ def test_live_tail_streams_incrementally(tmp_path):
     ex.jobs_add({"job_specifications": [{
         "id": "lj", "tasks": [{
             "id": "t",
-            "command": "echo first; sleep 0.4; echo second"}]}]}, "lp")
+            "command": "echo first; sleep 1.5; echo second"}]}]}, "lp")
     seen = []
-    out = ex.stream_task_file("lj", "t", sink=seen.append, timeout=30)
+    out = ex.stream_task_file("lj", "t", sink=seen.append, timeout=60)
     assert "first" in out and "second" in out
-    # incremental: 'first' arrived in an earlier chunk than 'second'
-    assert len(seen) >= 2
-    assert "first" in seen[0]
+    # incremental: 'first' arrived in a chunk without 'second' (the
+    # 1.5s gap makes chunk-merging effectively impossible even under
+    # heavy host load)
+    assert len(seen) >= 2, seen
+    assert "second" not in seen[0]
     ex.store.close()
