@@ -224,23 +224,6 @@ def egress_from_object_store(store: ObjectStore, remote_path: str,
 # ---------------------------------------------------------------------
 # task-level input/output data (reference convoy/data.py:219/447)
 # ---------------------------------------------------------------------
-def process_input_data(store_roots: Dict[str, ObjectStore],
-                       specs, default_dest: Path) -> None:
-    """Materialize a task's input_data before launch."""
-    for ds in specs:
-        spec = ds.spec
-        if ds.kind == "local_storage":
-            store = store_roots[spec.get("storage_account_settings",
-                                         "default")]
-            dest = Path(utils.expand_env(spec.get("local_path")
-                                         or str(default_dest)))
-            egress_from_object_store(
-                store, spec["remote_path"], dest,
-                include=spec.get("include") or (),
-                exclude=spec.get("exclude") or (),
-                verify=spec.get("verify", True))
-
-
 def process_output_data(store_roots: Dict[str, ObjectStore],
                         specs, task_wd: Path,
                         task_succeeded: bool) -> None:
