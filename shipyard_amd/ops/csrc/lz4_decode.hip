@@ -57,7 +57,10 @@ enum : uint32_t {
 // s_waitcnt immediate: lgkmcnt(0) only (vmcnt/expcnt unconstrained)
 constexpr int kWaitLgkm0 = 0xC07F;
 
-template <int RAWCAP>
+// PROBE: 0 = real decode; 1 = skip literal copies; 2 = skip match
+// copies (both produce WRONG output — perf attribution only, selected
+// via SY_LZ4_SKIP for experiments)
+template <int RAWCAP, int PROBE = 0>
 __global__ __launch_bounds__(SY_WAVE) void lz4_decode_kernel(
     const uint8_t* __restrict__ comp, const uint64_t* __restrict__ in_off,
     const uint32_t* __restrict__ in_len, uint8_t* __restrict__ out,
@@ -127,8 +130,10 @@ __global__ __launch_bounds__(SY_WAVE) void lz4_decode_kernel(
         // (byte-granular; funnel-shift wide copies measured 15-20%
         // SLOWER overall — typical runs are 5-30 B and the head/word/
         // tail fragmentation costs more than 4 B/lane saves)
-        for (uint32_t i = lane; i < litlen; i += SY_WAVE) {
-          dst[dpos + i] = src[pos + i];
+        if (PROBE != 1) {
+          for (uint32_t i = lane; i < litlen; i += SY_WAVE) {
+            dst[dpos + i] = src[pos + i];
+          }
         }
         pos += litlen;
         dpos += litlen;
@@ -164,14 +169,16 @@ __global__ __launch_bounds__(SY_WAVE) void lz4_decode_kernel(
 
         // ---- cooperative match copy, doubling over overlap ----
         uint32_t done = 0;
-        while (done < mlen) {
-          const uint32_t dist = done + offset;  // multiple of offset
-          const uint32_t n = min(mlen - done, dist);
-          for (uint32_t i = lane; i < n; i += SY_WAVE) {
-            dst[dpos + done + i] = dst[dpos + done + i - dist];
+        if (PROBE != 2) {
+          while (done < mlen) {
+            const uint32_t dist = done + offset;  // multiple of offset
+            const uint32_t n = min(mlen - done, dist);
+            for (uint32_t i = lane; i < n; i += SY_WAVE) {
+              dst[dpos + done + i] = dst[dpos + done + i - dist];
+            }
+            done += n;
+            if (done < mlen) __builtin_amdgcn_s_waitcnt(kWaitLgkm0);
           }
-          done += n;
-          if (done < mlen) __builtin_amdgcn_s_waitcnt(kWaitLgkm0);
         }
         dpos += mlen;
       }
@@ -217,6 +224,24 @@ SY_EXPORT int sy_lz4_decode_blocks(const void* d_comp, const uint64_t* d_in_off,
   uint32_t grid = n_blocks < grid_cap ? n_blocks : grid_cap;
   const uint8_t* c = static_cast<const uint8_t*>(d_comp);
   uint8_t* o = static_cast<uint8_t*>(d_out);
+  // perf-attribution probes (WRONG OUTPUT; experiments only)
+  static int probe = -1;
+  if (probe < 0) {
+    const char* e = getenv("SY_LZ4_SKIP");
+    probe = e ? atoi(e) : 0;
+  }
+  if (probe == 1 && raw_cap <= 8 * 1024) {
+    hipLaunchKernelGGL((lz4_decode_kernel<8 * 1024, 1>), dim3(grid),
+                       dim3(SY_WAVE), 0, stream, c, d_in_off, d_in_len, o,
+                       d_out_off, d_out_len, d_status, n_blocks);
+    return sy_check(hipGetLastError());
+  }
+  if (probe == 2 && raw_cap <= 8 * 1024) {
+    hipLaunchKernelGGL((lz4_decode_kernel<8 * 1024, 2>), dim3(grid),
+                       dim3(SY_WAVE), 0, stream, c, d_in_off, d_in_len, o,
+                       d_out_off, d_out_len, d_status, n_blocks);
+    return sy_check(hipGetLastError());
+  }
   if (raw_cap <= 4 * 1024) {
     hipLaunchKernelGGL((lz4_decode_kernel<4 * 1024>), dim3(grid),
                        dim3(SY_WAVE), 0, stream, c, d_in_off, d_in_len, o,
