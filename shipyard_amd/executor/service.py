@@ -1090,6 +1090,30 @@ class LocalExecutor:
             env[f"SHIPYARD_VOLUME_{name.upper()}"] = host
         return binds
 
+    def _ensure_image(self, ps: cfg.PoolSettings, js: cfg.JobSettings,
+                      ts: cfg.TaskSettings, env: Dict[str, str]) -> None:
+        """On-demand image staging (the wait_for_images.sh +
+        delay_docker_image_preload path): a task naming a registered
+        local_image gets it staged into the pool cache before launch
+        (lease-arbitrated, cached thereafter); missing unregistered
+        images fail the launch unless allow_run_on_missing_image."""
+        if not ts.image:
+            return
+        gc = self.global_config()
+        if gc is None:
+            return
+        gs = cfg.global_settings(gc)
+        names = {im["name"] for im in gs.local_images}
+        if ts.image not in names:
+            return  # docker/singularity images: runtime pulls or fails
+        cache = self.pool_root(ps.id) / "images" / ts.image
+        if not (cache / ".complete").exists():
+            rep = self.replicator(
+                ps.id, concurrency=gs.concurrent_source_downloads,
+                account=gs.storage_account)
+            rep.stage_image(ts.image)
+        env["SHIPYARD_IMAGE_DIR"] = str(cache)
+
     def _launch_task(self, ps: cfg.PoolSettings, js: cfg.JobSettings,
                      ts: cfg.TaskSettings, jid: str, tid: str,
                      device_ids: List[int], ranks: int) -> TaskHandle:
@@ -1097,6 +1121,7 @@ class LocalExecutor:
         env = dict(ps.environment_variables)
         env.update(ts.environment_variables)
         volumes = self._resolve_volumes(ts, env)
+        self._ensure_image(ps, js, ts, env)
         if ts.input_data or js.input_data:
             self._process_input_data(ps, jid, tid,
                                      list(js.input_data) + list(ts.input_data))

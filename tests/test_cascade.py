@@ -93,3 +93,37 @@ def test_image_name_traversal_rejected(tmp_path):
     for bad in ("../evil", "/abs", "a/../../b", ""):
         with pytest.raises(ValueError):
             rep.stage_image(bad)
+
+
+def test_on_demand_image_staging(tmp_path):
+    """delay_image_preload + a task naming a local_image: the scheduler
+    stages it at first launch (wait_for_images analogue)."""
+    ex = LocalExecutor(tmp_path / "od", detect_gpus=False)
+    src = tmp_path / "src"
+    src.mkdir()
+    (src / "model.bin").write_bytes(b"weights " * 1000)
+    ex.replicator("pre").pack_image("lazy-img", src)
+    ex.pool_add(
+        {"pool_specification": {
+            "id": "odp", "gpus": {"dedicated": 0}, "cpu_slots": 1,
+            "node_configuration": {"rocm": {"verify": False}}}},
+        config_conf={
+            "batch_shipyard": {"storage_account_settings": "default",
+                               "delay_image_preload": True},
+            "global_resources": {
+                "local_images": [{"name": "lazy-img",
+                                  "source": "prepacked"}]},
+        })
+    # not staged at pool add
+    assert not (ex.pool_root("odp") / "images" / "lazy-img" /
+                ".complete").exists()
+    ex.jobs_add({"job_specifications": [{
+        "id": "odj",
+        "tasks": [{"id": "t", "image": "lazy-img",
+                   "command": "test -f $SHIPYARD_IMAGE_DIR/model.bin"}],
+    }]}, "odp")
+    ex.run_until_idle(timeout=60)
+    assert ex.tasks_list("odj")[0]["state"] == "completed"
+    assert (ex.pool_root("odp") / "images" / "lazy-img" /
+            ".complete").exists()
+    ex.store.close()
