@@ -720,6 +720,19 @@ def pool_nodes():
     """Slot remediation (the `pool nodes` analogue)."""
 
 
+@pool_nodes.command("list")
+@click.option("--poolid", required=True)
+@_common
+@pass_ctx
+def nodes_list(ctx, poolid, configdir, root, raw):
+    """Per-slot state/device (reference `pool nodes list`)."""
+    _apply(ctx, configdir, root, raw)
+    rows = ctx.executor.store.query(
+        "SELECT slot_id, kind, device_id, dedicated, state FROM slots "
+        "WHERE pool_id=? ORDER BY slot_id", (poolid,))
+    ctx.emit([dict(r) for r in rows])
+
+
 @pool_nodes.command("offline")
 @click.option("--poolid", required=True)
 @click.option("--slot", type=int, required=True)

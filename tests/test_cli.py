@@ -173,3 +173,29 @@ def test_secret_id_dereference_on_load(configdir, tmp_path, monkeypatch):
     reg = doc["credentials"]["registries"]["docker"]["r.example.com"]
     assert reg["password"] == "plain-secret"
     assert "password_secret_id" not in reg
+
+
+def test_keyvault_cli_roundtrip(configdir, tmp_path, monkeypatch):
+    creds = yaml.safe_load((configdir / "credentials.yaml").read_text())
+    creds["credentials"]["secrets_store"] = {
+        "file": str(tmp_path / "kv.bin")}
+    (configdir / "credentials.yaml").write_text(yaml.safe_dump(creds))
+    monkeypatch.setenv("SHIPYARD_SECRETS_PASSPHRASE", "pw")
+    r = run(["keyvault", "set", "--name", "tok", "--value", "s3cr3t"],
+            configdir, tmp_path)
+    assert r.exit_code == 0, r.output
+    r = run(["keyvault", "get", "--name", "tok"], configdir, tmp_path)
+    assert r.output.strip() == "s3cr3t"
+    r = run(["keyvault", "list"], configdir, tmp_path)
+    assert "tok" in r.output
+    r = run(["keyvault", "del", "--name", "tok"], configdir, tmp_path)
+    assert r.exit_code == 0
+
+
+def test_pool_nodes_list(configdir, tmp_path):
+    run(["pool", "add"], configdir, tmp_path)
+    r = run(["pool", "nodes", "list", "--poolid", "clipool"],
+            configdir, tmp_path)
+    assert r.exit_code == 0
+    rows = json.loads(r.output)
+    assert len(rows) == 2 and rows[0]["state"] == "idle"
