@@ -371,6 +371,9 @@ __global__ __launch_bounds__(2 * SY_WAVE) void lz4_decode_pc_kernel(
 
     if (wave == 0) {
       // ---------------- producer: parse into the ring ----------------
+      LdsWindow w;
+      w.base = sbuf;
+      w.win_base = 0xFFFFFFF0u;
       uint32_t pos = 0;
       uint32_t produced = 0;
       uint32_t st = SY_LZ4_OK;

@@ -44,7 +44,7 @@ struct PinnedRing {
 
   void release() {
     for (int i = 0; i < kRing; ++i) {
-      if (buf[i]) hipHostFree(buf[i]);
+      if (buf[i]) (void)hipHostFree(buf[i]);
       buf[i] = nullptr;
     }
     bytes = 0;
@@ -146,7 +146,7 @@ SY_EXPORT int sy_stage_file(const char* path, void* d_dst,
     *out_seconds = std::chrono::duration<double>(
         std::chrono::steady_clock::now() - t0).count();
   }
-  for (int i = 0; i < kRing; ++i) hipEventDestroy(ev[i]);
+  for (int i = 0; i < kRing; ++i) (void)hipEventDestroy(ev[i]);
   close(fd);
   return err;
 }
