@@ -58,28 +58,6 @@ enum : uint32_t {
 constexpr int kWaitLgkm0 = 0xC07F;
 
 
-// Parse-side byte fetch through a 16 B register window over the LDS
-// stream: the parse chain's dependent loads drop from ~4-5 ds_read_u8
-// per sequence (~60 cyc each) to ~1 ds_read_b128 per 16 header bytes;
-// extraction is a handful of VALU selects.  (Attribution probe: parse
-// was 58% of decode time with per-byte LDS reads.)
-struct LdsWindow {
-  const uint8_t* base;   // 16 B-aligned LDS array base
-  uint32_t win_base = 0xFFFFFFF0u;
-  uint4 win;
-
-  __device__ __forceinline__ uint8_t get(uint32_t pos) {
-    const uint32_t wb = pos & ~15u;
-    if (wb != win_base) {
-      win_base = wb;
-      win = *reinterpret_cast<const uint4*>(base + wb);
-    }
-    const uint32_t word = (pos & 8u)
-        ? ((pos & 4u) ? win.w : win.z)
-        : ((pos & 4u) ? win.y : win.x);
-    return (uint8_t)(word >> ((pos & 3u) * 8u));
-  }
-};
 
 // PROBE: 0 = real decode; 1 = skip literal copies; 2 = skip match
 // copies (both produce WRONG output — perf attribution only, selected
@@ -122,13 +100,12 @@ __global__ __launch_bounds__(SY_WAVE) void lz4_decode_kernel(
       for (uint32_t i = lane; i < n16; i += SY_WAVE) s4[i] = g4[i];
       // all lanes' vm loads -> lds writes must land before parsing
       __builtin_amdgcn_s_waitcnt(0);
-      // parse below uses src = sbuf + srcoff; header bytes go through
-      // a register window (parse chain), literal copies read sbuf
-      // directly (independent of the chain)
+      // parse below uses src = sbuf + srcoff.  NOTE (measured): a
+      // 16 B register-window fetch for header bytes is 45% SLOWER than
+      // these per-byte LDS reads — at 9+ waves/CU the read latency is
+      // TLP-hidden and the window's extraction VALU + refill join the
+      // serial chain instead.
       const uint8_t* src = sbuf + srcoff;
-      LdsWindow w;
-      w.base = sbuf;
-      w.win_base = 0xFFFFFFF0u;
 
       // Uniform parse state (identical in every lane; LDS byte reads
       // of the same address broadcast).
@@ -139,15 +116,13 @@ __global__ __launch_bounds__(SY_WAVE) void lz4_decode_kernel(
           if (dpos != rawlen) st = SY_LZ4_ERR_MISMATCH;
           break;
         }
-        const uint32_t token = w.get(srcoff + pos);
-        ++pos;
+        const uint32_t token = src[pos++];
         uint32_t litlen = token >> 4;
         if (litlen == 15) {
           uint8_t b;
           do {
             if (pos >= slen) { st = SY_LZ4_ERR_TRUNC; break; }
-            b = w.get(srcoff + pos);
-            ++pos;
+            b = src[pos++];
             litlen += b;
           } while (b == 255);
           if (st != SY_LZ4_OK) break;
@@ -175,16 +150,14 @@ __global__ __launch_bounds__(SY_WAVE) void lz4_decode_kernel(
 
         if (pos + 2 > slen) { st = SY_LZ4_ERR_TRUNC; break; }
         const uint32_t offset =
-            (uint32_t)w.get(srcoff + pos) |
-            ((uint32_t)w.get(srcoff + pos + 1) << 8);
+            (uint32_t)src[pos] | ((uint32_t)src[pos + 1] << 8);
         pos += 2;
         uint32_t mlen = (token & 0xFu) + 4;
         if ((token & 0xFu) == 15) {
           uint8_t b;
           do {
             if (pos >= slen) { st = SY_LZ4_ERR_TRUNC; break; }
-            b = w.get(srcoff + pos);
-            ++pos;
+            b = src[pos++];
             mlen += b;
           } while (b == 255);
           if (st != SY_LZ4_OK) break;
@@ -371,9 +344,6 @@ __global__ __launch_bounds__(2 * SY_WAVE) void lz4_decode_pc_kernel(
 
     if (wave == 0) {
       // ---------------- producer: parse into the ring ----------------
-      LdsWindow w;
-      w.base = sbuf;
-      w.win_base = 0xFFFFFFF0u;
       uint32_t pos = 0;
       uint32_t produced = 0;
       uint32_t st = SY_LZ4_OK;
@@ -383,15 +353,13 @@ __global__ __launch_bounds__(2 * SY_WAVE) void lz4_decode_pc_kernel(
           if (dtotal != rawlen) st = SY_LZ4_ERR_MISMATCH;
           break;
         }
-        const uint32_t token = w.get(srcoff + pos);
-        ++pos;
+        const uint32_t token = src[pos++];
         uint32_t litlen = token >> 4;
         if (litlen == 15) {
           uint8_t b;
           do {
             if (pos >= slen) { st = SY_LZ4_ERR_TRUNC; break; }
-            b = w.get(srcoff + pos);
-            ++pos;
+            b = src[pos++];
             litlen += b;
           } while (b == 255);
           if (st != SY_LZ4_OK) break;
