@@ -225,7 +225,7 @@ def pool_autoscale_evaluate(ctx, poolid, configdir, root, raw):
 
     pid = poolid or ctx.conf(ConfigType.pool)["pool_specification"]["id"]
     ctl = AutoscaleController(ctx.executor, pid,
-                              ctx.executor._pool_settings(pid).autoscale)
+                              ctx.executor.pool_settings_of(pid).autoscale)
     dec = ctl.maybe_evaluate(_time.time() + 10**9)
     ctx.emit(dec.__dict__ if dec else {"autoscale": "disabled"})
 
@@ -476,7 +476,7 @@ def data_files():
 def files_list(ctx, jobid, taskid, configdir, root, raw):
     """List a task's files (reference `data files list`)."""
     _apply(ctx, configdir, root, raw)
-    pool_id = ctx.executor._job_pool(jobid)
+    pool_id = ctx.executor.job_pool(jobid)
     base = (ctx.executor.pool_root(pool_id) / "jobs" / jobid / "tasks" /
             taskid)
     out = []
@@ -499,7 +499,7 @@ def files_getall(ctx, jobid, taskid, dest, configdir, root, raw):
     _apply(ctx, configdir, root, raw)
     from shipyard_amd.data import mover as _mover
 
-    pool_id = ctx.executor._job_pool(jobid)
+    pool_id = ctx.executor.job_pool(jobid)
     base = (ctx.executor.pool_root(pool_id) / "jobs" / jobid / "tasks" /
             taskid)
     res = _mover.ingress_directory(base, dest)
@@ -930,7 +930,7 @@ def misc_tensorboard(ctx, jobid, taskid, logdir, port, configdir, root,
     import shutil as _shutil
     import subprocess as _sp
 
-    pool_id = ctx.executor._job_pool(jobid)
+    pool_id = ctx.executor.job_pool(jobid)
     path = (ctx.executor.pool_root(pool_id) / "jobs" / jobid / "tasks" /
             taskid / logdir)
     if _shutil.which("tensorboard") is None:

@@ -304,6 +304,13 @@ class LocalExecutor:
     def pool_root(self, pool_id: str) -> Path:
         return self.root / "pools" / pool_id
 
+    # public accessors (CLI/monitoring use)
+    def job_pool(self, job_id: str) -> str:
+        return self._job_pool(job_id)
+
+    def pool_settings_of(self, pool_id: str) -> cfg.PoolSettings:
+        return self._pool_settings(pool_id)
+
     # ----------------------------------------------------------------
     # jobs (reference convoy/batch.py:5056 add_jobs)
     # ----------------------------------------------------------------
