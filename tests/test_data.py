@@ -170,3 +170,28 @@ class TestExecutorDataFlow:
         states = {t["id"]: t["state"] for t in ex.tasks_list("jchain")}
         assert states == {"producer": "completed", "consumer": "completed"}
         ex.store.close()
+
+
+class TestObjectStoreOps:
+    def test_delete_and_exists(self, tmp_path):
+        st = ObjectStore(tmp_path / "s")
+        st.upload_bytes("c/x.bin", b"1")
+        st.upload_bytes("c/y.bin", b"2")
+        assert st.exists("c/x.bin")
+        assert st.delete("c/x.bin")
+        assert not st.exists("c/x.bin")
+        assert not st.delete("c/x.bin")  # already gone
+        assert st.delete("c")            # directory delete
+        assert list(st.list("")) == []
+
+    def test_manifest_files_hidden_from_list(self, tmp_path):
+        st = ObjectStore(tmp_path / "s")
+        st.upload_bytes("c/a.bin", b"x", manifest=True)
+        assert list(st.list("c")) == ["c/a.bin"]
+
+    def test_download_missing_raises(self, tmp_path):
+        st = ObjectStore(tmp_path / "s")
+        import pytest as _pytest
+
+        with _pytest.raises(FileNotFoundError):
+            st.download_bytes("nope.bin")
