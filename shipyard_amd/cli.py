@@ -639,13 +639,17 @@ def monitor_scrape(ctx, configdir, root, raw):
 
 @monitor.command("start")
 @click.option("--port", type=int, default=9400)
+@click.option("--tls-cert", default=None)
+@click.option("--tls-key", default=None)
 @_common
 @pass_ctx
-def monitor_start(ctx, port, configdir, root, raw):  # pragma: no cover
+def monitor_start(ctx, port, tls_cert, tls_key, configdir, root,
+                  raw):  # pragma: no cover
     _apply(ctx, configdir, root, raw)
     from shipyard_amd.monitor.exporter import Exporter
 
-    Exporter(store=ctx.executor.store, port=port).serve_forever()
+    Exporter(store=ctx.executor.store, port=port, tls_cert=tls_cert,
+             tls_key=tls_key).serve_forever()
 
 
 @monitor.command("targets")
@@ -908,6 +912,71 @@ def kv_list(ctx, configdir, root, raw):
 def kv_del(ctx, name, configdir, root, raw):
     _apply(ctx, configdir, root, raw)
     ctx.emit({"deleted": _secrets(ctx).delete(name)})
+
+
+# ---------------------------------------------------------------- cert
+@cli.group()
+def cert():
+    """TLS certificates (reference `cert` group, convoy/crypto.py:445;
+    local targets: monitoring-exporter TLS, PFX export)."""
+
+
+def _cert_dir(ctx) -> Path:
+    return Path(ctx.executor.root) / "certs"
+
+
+@cert.command("create")
+@click.option("--cn", default="shipyard-amd")
+@click.option("--days", type=int, default=365)
+@click.option("--prefix", default="shipyard_cert")
+@click.option("--pfx-password", default=None,
+              help="also emit a PFX bundle with this password")
+@_common
+@pass_ctx
+def cert_create(ctx, cn, days, prefix, pfx_password, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.utils import crypto
+
+    key, crt = crypto.generate_self_signed_cert(
+        _cert_dir(ctx), cn=cn, days=days, prefix=prefix)
+    out = {"key": str(key), "cert": str(crt),
+           "sha256": crypto.cert_fingerprint(crt)}
+    if pfx_password is not None:
+        out["pfx"] = str(crypto.export_pfx(
+            key, crt, _cert_dir(ctx) / f"{prefix}.pfx", pfx_password))
+    ctx.emit(out)
+
+
+@cert.command("list")
+@_common
+@pass_ctx
+def cert_list(ctx, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.utils import crypto
+
+    d = _cert_dir(ctx)
+    rows = []
+    if d.is_dir():
+        for p in sorted(d.glob("*.crt")):
+            rows.append({"cert": str(p),
+                         "sha256": crypto.cert_fingerprint(p)})
+    ctx.emit(rows)
+
+
+@cert.command("del")
+@click.option("--prefix", required=True)
+@_common
+@pass_ctx
+def cert_del(ctx, prefix, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    d = _cert_dir(ctx)
+    gone = []
+    for suf in (".key", ".crt", ".pfx"):
+        p = d / f"{prefix}{suf}"
+        if p.exists():
+            p.unlink()
+            gone.append(str(p))
+    ctx.emit({"deleted": gone})
 
 
 # ---------------------------------------------------------------- misc

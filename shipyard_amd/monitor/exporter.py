@@ -142,10 +142,13 @@ class Exporter:
     binding a port via scrape())."""
 
     def __init__(self, store=None, port: int = 9400,
-                 interval_s: float = 1.0):
+                 interval_s: float = 1.0, tls_cert: Optional[str] = None,
+                 tls_key: Optional[str] = None):
         self.store = store
         self.port = port
         self.interval_s = interval_s
+        self.tls_cert = tls_cert
+        self.tls_key = tls_key
         self._gauges = {}
         self._registry = None
 
@@ -174,7 +177,9 @@ class Exporter:
                 self._g_ex.labels(name=k).set(v)
         return generate_latest(self._registry)
 
-    def serve_forever(self):  # pragma: no cover - long-running daemon
+    def make_server(self, port: Optional[int] = None):
+        """Build the (optionally TLS-wrapped) HTTP server without
+        blocking; tests bind port 0 and scrape over the wire."""
         import http.server
 
         exporter = self
@@ -191,6 +196,18 @@ class Exporter:
             def log_message(self, *a):
                 pass
 
-        srv = http.server.HTTPServer(("0.0.0.0", self.port), Handler)
-        logger.info("exporter serving on :%d", self.port)
+        srv = http.server.HTTPServer(
+            ("0.0.0.0", self.port if port is None else port), Handler)
+        if self.tls_cert and self.tls_key:
+            import ssl
+
+            sctx = ssl.SSLContext(ssl.PROTOCOL_TLS_SERVER)
+            sctx.load_cert_chain(self.tls_cert, self.tls_key)
+            srv.socket = sctx.wrap_socket(srv.socket, server_side=True)
+        return srv
+
+    def serve_forever(self):  # pragma: no cover - long-running daemon
+        srv = self.make_server()
+        logger.info("exporter serving on :%d%s", srv.server_address[1],
+                    " (tls)" if self.tls_cert else "")
         srv.serve_forever()

@@ -40,6 +40,55 @@ def generate_ssh_keypair(export_path, prefix: str = DEFAULT_KEY_NAME,
     return priv, pub
 
 
+def generate_self_signed_cert(export_path, cn: str = "shipyard-amd",
+                              days: int = 365,
+                              prefix: str = "shipyard_cert",
+                              ) -> Tuple[Path, Path]:
+    """Generate a PEM key + self-signed cert via openssl (reference
+    crypto.py:445 `generate_pem_pfx_certificates`; here the target is
+    local TLS — e.g. the monitoring exporter — not the Azure cert
+    store).  Returns (key_path, cert_path)."""
+    if shutil.which("openssl") is None:
+        raise RuntimeError("openssl not installed")
+    out = Path(export_path)
+    out.mkdir(parents=True, exist_ok=True)
+    key = out / f"{prefix}.key"
+    cert = out / f"{prefix}.crt"
+    subprocess.run(
+        ["openssl", "req", "-x509", "-newkey", "rsa:3072", "-nodes",
+         "-keyout", str(key), "-out", str(cert), "-days", str(days),
+         "-subj", f"/CN={cn}",
+         "-addext", "subjectAltName=DNS:localhost,IP:127.0.0.1"],
+        check=True, capture_output=True)
+    key.chmod(0o600)
+    return key, cert
+
+
+def export_pfx(key_path, cert_path, out_path, password: str = "") -> Path:
+    """Bundle key+cert into a PKCS#12/PFX file (reference crypto.py:445
+    emits PFX for the Azure certificate store)."""
+    if shutil.which("openssl") is None:
+        raise RuntimeError("openssl not installed")
+    out = Path(out_path)
+    subprocess.run(
+        ["openssl", "pkcs12", "-export", "-out", str(out),
+         "-inkey", str(key_path), "-in", str(cert_path),
+         "-passout", f"pass:{password}"],
+        check=True, capture_output=True)
+    out.chmod(0o600)
+    return out
+
+
+def cert_fingerprint(cert_path) -> str:
+    """SHA-256 fingerprint of a PEM cert (reference prints the SHA-1
+    thumbprint used as the Batch cert id; SHA-256 here)."""
+    res = subprocess.run(
+        ["openssl", "x509", "-in", str(cert_path), "-noout",
+         "-fingerprint", "-sha256"],
+        check=True, capture_output=True, text=True)
+    return res.stdout.strip().split("=", 1)[1].replace(":", "").lower()
+
+
 def ssh_command(host: str, command: str, username: Optional[str] = None,
                 private_key: Optional[str] = None,
                 extra_options: Optional[List[str]] = None) -> List[str]:

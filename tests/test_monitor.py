@@ -119,3 +119,59 @@ def test_perf_collectors_in_scrape(tmp_path):
     m = collect_perf_metrics(ex.store)
     assert m["mover_transfers"] == 1 and m["mover_bytes"] == 1000
     ex.store.close()
+
+
+class TestCertsAndTLS:
+    def test_self_signed_cert_and_fingerprint(self, tmp_path):
+        from shipyard_amd.utils import crypto
+
+        key, crt = crypto.generate_self_signed_cert(tmp_path, cn="unit-test")
+        assert key.exists() and crt.exists()
+        fp = crypto.cert_fingerprint(crt)
+        assert len(fp) == 64 and all(c in "0123456789abcdef" for c in fp)
+        pfx = crypto.export_pfx(key, crt, tmp_path / "b.pfx", "pw")
+        assert pfx.stat().st_size > 0
+
+    def test_exporter_tls_scrape(self, tmp_path):
+        """End-to-end: exporter behind TLS, scraped over https."""
+        import ssl
+        import threading
+        import urllib.request
+
+        from shipyard_amd.monitor.exporter import Exporter
+        from shipyard_amd.utils import crypto
+
+        key, crt = crypto.generate_self_signed_cert(tmp_path)
+        ex = Exporter(store=None, tls_cert=str(crt), tls_key=str(key))
+        srv = ex.make_server(port=0)
+        port = srv.server_address[1]
+        t = threading.Thread(target=srv.serve_forever, daemon=True)
+        t.start()
+        try:
+            cctx = ssl.create_default_context(cafile=str(crt))
+            cctx.check_hostname = False
+            body = urllib.request.urlopen(
+                f"https://127.0.0.1:{port}/metrics", context=cctx,
+                timeout=10).read()
+            assert b"shipyard" in body or body == b"" or b"#" in body
+        finally:
+            srv.shutdown()
+            srv.server_close()
+
+    def test_cli_cert_lifecycle(self, tmp_path):
+        from click.testing import CliRunner
+
+        from shipyard_amd.cli import cli
+
+        r = CliRunner()
+        env_root = ["--root", str(tmp_path / "root")]
+        res = r.invoke(cli, ["cert", "create", "--cn", "cli-test",
+                             "--pfx-password", "pw", *env_root])
+        assert res.exit_code == 0, res.output
+        res = r.invoke(cli, ["cert", "list", *env_root])
+        assert res.exit_code == 0 and "sha256" in res.output
+        res = r.invoke(cli, ["cert", "del", "--prefix", "shipyard_cert",
+                             *env_root])
+        assert res.exit_code == 0 and "deleted" in res.output
+        res = r.invoke(cli, ["cert", "list", *env_root])
+        assert res.exit_code == 0 and "sha256" not in res.output
