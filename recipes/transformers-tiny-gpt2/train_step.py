@@ -1,0 +1,31 @@
+"""Random-init tiny GPT-2 training steps (no network: config-built
+model, random token ids).  Exemplar for the reference's ML recipe
+family (TensorFlow-GPU/Keras+Theano-GPU/etc -> PyTorch on MI355X)."""
+import time
+
+import torch
+from transformers import GPT2Config, GPT2LMHeadModel
+
+cfg = GPT2Config(n_layer=4, n_head=8, n_embd=512, n_positions=512,
+                 vocab_size=32000)
+model = GPT2LMHeadModel(cfg).to("cuda", dtype=torch.bfloat16)
+opt = torch.optim.AdamW(model.parameters(), lr=1e-4)
+B, S = 8, 512
+ids = torch.randint(0, cfg.vocab_size, (B, S), device="cuda")
+for _ in range(3):  # warmup
+    loss = model(input_ids=ids, labels=ids).loss
+    loss.backward()
+    opt.step()
+    opt.zero_grad(set_to_none=True)
+torch.cuda.synchronize()
+t0 = time.perf_counter()
+STEPS = 10
+for _ in range(STEPS):
+    loss = model(input_ids=ids, labels=ids).loss
+    loss.backward()
+    opt.step()
+    opt.zero_grad(set_to_none=True)
+torch.cuda.synchronize()
+dt = time.perf_counter() - t0
+print(f"tiny-gpt2 {B * S * STEPS / dt:.0f} tokens/s "
+      f"loss={float(loss):.3f}")
