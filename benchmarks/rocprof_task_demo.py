@@ -54,7 +54,7 @@ if stats:
     for r in rows[:10]:
         name = (r.get("Name") or r.get("Kernel_Name") or "?")[:80]
         print(f"  {name}  {r.get(key, '?')}")
-    out = Path("gpurun_out")
+    out = repo / "gpurun_out"
     out.mkdir(exist_ok=True)
     import shutil
     shutil.copy(stats[-1], out / "task_rocprof_kernel_stats.csv")
