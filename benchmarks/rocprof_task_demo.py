@@ -42,6 +42,12 @@ if not stats:
     stats = sorted(glob.glob(str(prof_dir / "**" / "*.csv"),
                              recursive=True))
 print("stats files:", [Path(s).name for s in stats])
+if not stats:
+    print("prof dir listing:")
+    for f in sorted(prof_dir.rglob("*")):
+        print("  ", f.relative_to(prof_dir))
+    err = ex.task_file("p", "prof-demo", "gpt2", "stderr.txt")
+    print("task stderr tail:", err.read_text()[-500:])
 if stats:
     with open(stats[-1]) as f:
         rows = list(csv.DictReader(f))

@@ -1151,8 +1151,10 @@ class LocalExecutor:
                 prof_dir = (self.pool_root(ps.id) / "jobs" / jid / "tasks" /
                             tid / "prof")
                 prof_dir.mkdir(parents=True, exist_ok=True)
-                opts = list(ts.rocprof_options) or ["--kernel-trace",
-                                                    "--stats"]
+                # rocprofv3 defaults to rocpd/sqlite output; ask for
+                # csv so the stats land as grep-able files per task
+                opts = list(ts.rocprof_options) or [
+                    "--kernel-trace", "--stats", "--output-format", "csv"]
                 wrapper = ["rocprofv3", *opts, "-d", str(prof_dir), "--"]
                 env.setdefault("TMPDIR", "/tmp")
             else:
