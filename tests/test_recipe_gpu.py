@@ -77,3 +77,40 @@ def test_gemm_burn_recipe(tmp_path):
     assert "TFLOP/s" in r.output, r.output
     tflops = float(r.output.split("bf16 GEMM")[1].split("TFLOP/s")[0])
     assert tflops > 100, r.output
+
+
+def test_rocprof_task_wrapper(tmp_path):
+    """jobs.yaml `rocprof` feature: the executor wraps the task in
+    rocprofv3 and per-task kernel stats appear under the task dir."""
+    import shutil
+
+    if shutil.which("rocprofv3") is None:
+        pytest.skip("rocprofv3 not installed")
+    import os
+
+    from shipyard_amd.executor import LocalExecutor
+
+    os.environ.setdefault("TMPDIR", "/tmp")
+    ex = LocalExecutor(tmp_path / "root", detect_gpus=True)
+    ex.pool_add({"pool_specification": {
+        "id": "p", "gpus": {"dedicated": 1},
+        "node_configuration": {"rocm": {"verify": False}}}})
+    ex.jobs_add({"job_specifications": [{
+        "id": "jprof",
+        "tasks": [{
+            "id": "t", "gpus": 1, "rocprof": {"enabled": True},
+            "command": "python3 -c \"import torch; "
+                       "a = torch.randn(512, 512, device='cuda'); "
+                       "print(float((a @ a).sum()))\"",
+        }],
+    }]}, "p")
+    ex.run_until_idle(timeout=300)
+    t = ex.tasks_list("jprof")[0]
+    assert t["state"] == "completed", t
+    prof = (tmp_path / "root" / "pools" / "p" / "jobs" / "jprof" /
+            "tasks" / "t" / "prof")
+    stats = list(prof.rglob("*kernel_stats.csv"))
+    assert stats, list(prof.rglob("*"))
+    assert "Cijk" in stats[0].read_text() or "kernel" in \
+        stats[0].read_text().lower()
+    ex.store.close()
