@@ -1,0 +1,139 @@
+"""Node agent — the per-node execution daemon of multi-node pools.
+
+Analogue of the reference's node-side stack (reference
+scripts/shipyard_nodeprep.sh start-task bootstrap + the Batch agent
+that runs tasks on each VM): the coordinator (LocalExecutor) compiles
+every launch into a LaunchSpec and enqueues it on the shared store's
+``assignments`` table, tagged with a node id; the agent on that node
+claims its assignments, launches them with the exact same runner the
+coordinator uses locally, heartbeats, and reports exit codes back
+through the store.  Cross-node gangs arrive as per-node rank windows
+(LaunchSpec.rank_start/world_size/master_addr) whose rendezvous is
+torch.distributed over RCCL (or gloo on CPU nodes).
+
+Deployment model: coordinator and nodes share the storage root (the
+pool root and store.db are on a shared filesystem, the reference's
+storage-account analogue); each node runs
+
+    python -m shipyard_amd.agent --root ROOT --pool POOL --node NODE
+"""
+from __future__ import annotations
+
+import os
+import time
+from pathlib import Path
+from typing import Dict, Optional
+
+from shipyard_amd import utils
+from shipyard_amd.executor.store import Store
+from shipyard_amd.runner.task_runner import (TaskHandle, launch,
+                                             spec_from_json)
+
+logger = utils.get_logger(__name__)
+
+
+class NodeAgent:
+    def __init__(self, root, pool_id: str, node_id: str) -> None:
+        self.root = Path(root)
+        self.pool_id = pool_id
+        self.node_id = node_id
+        self.store = Store(self.root / "store.db")
+        self._handles: Dict[int, TaskHandle] = {}
+        self._stop = False
+
+    # -- lifecycle ----------------------------------------------------
+    def heartbeat(self) -> None:
+        state = "running" if self._handles else "idle"
+        self.store.execute(
+            "UPDATE nodes SET heartbeat=?, state=?, agent_pid=? "
+            "WHERE pool_id=? AND node_id=?",
+            (time.time(), state, os.getpid(), self.pool_id, self.node_id))
+
+    def pool_root(self) -> Path:
+        return self.root / "pools" / self.pool_id
+
+    # -- assignment processing ---------------------------------------
+    def claim(self) -> int:
+        """Atomically claim queued assignments for this node and launch
+        them."""
+        claimed = []
+        with self.store.transaction() as conn:
+            rows = list(conn.execute(
+                "SELECT id, spec_json FROM assignments WHERE pool_id=? "
+                "AND node_id=? AND state='queued' ORDER BY id",
+                (self.pool_id, self.node_id)))
+            for r in rows:
+                conn.execute(
+                    "UPDATE assignments SET state='running', updated_at=? "
+                    "WHERE id=?", (time.time(), r["id"]))
+                claimed.append((r["id"], r["spec_json"]))
+        for aid, spec_json in claimed:
+            try:
+                spec = spec_from_json(spec_json)
+                self._handles[aid] = launch(spec, self.pool_root())
+                logger.info("node %s launched assignment %d (%s/%s)",
+                            self.node_id, aid, spec.job_id, spec.task_id)
+            except Exception as exc:
+                logger.error("launch of assignment %d failed: %s", aid, exc)
+                self._finish(aid, -1)
+        return len(claimed)
+
+    def poll(self) -> int:
+        """Poll running handles; honor cancellation; report exits."""
+        n = 0
+        if self._handles:
+            qmarks = ",".join("?" * len(self._handles))
+            cancelling = {r["id"] for r in self.store.query(
+                f"SELECT id FROM assignments WHERE id IN ({qmarks}) "
+                "AND state='cancelling'", list(self._handles))}
+        else:
+            cancelling = set()
+        for aid, h in list(self._handles.items()):
+            if aid in cancelling:
+                h.kill()
+            rc = h.poll()
+            if rc is None:
+                continue
+            del self._handles[aid]
+            self._finish(aid, rc)
+            n += 1
+        return n
+
+    def _finish(self, aid: int, rc: int) -> None:
+        self.store.execute(
+            "UPDATE assignments SET state='done', rc=?, updated_at=? "
+            "WHERE id=?", (rc, time.time(), aid))
+
+    def run_once(self) -> int:
+        self.heartbeat()
+        return self.claim() + self.poll()
+
+    def serve(self, poll_s: float = 0.05,
+              idle_exit_s: Optional[float] = None) -> None:
+        """Main loop.  idle_exit_s: exit after that long with no work
+        (None = run until stopped)."""
+        last_work = time.monotonic()
+        while not self._stop:
+            if self.run_once():
+                last_work = time.monotonic()
+            if (idle_exit_s is not None and not self._handles
+                    and time.monotonic() - last_work > idle_exit_s):
+                break
+            time.sleep(poll_s)
+        self.shutdown()
+
+    def stop(self) -> None:
+        self._stop = True
+
+    def shutdown(self) -> None:
+        """Tear down any still-running work (reported as rc -9 so the
+        coordinator's retry policy decides) and go offline."""
+        for aid, h in list(self._handles.items()):
+            h.kill()
+            rc = h.poll()
+            self._finish(aid, rc if rc is not None else -9)
+        self._handles.clear()
+        self.store.execute(
+            "UPDATE nodes SET state='offline', agent_pid=NULL "
+            "WHERE pool_id=? AND node_id=?", (self.pool_id, self.node_id))
+        self.store.close()

@@ -1,0 +1,26 @@
+"""Entry point: `python -m shipyard_amd.agent --root R --pool P --node N`."""
+import argparse
+import signal
+
+from shipyard_amd.agent import NodeAgent
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser(description="shipyard-amd node agent")
+    ap.add_argument("--root", required=True,
+                    help="shared storage root (contains state.db)")
+    ap.add_argument("--pool", required=True)
+    ap.add_argument("--node", required=True)
+    ap.add_argument("--poll", type=float, default=0.05)
+    ap.add_argument("--idle-exit", type=float, default=None,
+                    help="exit after this many idle seconds")
+    args = ap.parse_args()
+
+    agent = NodeAgent(args.root, args.pool, args.node)
+    signal.signal(signal.SIGTERM, lambda *a: agent.stop())
+    signal.signal(signal.SIGINT, lambda *a: agent.stop())
+    agent.serve(poll_s=args.poll, idle_exit_s=args.idle_exit)
+
+
+if __name__ == "__main__":
+    main()

@@ -59,6 +59,20 @@ class AutoscaleSettings:
 
 
 @dataclasses.dataclass(frozen=True)
+class NodeSettings:
+    """One node of a multi-node pool (the reference's pools span VMs;
+    here a node = a host reachable over a shared filesystem whose agent
+    runs `python -m shipyard_amd.agent`)."""
+    id: str
+    host: str = "127.0.0.1"
+    gpus_dedicated: int = 0
+    cpu_slots: int = 0
+    device_ids: Optional[List[int]] = None
+    ssh_user: Optional[str] = None
+    ssh_private_key: Optional[str] = None
+
+
+@dataclasses.dataclass(frozen=True)
 class PoolSettings:
     id: str
     gpus_dedicated: int
@@ -84,11 +98,22 @@ class PoolSettings:
     prometheus_rocm_exporter: bool
     prometheus_rocm_port: int
     prometheus_rocm_interval: float
+    nodes: Tuple["NodeSettings", ...] = ()
 
 
 def pool_settings(conf: Dict[str, Any]) -> PoolSettings:
     """reference convoy/settings.py:1277 `pool_settings`"""
     p = conf["pool_specification"]
+    nodes = tuple(
+        NodeSettings(
+            id=n["id"],
+            host=n.get("host", "127.0.0.1"),
+            gpus_dedicated=_get(n, "gpus", "dedicated", default=0),
+            cpu_slots=n.get("cpu_slots", 0),
+            device_ids=_get(n, "gpus", "device_ids"),
+            ssh_user=_get(n, "ssh", "username"),
+            ssh_private_key=_get(n, "ssh", "private_key"),
+        ) for n in (p.get("nodes") or []))
     scen = _get(p, "autoscale", "scenario")
     scenario = None
     if scen:
@@ -135,11 +160,15 @@ def pool_settings(conf: Dict[str, Any]) -> PoolSettings:
         runtimes = ["process"] + list(runtimes)
     return PoolSettings(
         id=p["id"],
-        gpus_dedicated=_get(p, "gpus", "dedicated", default=0),
-        gpus_low_priority=_get(p, "gpus", "low_priority", default=0),
+        # multi-node pools derive totals from the node list
+        gpus_dedicated=(sum(n.gpus_dedicated for n in nodes) if nodes
+                        else _get(p, "gpus", "dedicated", default=0)),
+        gpus_low_priority=(0 if nodes
+                           else _get(p, "gpus", "low_priority", default=0)),
         device_ids=_get(p, "gpus", "device_ids"),
         max_tasks_per_gpu=p.get("max_tasks_per_gpu", 1),
-        cpu_slots=p.get("cpu_slots", 0),
+        cpu_slots=(sum(n.cpu_slots for n in nodes) if nodes
+                   else p.get("cpu_slots", 0)),
         node_fill_type=p.get("node_fill_type", "pack"),
         resize_timeout=utils.parse_timedelta(
             p.get("resize_timeout", "00:20:00")),
@@ -172,6 +201,7 @@ def pool_settings(conf: Dict[str, Any]) -> PoolSettings:
                                   default=9400),
         prometheus_rocm_interval=_get(p, "prometheus", "rocm_exporter",
                                       "interval_seconds", default=1.0),
+        nodes=nodes,
     )
 
 

@@ -28,7 +28,9 @@ from shipyard_amd.data import mover
 from shipyard_amd.data.storage import ObjectStore
 from shipyard_amd.executor import task_factory
 from shipyard_amd.executor.store import Store
-from shipyard_amd.runner.task_runner import LaunchSpec, TaskHandle, launch
+from shipyard_amd.runner.task_runner import (LaunchSpec, TaskHandle,
+                                             _free_port, launch,
+                                             spec_to_json)
 
 logger = utils.get_logger(__name__)
 
@@ -40,6 +42,12 @@ class ExecutorError(RuntimeError):
     pass
 
 
+class _RemoteHandle:
+    """Stand-in handle for tasks executed by node agents (exit codes
+    arrive through the assignments table, not a local process)."""
+    timed_out = False
+
+
 class LocalExecutor:
     def __init__(self, root, detect_gpus: bool = True,
                  credentials_conf: Optional[Dict[str, Any]] = None):
@@ -47,6 +55,7 @@ class LocalExecutor:
         self.root.mkdir(parents=True, exist_ok=True)
         self.store = Store(self.root / "store.db")
         self._handles: Dict[Tuple[str, str], TaskHandle] = {}
+        self._agents: Dict[str, List[Any]] = {}  # pool -> agent Popens
         self._detect_gpus = detect_gpus
         self._n_host_gpus: Optional[int] = None
         # object stores (Azure Storage analogue); "default" always exists
@@ -94,8 +103,11 @@ class LocalExecutor:
         if len(device_ids) < total_gpus:
             raise ExecutorError("device_ids shorter than gpu slot count")
         # oversubscription guard (the analogue of the reference's
-        # vm_size capability checks, settings.py:4231-4289)
-        if host and total_gpus > host * ps.max_tasks_per_gpu:
+        # vm_size capability checks, settings.py:4231-4289).  Multi-node
+        # pools place slots on OTHER hosts, so the local-host bound only
+        # applies to single-node pools.
+        if (not ps.nodes and host
+                and total_gpus > host * ps.max_tasks_per_gpu):
             raise ExecutorError(
                 f"pool {ps.id} wants {total_gpus} gpu slots but host has "
                 f"{host} GPUs x {ps.max_tasks_per_gpu} tasks/gpu")
@@ -142,22 +154,45 @@ class LocalExecutor:
                       device_ids: Sequence[int]) -> None:
         rows = []
         slot = 0
-        # one slot per (gpu, task-slot) pair; dedicated slots first
-        for i in range(ps.gpus_dedicated):
-            for rep in range(ps.max_tasks_per_gpu):
-                rows.append((ps.id, slot, "gpu", device_ids[i], 1, "idle"))
+        if ps.nodes:
+            # multi-node pool: slots are (node x device); agents on each
+            # node execute (reference: pools span VMs)
+            for node in ps.nodes:
+                devs = node.device_ids
+                if devs is None:
+                    devs = list(range(node.gpus_dedicated))
+                for i in range(node.gpus_dedicated):
+                    for _ in range(ps.max_tasks_per_gpu):
+                        rows.append((ps.id, slot, "gpu", devs[i], 1,
+                                     "idle", node.id))
+                        slot += 1
+                for _ in range(node.cpu_slots):
+                    rows.append((ps.id, slot, "cpu", None, 1, "idle",
+                                 node.id))
+                    slot += 1
+            self.store.executemany(
+                "INSERT INTO nodes (pool_id, node_id, host) "
+                "VALUES (?,?,?)",
+                [(ps.id, n.id, n.host) for n in ps.nodes])
+        else:
+            # one slot per (gpu, task-slot) pair; dedicated slots first
+            for i in range(ps.gpus_dedicated):
+                for rep in range(ps.max_tasks_per_gpu):
+                    rows.append((ps.id, slot, "gpu", device_ids[i], 1,
+                                 "idle", "local"))
+                    slot += 1
+            for i in range(ps.gpus_dedicated,
+                           ps.gpus_dedicated + ps.gpus_low_priority):
+                for rep in range(ps.max_tasks_per_gpu):
+                    rows.append((ps.id, slot, "gpu", device_ids[i], 0,
+                                 "idle", "local"))
+                    slot += 1
+            for _ in range(ps.cpu_slots):
+                rows.append((ps.id, slot, "cpu", None, 1, "idle", "local"))
                 slot += 1
-        for i in range(ps.gpus_dedicated,
-                       ps.gpus_dedicated + ps.gpus_low_priority):
-            for rep in range(ps.max_tasks_per_gpu):
-                rows.append((ps.id, slot, "gpu", device_ids[i], 0, "idle"))
-                slot += 1
-        for _ in range(ps.cpu_slots):
-            rows.append((ps.id, slot, "cpu", None, 1, "idle"))
-            slot += 1
         self.store.executemany(
             "INSERT INTO slots (pool_id, slot_id, kind, device_id, "
-            "dedicated, state) VALUES (?,?,?,?,?,?)", rows)
+            "dedicated, state, node_id) VALUES (?,?,?,?,?,?,?)", rows)
 
     def wait_for_pool_ready(self, pool_id: str,
                             timeout: float = 300.0) -> None:
@@ -172,7 +207,9 @@ class LocalExecutor:
         ps = cfg.pool_settings(conf)
         t0 = time.time()
         try:
-            if ps.rocm_verify and (ps.gpus_dedicated + ps.gpus_low_priority):
+            # multi-node pools: GPUs live on the agents' hosts, not here
+            if (ps.rocm_verify and not ps.nodes
+                    and (ps.gpus_dedicated + ps.gpus_low_priority)):
                 self._verify_rocm(ps)
             for cmd in ps.start_task_pre + ps.start_task_post:
                 rc, out, err = utils.subprocess_with_output(
@@ -219,6 +256,7 @@ class LocalExecutor:
             h = self._handles.pop((r["job_id"], r["id"]), None)
             if h:
                 h.kill()
+        self.stop_local_agents(pool_id)
         with self.store.transaction() as conn:
             conn.execute("DELETE FROM slots WHERE pool_id=?", (pool_id,))
             conn.execute(
@@ -226,6 +264,9 @@ class LocalExecutor:
                 "(SELECT id FROM jobs WHERE pool_id=?)", (pool_id,))
             conn.execute("DELETE FROM jobs WHERE pool_id=?", (pool_id,))
             conn.execute("DELETE FROM pools WHERE id=?", (pool_id,))
+            conn.execute("DELETE FROM nodes WHERE pool_id=?", (pool_id,))
+            conn.execute("DELETE FROM assignments WHERE pool_id=?",
+                         (pool_id,))
 
     def pool_resize(self, pool_id: str, dedicated: Optional[int] = None,
                     low_priority: Optional[int] = None) -> None:
@@ -493,6 +534,7 @@ class LocalExecutor:
                     self._release_slots(
                         self._job_pool(jid),
                         json.loads(row["slots_json"] or "[]"))
+        self._cancel_assignments(job_id)
         with self.store.transaction() as conn:
             conn.execute(
                 "UPDATE tasks SET state='cancelled', end_time=? WHERE "
@@ -502,6 +544,32 @@ class LocalExecutor:
             conn.execute(
                 "UPDATE jobs SET state='terminated', completed_at=? "
                 "WHERE id=?", (time.time(), job_id))
+
+    def _cancel_assignments(self, job_id: str,
+                            task_id: Optional[str] = None) -> None:
+        """Remote termination: unclaimed windows are dropped, running
+        ones flip to cancelling so the node agent tears them down."""
+        extra = " AND task_id=?" if task_id else ""
+        args = [job_id] + ([task_id] if task_id else [])
+        # release the slots of affected running tasks (the local path
+        # does this via the handle; remote tasks have no handle here)
+        for r in self.store.query(
+                "SELECT DISTINCT job_id, task_id FROM assignments "
+                f"WHERE job_id=?{extra}", args):
+            row = self.store.query_one(
+                "SELECT slots_json, state FROM tasks WHERE job_id=? AND "
+                "id=?", (r["job_id"], r["task_id"]))
+            if row and row["state"] == "running":
+                self._release_slots(
+                    self._job_pool(r["job_id"]),
+                    json.loads(row["slots_json"] or "[]"))
+        self.store.execute(
+            f"DELETE FROM assignments WHERE job_id=?{extra} AND "
+            "state='queued'", args)
+        self.store.execute(
+            "UPDATE assignments SET state='cancelling', updated_at=? "
+            f"WHERE job_id=?{extra} AND state='running'",
+            [time.time()] + args)
 
     def task_terminate(self, job_id: str, task_id: str) -> None:
         """Terminate one task (reference `jobs tasks term`): kill if
@@ -515,6 +583,7 @@ class LocalExecutor:
             if row:
                 self._release_slots(self._job_pool(job_id),
                                     json.loads(row["slots_json"] or "[]"))
+        self._cancel_assignments(job_id, task_id)
         self.store.execute(
             "UPDATE tasks SET state='cancelled', end_time=? WHERE "
             "job_id=? AND id=? AND state IN "
@@ -564,6 +633,79 @@ class LocalExecutor:
         self.store.execute(
             "UPDATE slots SET state='idle' WHERE pool_id=? AND slot_id=? "
             "AND state='offline'", (pool_id, slot_id))
+
+    # ----------------------------------------------------------------
+    # multi-node agents (reference: the Batch agent per VM + nodeprep;
+    # here `python -m shipyard_amd.agent` per node over a shared root)
+    # ----------------------------------------------------------------
+    def nodes_list(self, pool_id: str) -> List[dict]:
+        return [dict(r) for r in self.store.query(
+            "SELECT node_id, host, state, heartbeat, agent_pid FROM nodes "
+            "WHERE pool_id=? ORDER BY node_id", (pool_id,))]
+
+    def agent_command(self, pool_id: str,
+                      node: cfg.NodeSettings) -> List[str]:
+        """The command that starts `node`'s agent — ssh-wrapped for
+        remote hosts (reference fleet.py:2045 SSH fan-out)."""
+        import sys as _sys
+
+        base = [_sys.executable, "-m", "shipyard_amd.agent",
+                "--root", str(self.root), "--pool", pool_id,
+                "--node", node.id]
+        if node.host in ("127.0.0.1", "localhost"):
+            return base
+        from shipyard_amd.utils import crypto
+
+        return crypto.ssh_command(
+            node.host, " ".join(base), username=node.ssh_user,
+            private_key=node.ssh_private_key)
+
+    def start_local_agents(self, pool_id: str,
+                           poll: float = 0.05) -> List[Any]:
+        """Spawn agent processes for this pool's localhost nodes
+        (remote hosts: run agent_command there instead)."""
+        import os as _os
+        import subprocess as _sp
+        import sys as _sys
+
+        ps = self._pool_settings(pool_id)
+        procs = []
+        pkg_root = str(Path(__file__).resolve().parents[2])
+        env = dict(_os.environ)
+        pp = env.get("PYTHONPATH", "")
+        if pkg_root not in pp.split(_os.pathsep):
+            env["PYTHONPATH"] = pkg_root + (_os.pathsep + pp if pp else "")
+        for node in ps.nodes:
+            if node.host not in ("127.0.0.1", "localhost"):
+                continue
+            cmd = [_sys.executable, "-m", "shipyard_amd.agent",
+                   "--root", str(self.root), "--pool", pool_id,
+                   "--node", node.id, "--poll", str(poll)]
+            procs.append(_sp.Popen(cmd, env=env, start_new_session=True))
+        self._agents.setdefault(pool_id, []).extend(procs)
+        return procs
+
+    def stop_local_agents(self, pool_id: Optional[str] = None) -> None:
+        """SIGTERM this process's agents (they tear down their work and
+        mark their nodes offline) and wait briefly for exit."""
+        import signal as _signal
+        import subprocess as _sp
+
+        pools = [pool_id] if pool_id else list(self._agents)
+        procs = []
+        for pid in pools:
+            procs.extend(self._agents.pop(pid, []))
+        for p in procs:
+            if p.poll() is None:
+                try:
+                    p.send_signal(_signal.SIGTERM)
+                except ProcessLookupError:
+                    pass
+        for p in procs:
+            try:
+                p.wait(timeout=10)
+            except _sp.TimeoutExpired:
+                p.kill()
 
     def clean_retained(self, now: Optional[float] = None) -> int:
         """Delete task directories past their retention_time (the
@@ -637,6 +779,7 @@ class LocalExecutor:
         dependency-satisfied tasks, assign ready tasks to idle slots,
         launch.  Returns number of state transitions made."""
         n = 0
+        self.reap_dead_agents()
         n += self._collect_finished()
         n += self._promote_pending()
         n += self._assign_and_launch()
@@ -768,6 +911,79 @@ class LocalExecutor:
             del self._handles[(jid, tid)]
             self._finish_task(jid, tid, rc, h)
             n += 1
+        n += self._collect_remote()
+        return n
+
+    def _collect_remote(self) -> int:
+        """Aggregate agent-reported assignment exits into task results.
+        Gang semantics across nodes: any failed window cancels the
+        rest; the task's exit code is the first failure's."""
+        groups = self.store.query(
+            "SELECT job_id, task_id, COUNT(*) AS total, "
+            "SUM(state='done') AS done FROM assignments "
+            "GROUP BY job_id, task_id")
+        n = 0
+        for g in groups:
+            jid, tid = g["job_id"], g["task_id"]
+            trow = self.store.query_one(
+                "SELECT state FROM tasks WHERE job_id=? AND id=?",
+                (jid, tid))
+            if trow is None or trow["state"] != "running":
+                # job deleted / task collected or terminated: drop rows
+                # the agent no longer owns; running/cancelling ones are
+                # reaped on a later pass once the agent reports them
+                self.store.execute(
+                    "DELETE FROM assignments WHERE job_id=? AND task_id=? "
+                    "AND state IN ('queued','done')", (jid, tid))
+                continue
+            if g["done"] == g["total"]:
+                rcs = [r["rc"] for r in self.store.query(
+                    "SELECT rc FROM assignments WHERE job_id=? AND "
+                    "task_id=? ORDER BY id", (jid, tid))]
+                rc = next((c for c in rcs if c), 0)
+                self.store.execute(
+                    "DELETE FROM assignments WHERE job_id=? AND task_id=?",
+                    (jid, tid))
+                self._finish_task(jid, tid, rc or 0, _RemoteHandle())
+                n += 1
+            elif self.store.query_one(
+                    "SELECT 1 FROM assignments WHERE job_id=? AND "
+                    "task_id=? AND state='done' AND rc!=0", (jid, tid)):
+                # a window failed: tear down the rest of the gang
+                self.store.execute(
+                    "UPDATE assignments SET state='done', rc=-15, "
+                    "updated_at=? WHERE job_id=? AND task_id=? AND "
+                    "state='queued'", (time.time(), jid, tid))
+                self.store.execute(
+                    "UPDATE assignments SET state='cancelling', "
+                    "updated_at=? WHERE job_id=? AND task_id=? AND "
+                    "state='running'", (time.time(), jid, tid))
+        return n
+
+    def reap_dead_agents(self, max_age_s: float = 10.0) -> int:
+        """Fail work stuck on nodes whose agent stopped heartbeating
+        (the analogue of the reference's unusable-node recovery,
+        convoy/fleet.py attempt_recovery_on_unusable).  Nodes that
+        never had an agent (heartbeat 0) are left alone — their work
+        queues until an agent arrives, as pools wait for nodes."""
+        cutoff = time.time() - max_age_s
+        dead = self.store.query(
+            "SELECT pool_id, node_id FROM nodes WHERE heartbeat > 0 "
+            "AND heartbeat < ? AND state != 'offline'", (cutoff,))
+        n = 0
+        for d in dead:
+            self.store.execute(
+                "UPDATE nodes SET state='offline' WHERE pool_id=? AND "
+                "node_id=?", (d["pool_id"], d["node_id"]))
+            cur = self.store.execute(
+                "UPDATE assignments SET state='done', rc=-9, updated_at=? "
+                "WHERE pool_id=? AND node_id=? AND state IN "
+                "('queued','running','cancelling')",
+                (time.time(), d["pool_id"], d["node_id"]))
+            n += cur.rowcount
+            self.store.add_event(f"node:{d['pool_id']}/{d['node_id']}",
+                                 "agent_dead", {"failed_assignments":
+                                                cur.rowcount})
         return n
 
     def _finish_task(self, jid: str, tid: str, rc: int,
@@ -931,20 +1147,29 @@ class LocalExecutor:
                 ranks = 1
                 gpus_needed = ts.gpus
 
-            slots = self._try_allocate(pool_id, gpus_needed,
-                                       cpu_ok=(gpus_needed == 0))
+            slots = self._try_allocate(
+                pool_id, gpus_needed, cpu_ok=(gpus_needed == 0),
+                ranks=ranks, span_ok=(ts.multi_instance is not None),
+                chunk=(ts.multi_instance.gang.gpus_per_rank
+                       if ts.multi_instance else 1))
             if slots is None:
                 continue
             slot_ids = [s["slot_id"] for s in slots]
             device_ids = [s["device_id"] for s in slots
                           if s["kind"] == "gpu"]
+            node_ids = {s.get("node_id", "local") for s in slots}
             self.store.execute(
                 "UPDATE tasks SET state='running', start_time=?, "
                 "slots_json=? WHERE job_id=? AND id=?",
                 (time.time(), json.dumps(slot_ids), jid, tid))
             try:
-                handle = self._launch_task(ps, js, ts, jid, tid, device_ids,
-                                           ranks)
+                if node_ids == {"local"}:
+                    handle = self._launch_task(ps, js, ts, jid, tid,
+                                               device_ids, ranks)
+                    self._handles[(jid, tid)] = handle
+                else:
+                    self._dispatch_remote(ps, js, ts, jid, tid, slots,
+                                          ranks)
             except Exception as exc:  # launch failure = task failure
                 logger.error("launch failed for %s/%s: %s", jid, tid, exc)
                 self.store.execute(
@@ -953,45 +1178,118 @@ class LocalExecutor:
                     (time.time(), jid, tid))
                 self._release_slots(pool_id, slot_ids)
                 continue
-            self._handles[(jid, tid)] = handle
             self.store.add_event(f"task:{jid}/{tid}", "launched",
-                                 {"slots": slot_ids, "devices": device_ids})
+                                 {"slots": slot_ids, "devices": device_ids,
+                                  "nodes": sorted(node_ids)})
             n += 1
         return n
 
-    def _try_allocate(self, pool_id: str, gpus: int,
-                      cpu_ok: bool) -> Optional[List[dict]]:
-        with self.store.transaction() as conn:
-            if gpus > 0:
-                rows = list(conn.execute(
-                    "SELECT slot_id, kind, device_id FROM slots WHERE "
-                    "pool_id=? AND state='idle' AND kind='gpu' "
-                    "ORDER BY dedicated DESC, device_id ASC LIMIT ?",
-                    (pool_id, gpus)))
-                if len(rows) < gpus:
-                    return None
-                # prefer distinct devices for multi-gpu tasks
-                if gpus > 1:
-                    seen = set()
-                    distinct = []
-                    for row in conn.execute(
-                            "SELECT slot_id, kind, device_id FROM slots "
-                            "WHERE pool_id=? AND state='idle' AND "
-                            "kind='gpu' ORDER BY dedicated DESC, "
-                            "device_id ASC", (pool_id,)):
-                        if row["device_id"] not in seen:
-                            seen.add(row["device_id"])
-                            distinct.append(row)
-                    if len(distinct) >= gpus:
-                        rows = distinct[:gpus]
+    def _dispatch_remote(self, ps: cfg.PoolSettings, js: cfg.JobSettings,
+                         ts: cfg.TaskSettings, jid: str, tid: str,
+                         slots: List[dict], ranks: int) -> None:
+        """Compile per-node LaunchSpec windows and enqueue them on the
+        assignments table for the node agents (multi-node pools; gangs
+        get consecutive rank windows with the rendezvous on the node
+        holding rank 0)."""
+        import dataclasses as _dc
+
+        hosts = {nd.id: nd.host for nd in ps.nodes}
+        by_node: Dict[str, List[dict]] = {}
+        for s in slots:  # slot order defines rank order
+            by_node.setdefault(s["node_id"], []).append(s)
+        mi = ts.multi_instance
+        per_rank = mi.gang.gpus_per_rank if mi else 1
+        spec0 = self._build_spec(ps, js, ts, jid, tid, [], ranks)
+        master_addr = None
+        port = None
+        if ranks > 1:
+            first = next(iter(by_node))
+            master_addr = hosts.get(first, "127.0.0.1")
+            port = (mi.gang.master_port if mi and mi.gang.master_port
+                    else _free_port())
+        rows = []
+        rank_start = 0
+        for node_id, node_slots in by_node.items():
+            devs = [s["device_id"] for s in node_slots
+                    if s["kind"] == "gpu"]
+            if ranks > 1:
+                ranks_here = (len(devs) // per_rank if per_rank
+                              else len(node_slots))
             else:
-                rows = list(conn.execute(
-                    "SELECT slot_id, kind, device_id FROM slots WHERE "
-                    "pool_id=? AND state='idle' "
-                    "ORDER BY kind='gpu', slot_id LIMIT 1", (pool_id,)))
-                if not rows:
+                ranks_here = 1
+            spec = _dc.replace(
+                spec0, device_ids=devs, num_instances=ranks_here,
+                rank_start=rank_start,
+                world_size=(ranks if ranks > 1 else None),
+                master_addr=master_addr or "127.0.0.1",
+                master_port=port if port else spec0.master_port)
+            rows.append((ps.id, node_id, jid, tid, spec_to_json(spec),
+                         time.time()))
+            rank_start += ranks_here
+        if ranks > 1 and rank_start != ranks:
+            raise ExecutorError(
+                f"gang window mismatch for {jid}/{tid}: "
+                f"{rank_start} != {ranks}")
+        self.store.executemany(
+            "INSERT INTO assignments (pool_id, node_id, job_id, task_id, "
+            "spec_json, created_at) VALUES (?,?,?,?,?,?)", rows)
+
+    def _try_allocate(self, pool_id: str, gpus: int, cpu_ok: bool,
+                      ranks: int = 1, span_ok: bool = False,
+                      chunk: int = 1) -> Optional[List[dict]]:
+        """Claim slots for a task.  Single tasks stay on one node;
+        gangs (span_ok) may span nodes, each node contributing a
+        multiple of `chunk` (= gpus_per_rank) devices so every rank's
+        devices are co-resident."""
+        with self.store.transaction() as conn:
+            multi = conn.execute(
+                "SELECT COUNT(*) FROM nodes WHERE pool_id=?",
+                (pool_id,)).fetchone()[0] > 0
+            if gpus > 0:
+                idle = list(conn.execute(
+                    "SELECT slot_id, kind, device_id, node_id FROM slots "
+                    "WHERE pool_id=? AND state='idle' AND kind='gpu' "
+                    "ORDER BY node_id, dedicated DESC, device_id ASC",
+                    (pool_id,)))
+                # distinct devices per node, preserving order
+                per_node: Dict[str, List] = {}
+                for row in idle:
+                    bucket = per_node.setdefault(row["node_id"], [])
+                    if row["device_id"] not in {r["device_id"]
+                                                for r in bucket}:
+                        bucket.append(row)
+                rows = None
+                # single-node fit on distinct devices
+                for bucket in per_node.values():
+                    if len(bucket) >= gpus:
+                        rows = bucket[:gpus]
+                        break
+                # same-device oversubscription (max_tasks_per_gpu > 1)
+                # is a single-host concept
+                if rows is None and not multi and len(idle) >= gpus:
+                    rows = idle[:gpus]
+                # gang spanning nodes, chunk-aligned per node
+                if rows is None and span_ok and multi:
+                    acc: List = []
+                    for bucket in per_node.values():
+                        take = (len(bucket) // chunk) * chunk \
+                            if chunk > 1 else len(bucket)
+                        acc.extend(bucket[:take])
+                    if len(acc) >= gpus:
+                        rows = acc[:gpus]
+                if rows is None:
                     return None
-                rows = rows[:1]
+            else:
+                # cpu task: 1 slot; cpu gang on a multi-node pool: one
+                # slot per rank (possibly across nodes)
+                want = ranks if (span_ok and multi and ranks > 1) else 1
+                rows = list(conn.execute(
+                    "SELECT slot_id, kind, device_id, node_id FROM slots "
+                    "WHERE pool_id=? AND state='idle' "
+                    "ORDER BY kind='gpu', node_id, slot_id LIMIT ?",
+                    (pool_id, want)))
+                if len(rows) < want:
+                    return None
             for row in rows:
                 conn.execute(
                     "UPDATE slots SET state='busy' WHERE pool_id=? AND "
@@ -1124,6 +1422,12 @@ class LocalExecutor:
     def _launch_task(self, ps: cfg.PoolSettings, js: cfg.JobSettings,
                      ts: cfg.TaskSettings, jid: str, tid: str,
                      device_ids: List[int], ranks: int) -> TaskHandle:
+        spec = self._build_spec(ps, js, ts, jid, tid, device_ids, ranks)
+        return launch(spec, self.pool_root(ps.id))
+
+    def _build_spec(self, ps: cfg.PoolSettings, js: cfg.JobSettings,
+                    ts: cfg.TaskSettings, jid: str, tid: str,
+                    device_ids: List[int], ranks: int) -> LaunchSpec:
         mi = ts.multi_instance
         env = dict(ps.environment_variables)
         env.update(ts.environment_variables)
@@ -1184,7 +1488,7 @@ class LocalExecutor:
                              if ts.max_wall_time else None),
             wrapper=wrapper,
         )
-        return launch(spec, self.pool_root(ps.id))
+        return spec
 
     def _complete_auto_jobs(self) -> None:
         rows = self.store.query(

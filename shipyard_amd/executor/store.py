@@ -37,7 +37,30 @@ CREATE TABLE IF NOT EXISTS slots (
     dedicated INTEGER NOT NULL DEFAULT 1,
     state TEXT NOT NULL DEFAULT 'idle',  -- idle|busy|offline|starting
     task_ref TEXT,
+    node_id TEXT NOT NULL DEFAULT 'local',
     PRIMARY KEY (pool_id, slot_id)
+);
+CREATE TABLE IF NOT EXISTS nodes (
+    pool_id TEXT NOT NULL,
+    node_id TEXT NOT NULL,
+    host TEXT NOT NULL DEFAULT '127.0.0.1',
+    state TEXT NOT NULL DEFAULT 'offline',  -- offline|idle|running
+    heartbeat REAL NOT NULL DEFAULT 0,
+    agent_pid INTEGER,
+    PRIMARY KEY (pool_id, node_id)
+);
+CREATE TABLE IF NOT EXISTS assignments (
+    id INTEGER PRIMARY KEY AUTOINCREMENT,
+    pool_id TEXT NOT NULL,
+    node_id TEXT NOT NULL,
+    job_id TEXT NOT NULL,
+    task_id TEXT NOT NULL,
+    state TEXT NOT NULL DEFAULT 'queued',
+        -- queued|running|cancelling|done
+    spec_json TEXT NOT NULL,
+    rc INTEGER,
+    created_at REAL NOT NULL,
+    updated_at REAL NOT NULL DEFAULT 0
 );
 CREATE TABLE IF NOT EXISTS jobs (
     id TEXT PRIMARY KEY,
@@ -109,7 +132,15 @@ class Store:
         with self._lock:
             self._conn.execute("PRAGMA journal_mode=WAL")
             self._conn.execute("PRAGMA synchronous=NORMAL")
+            # node agents are separate processes on the same DB
+            self._conn.execute("PRAGMA busy_timeout=10000")
             self._conn.executescript(_SCHEMA)
+            # migration: slots.node_id added after v0 databases existed
+            cols = {r[1] for r in self._conn.execute(
+                "PRAGMA table_info(slots)")}
+            if "node_id" not in cols:
+                self._conn.execute("ALTER TABLE slots ADD COLUMN node_id "
+                                   "TEXT NOT NULL DEFAULT 'local'")
             self._conn.commit()
 
     def close(self) -> None:

@@ -83,6 +83,28 @@ class LaunchSpec:
     # command-vector prefix, e.g. ["rocprofv3", ..., "--"] for per-task
     # kernel tracing (vector level: no shell quoting hazards)
     wrapper: List[str] = field(default_factory=list)
+    # multi-node gang window: this launch starts ranks
+    # [rank_start, rank_start + num_instances) of world_size total, with
+    # the rendezvous at master_addr (rank 0's node).  Defaults reproduce
+    # the single-node behavior (window = whole gang).
+    rank_start: int = 0
+    world_size: Optional[int] = None
+    master_addr: str = "127.0.0.1"
+
+
+def spec_to_json(spec: LaunchSpec) -> str:
+    """Serialize for the multi-node assignment queue (the coordinator
+    compiles the LaunchSpec; the node agent re-hydrates and launches)."""
+    import dataclasses
+    import json
+
+    return json.dumps(dataclasses.asdict(spec))
+
+
+def spec_from_json(s: str) -> LaunchSpec:
+    import json
+
+    return LaunchSpec(**json.loads(s))
 
 
 @dataclass
@@ -273,13 +295,14 @@ def launch(spec: LaunchSpec, pool_root: Path) -> TaskHandle:
                                                         exist_ok=True)
     start = time.monotonic()
     ranks: List[RankProc] = []
-    n = max(1, spec.num_instances)
+    n = max(1, spec.num_instances)       # ranks launched HERE
+    world = spec.world_size or n         # whole-gang size
 
     command = spec.command
     if spec.pre_execution_command:
         command = f"{spec.pre_execution_command}; {command}"
 
-    if n == 1:
+    if world == 1:
         paths = TaskPaths.create(pool_root, spec.job_id, spec.task_id)
         env = _base_env(spec, paths, pool_root)
         env.update(rt.gpu_env(spec.device_ids))
@@ -288,22 +311,23 @@ def launch(spec: LaunchSpec, pool_root: Path) -> TaskHandle:
         ranks.append(_spawn(cmd, env, paths, 0))
     else:
         port = spec.master_port or _free_port()
-        # split granted devices across ranks
+        # split LOCALLY granted devices across local ranks
         per_rank = spec.gpus_per_rank
-        for rank in range(n):
+        for local in range(n):
+            rank = spec.rank_start + local
             paths = TaskPaths.create(pool_root, spec.job_id, spec.task_id,
                                      rank=rank)
             env = _base_env(spec, paths, pool_root)
-            devs = spec.device_ids[rank * per_rank:(rank + 1) * per_rank] \
+            devs = spec.device_ids[local * per_rank:(local + 1) * per_rank] \
                 if per_rank else []
             env.update(rt.gpu_env(devs))
             env.update({
                 "RANK": str(rank),
-                "LOCAL_RANK": "0" if devs else str(rank),
-                "WORLD_SIZE": str(n),
-                "MASTER_ADDR": "127.0.0.1",
+                "LOCAL_RANK": "0" if devs else str(local),
+                "WORLD_SIZE": str(world),
+                "MASTER_ADDR": spec.master_addr,
                 "MASTER_PORT": str(port),
-                "SHIPYARD_GANG_SIZE": str(n),
+                "SHIPYARD_GANG_SIZE": str(world),
                 "SHIPYARD_GANG_RANK": str(rank),
                 "SHIPYARD_GANG_BACKEND": spec.gang_backend,
             })
