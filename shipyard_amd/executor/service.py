@@ -937,9 +937,13 @@ class LocalExecutor:
                     "AND state IN ('queued','done')", (jid, tid))
                 continue
             if g["done"] == g["total"]:
+                # originating-failure semantics: the first window to
+                # fail in TIME carries the task's exit code (peers are
+                # torn down afterwards with -15, matching the
+                # single-node gang's _first_fail behavior)
                 rcs = [r["rc"] for r in self.store.query(
                     "SELECT rc FROM assignments WHERE job_id=? AND "
-                    "task_id=? ORDER BY id", (jid, tid))]
+                    "task_id=? ORDER BY updated_at, id", (jid, tid))]
                 rc = next((c for c in rcs if c), 0)
                 self.store.execute(
                     "DELETE FROM assignments WHERE job_id=? AND task_id=?",

@@ -1,0 +1,210 @@
+"""Multi-node pools: node agents over the shared store, assignment
+dispatch, cross-node gangs (gloo on CPU; RCCL path identical modulo
+backend), agent failure handling.  (Reference analogue: pools spanning
+VMs with the Batch agent per node + multi-instance tasks,
+convoy/batch.py:4590-4698.)"""
+import textwrap
+import time
+
+import pytest
+
+from shipyard_amd.executor import LocalExecutor
+
+
+def _mk_pool(ex, cpu_per_node=2, n_nodes=2):
+    ex.pool_add({"pool_specification": {
+        "id": "mp",
+        "nodes": [{"id": f"n{i}", "host": "127.0.0.1",
+                   "cpu_slots": cpu_per_node} for i in range(n_nodes)],
+        "node_configuration": {"rocm": {"verify": False}}}})
+
+
+@pytest.fixture
+def mx(tmp_path):
+    ex = LocalExecutor(tmp_path / "root", detect_gpus=False)
+    yield ex
+    ex.stop_local_agents()
+    ex.store.close()
+
+
+class TestMultiNodePool:
+    def test_slots_per_node(self, mx):
+        _mk_pool(mx)
+        rows = mx.store.query(
+            "SELECT node_id, COUNT(*) n FROM slots WHERE pool_id='mp' "
+            "GROUP BY node_id ORDER BY node_id")
+        assert [(r["node_id"], r["n"]) for r in rows] == \
+            [("n0", 2), ("n1", 2)]
+        assert [n["node_id"] for n in mx.nodes_list("mp")] == ["n0", "n1"]
+
+    def test_gpu_nodes_skip_local_rocm_verify(self, tmp_path):
+        # verify=True must NOT check the local host for a multi-node
+        # pool — its GPUs live on the agents' hosts
+        ex = LocalExecutor(tmp_path / "r2", detect_gpus=False)
+        ex.pool_add({"pool_specification": {
+            "id": "gp",
+            "nodes": [{"id": "a", "gpus": {"dedicated": 4}},
+                      {"id": "b", "gpus": {"dedicated": 4}}],
+            "node_configuration": {"rocm": {"verify": True}}}})
+        kinds = mxq = ex.store.query(
+            "SELECT kind, device_id, node_id FROM slots WHERE "
+            "pool_id='gp' ORDER BY slot_id")
+        assert len(mxq) == 8
+        assert {r["node_id"] for r in kinds} == {"a", "b"}
+        # per-node device ids restart at 0
+        assert [r["device_id"] for r in kinds] == [0, 1, 2, 3, 0, 1, 2, 3]
+        ex.store.close()
+
+    def test_agent_command_ssh_synthesis(self, mx):
+        from shipyard_amd.config import settings as cfg
+
+        _mk_pool(mx)
+        local = cfg.NodeSettings(id="n0", host="127.0.0.1")
+        remote = cfg.NodeSettings(id="nx", host="10.0.0.7",
+                                  ssh_user="ops", ssh_private_key="/k")
+        assert mx.agent_command("mp", local)[1:3] == \
+            ["-m", "shipyard_amd.agent"]
+        cmd = mx.agent_command("mp", remote)
+        assert cmd[0] == "ssh" and "ops@10.0.0.7" in cmd
+        assert "-i" in cmd and "/k" in cmd
+
+
+class TestAgentExecution:
+    def test_tasks_distributed_across_nodes(self, mx):
+        _mk_pool(mx)
+        mx.start_local_agents("mp")
+        mx.jobs_add({"job_specifications": [{
+            "id": "mj",
+            "tasks": [{"id": f"t{i}",
+                       "command": "echo ran-on $SHIPYARD_TASK_ID"}
+                      for i in range(4)],
+        }]}, "mp")
+        mx.run_until_idle(timeout=60)
+        states = {t["id"]: t["state"] for t in mx.tasks_list("mj")}
+        assert set(states.values()) == {"completed"}
+        # both agents did work (assignment events name the nodes)
+        used = {e["source"] for e in mx.store.query(
+            "SELECT source FROM events WHERE category='launched'")}
+        assert len(used) == 4
+
+    def test_cross_node_gang_allreduce(self, mx, tmp_path):
+        prog = tmp_path / "gang.py"
+        prog.write_text(textwrap.dedent("""
+            import torch, torch.distributed as dist
+            dist.init_process_group('gloo')
+            t = torch.tensor([float(dist.get_rank() + 1)])
+            dist.all_reduce(t)
+            assert t.item() == 3.0, t
+            print('rank', dist.get_rank(), 'sum', t.item(), flush=True)
+            dist.destroy_process_group()
+        """))
+        _mk_pool(mx, cpu_per_node=1)
+        mx.start_local_agents("mp")
+        mx.jobs_add({"job_specifications": [{
+            "id": "gj",
+            "tasks": [{
+                "id": "gang",
+                "command": f"python3 {prog}",
+                "multi_instance": {
+                    "num_instances": 2,
+                    "gang": {"backend": "gloo", "gpus_per_rank": 0},
+                },
+            }],
+        }]}, "mp")
+        mx.run_until_idle(timeout=120)
+        t = mx.tasks_list("gj")[0]
+        assert t["state"] == "completed", t
+        for rank in (0, 1):
+            out = (mx.pool_root("mp") / "jobs" / "gj" / "tasks" / "gang"
+                   / f"rank{rank:03d}" / "stdout.txt").read_text()
+            assert "sum 3.0" in out
+        # windows were separate assignments, cleaned up after collect
+        assert mx.store.query_one(
+            "SELECT COUNT(*) n FROM assignments")["n"] == 0
+
+    def test_failed_gang_window_cancels_peers(self, mx, tmp_path):
+        """Rank 1's window fails fast; rank 0 blocks in rendezvous on
+        its own node and must be torn down by the coordinator."""
+        prog = tmp_path / "failing.py"
+        prog.write_text(textwrap.dedent("""
+            import os, sys, time
+            if os.environ['RANK'] == '1':
+                sys.exit(7)
+            time.sleep(600)   # rank 0 waits forever
+        """))
+        _mk_pool(mx, cpu_per_node=1)
+        mx.start_local_agents("mp")
+        mx.jobs_add({"job_specifications": [{
+            "id": "fj",
+            "tasks": [{
+                "id": "gang",
+                "command": f"python3 {prog}",
+                "max_task_retries": 0,
+                "multi_instance": {
+                    "num_instances": 2,
+                    "gang": {"backend": "gloo", "gpus_per_rank": 0},
+                },
+            }],
+        }]}, "mp")
+        mx.run_until_idle(timeout=90)
+        t = mx.tasks_list("fj")[0]
+        assert t["state"] == "failed"
+        assert t["exit_code"] == 7  # originating rank's code, not -15
+
+    def test_dead_agent_reaped(self, mx):
+        _mk_pool(mx, cpu_per_node=1, n_nodes=1)
+        procs = mx.start_local_agents("mp")
+        mx.jobs_add({"job_specifications": [{
+            "id": "dj",
+            "tasks": [{"id": "t", "command": "sleep 600",
+                       "max_task_retries": 0}],
+        }]}, "mp")
+        # let the agent claim the work
+        deadline = time.monotonic() + 30
+        while time.monotonic() < deadline:
+            mx.schedule_once()
+            if mx.store.query_one(
+                    "SELECT 1 FROM assignments WHERE state='running'"):
+                break
+            time.sleep(0.05)
+        else:
+            pytest.fail("agent never claimed the task")
+        # kill the agent outright (no graceful shutdown)
+        procs[0].kill()
+        procs[0].wait(timeout=10)
+        assert mx.reap_dead_agents(max_age_s=0.0) >= 1
+        mx.run_until_idle(timeout=30)
+        t = mx.tasks_list("dj")[0]
+        assert t["state"] == "failed" and t["exit_code"] == -9
+        node = mx.nodes_list("mp")[0]
+        assert node["state"] == "offline"
+
+    def test_job_terminate_cancels_remote(self, mx):
+        _mk_pool(mx, cpu_per_node=1, n_nodes=1)
+        mx.start_local_agents("mp")
+        mx.jobs_add({"job_specifications": [{
+            "id": "tj", "tasks": [{"id": "t", "command": "sleep 600"}],
+        }]}, "mp")
+        deadline = time.monotonic() + 30
+        while time.monotonic() < deadline:
+            mx.schedule_once()
+            if mx.store.query_one(
+                    "SELECT 1 FROM assignments WHERE state='running'"):
+                break
+            time.sleep(0.05)
+        mx.job_terminate("tj")
+        t = mx.tasks_list("tj")[0]
+        assert t["state"] == "cancelled"
+        # the agent notices the cancelling flag and reports; rows drain
+        deadline = time.monotonic() + 30
+        while time.monotonic() < deadline:
+            mx.schedule_once()
+            if mx.store.query_one(
+                    "SELECT COUNT(*) n FROM assignments")["n"] == 0:
+                break
+            time.sleep(0.05)
+        assert mx.store.query_one(
+            "SELECT COUNT(*) n FROM assignments")["n"] == 0
+        # slots are idle again
+        assert mx.store.query_one(
+            "SELECT COUNT(*) n FROM slots WHERE state='idle'")["n"] == 1
