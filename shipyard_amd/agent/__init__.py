@@ -44,10 +44,16 @@ class NodeAgent:
     # -- lifecycle ----------------------------------------------------
     def heartbeat(self) -> None:
         state = "running" if self._handles else "idle"
-        self.store.execute(
+        cur = self.store.execute(
             "UPDATE nodes SET heartbeat=?, state=?, agent_pid=? "
             "WHERE pool_id=? AND node_id=?",
             (time.time(), state, os.getpid(), self.pool_id, self.node_id))
+        if cur.rowcount == 0:
+            # pool (or this node) was deleted: tear down and exit —
+            # `pool del` on the coordinator stops agents on every host
+            logger.info("node %s: pool %s gone, agent exiting",
+                        self.node_id, self.pool_id)
+            self.stop()
 
     def pool_root(self) -> Path:
         return self.root / "pools" / self.pool_id

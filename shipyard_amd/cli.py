@@ -737,6 +737,49 @@ def nodes_list(ctx, poolid, configdir, root, raw):
     ctx.emit([dict(r) for r in rows])
 
 
+@pool_nodes.command("hosts")
+@click.option("--poolid", required=True)
+@_common
+@pass_ctx
+def nodes_hosts(ctx, poolid, configdir, root, raw):
+    """Multi-node pools: per-node agent state + heartbeats."""
+    _apply(ctx, configdir, root, raw)
+    ctx.emit(ctx.executor.nodes_list(poolid))
+
+
+@pool.group("agents")
+def pool_agents():
+    """Node agents of multi-node pools (`python -m shipyard_amd.agent`
+    per node over the shared root)."""
+
+
+@pool_agents.command("start")
+@click.option("--poolid", required=True)
+@_common
+@pass_ctx
+def agents_start(ctx, poolid, configdir, root, raw):
+    """Spawn agents for this pool's localhost nodes; print the ssh
+    command for remote ones."""
+    _apply(ctx, configdir, root, raw)
+    procs = ctx.executor.start_local_agents(poolid)
+    ps = ctx.executor.pool_settings_of(poolid)
+    remote = {n.id: " ".join(ctx.executor.agent_command(poolid, n))
+              for n in ps.nodes
+              if n.host not in ("127.0.0.1", "localhost")}
+    ctx.emit({"started": [p.pid for p in procs],
+              "remote_commands": remote})
+
+
+@pool_agents.command("stop")
+@click.option("--poolid", required=True)
+@_common
+@pass_ctx
+def agents_stop(ctx, poolid, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ctx.executor.stop_local_agents(poolid)
+    ctx.emit({"stopped": poolid})
+
+
 @pool_nodes.command("offline")
 @click.option("--poolid", required=True)
 @click.option("--slot", type=int, required=True)

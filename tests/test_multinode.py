@@ -208,3 +208,58 @@ class TestAgentExecution:
         # slots are idle again
         assert mx.store.query_one(
             "SELECT COUNT(*) n FROM slots WHERE state='idle'")["n"] == 1
+
+
+class TestMultiNodeCLI:
+    def test_recipe_end_to_end(self, tmp_path):
+        """The multinode-two-agents recipe through the CLI verbs."""
+        from pathlib import Path
+
+        from click.testing import CliRunner
+
+        from shipyard_amd.cli import cli
+
+        recipes = Path(__file__).parents[1] / "recipes"
+        r = CliRunner()
+
+        def run(args):
+            return r.invoke(
+                cli, args + ["--configdir",
+                             str(recipes / "multinode-two-agents"),
+                             "--root", str(tmp_path / "root")],
+                catch_exceptions=False)
+
+        try:
+            res = run(["pool", "add"])
+            assert res.exit_code == 0, res.output
+            res = run(["pool", "agents", "start", "--poolid", "duo"])
+            assert res.exit_code == 0 and "started" in res.output
+            res = run(["jobs", "add", "--wait"])
+            assert res.exit_code == 0, res.output
+            res = run(["jobs", "stats", "--jobid", "duo-gang"])
+            assert '"completed": 1' in res.output, res.output
+            res = run(["pool", "nodes", "hosts", "--poolid", "duo"])
+            assert res.exit_code == 0 and "n0" in res.output
+            import json as _json
+
+            pids = [n["agent_pid"] for n in _json.loads(res.output)
+                    if n["agent_pid"]]
+            assert pids, res.output
+        finally:
+            res = run(["pool", "del", "--poolid", "duo", "--force"])
+        # detached agents notice their node rows vanished and exit
+        deadline = time.monotonic() + 30
+        import os
+
+        def alive(pid):
+            # zombies count as exited (the agents are children of this
+            # in-process CliRunner and are reaped lazily)
+            try:
+                with open(f"/proc/{pid}/stat") as f:
+                    return f.read().rsplit(")", 1)[1].split()[0] != "Z"
+            except (FileNotFoundError, ProcessLookupError, IndexError):
+                return False
+
+        while time.monotonic() < deadline and any(alive(p) for p in pids):
+            time.sleep(0.1)
+        assert not any(alive(p) for p in pids), "agents leaked"

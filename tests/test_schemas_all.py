@@ -133,3 +133,20 @@ def test_typed_accessors_for_aux_families():
     s = slurm_settings(yaml.safe_load(VALID[ConfigType.slurm]))
     assert s.cluster_id == "sy"
     assert s.partitions[0].batch_pools["p"]["max_compute_nodes"] == 4
+
+
+def test_pool_nodes_multinode_schema():
+    """Multi-node pool surface (nodes: per-host slots + ssh)."""
+    validate_config(ConfigType.pool, {"pool_specification": {
+        "id": "mp",
+        "nodes": [
+            {"id": "n0", "host": "10.0.0.4", "gpus": {"dedicated": 8},
+             "ssh": {"username": "ops", "private_key": "/k"}},
+            {"id": "n1", "cpu_slots": 4},
+        ]}})
+    with pytest.raises(SchemaViolation):
+        validate_config(ConfigType.pool, {"pool_specification": {
+            "id": "mp", "nodes": [{"id": "n0", "bogus": True}]}})
+    with pytest.raises(SchemaViolation):  # node id required
+        validate_config(ConfigType.pool, {"pool_specification": {
+            "id": "mp", "nodes": [{"host": "10.0.0.4"}]}})
