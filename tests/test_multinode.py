@@ -263,3 +263,85 @@ class TestMultiNodeCLI:
         while time.monotonic() < deadline and any(alive(p) for p in pids):
             time.sleep(0.1)
         assert not any(alive(p) for p in pids), "agents leaked"
+
+
+@pytest.mark.gpu
+class TestMultiNodeGPU:
+    def test_agent_launches_gpu_task(self, tmp_path):
+        """A GPU node executed by its agent: the ROCm binder env and
+        device slots flow through the assignment queue."""
+        import torch
+
+        assert torch.cuda.is_available()
+        ex = LocalExecutor(tmp_path / "root", detect_gpus=True)
+        try:
+            ex.pool_add({"pool_specification": {
+                "id": "gmp",
+                "nodes": [{"id": "g0", "host": "127.0.0.1",
+                           "gpus": {"dedicated": 1}}],
+                "node_configuration": {"rocm": {"verify": False}}}})
+            ex.start_local_agents("gmp")
+            ex.jobs_add({"job_specifications": [{
+                "id": "gput",
+                "tasks": [{
+                    "id": "t", "gpus": 1,
+                    "command": "python3 -c \"import os, torch; "
+                               "assert os.environ['HIP_VISIBLE_DEVICES']"
+                               " == '0'; "
+                               "a = torch.randn(256, 256, device='cuda');"
+                               " print('gpu-ok', float((a @ a).sum()))\"",
+                }],
+            }]}, "gmp")
+            ex.run_until_idle(timeout=300)
+            t = ex.tasks_list("gput")[0]
+            assert t["state"] == "completed", t
+            out = ex.task_file("gmp", "gput", "t").read_text()
+            assert "gpu-ok" in out
+        finally:
+            ex.stop_local_agents()
+            ex.store.close()
+
+    def test_agent_rccl_gang_one_gpu(self, tmp_path):
+        """A world-1 RCCL gang through the agent: init_process_group
+        with backend nccl (=RCCL) works under agent-set rendezvous."""
+        import textwrap as tw
+
+        import torch
+
+        assert torch.cuda.is_available()
+        prog = tmp_path / "rccl1.py"
+        prog.write_text(tw.dedent("""
+            import torch, torch.distributed as dist
+            dist.init_process_group('nccl')
+            t = torch.ones(1024, device='cuda')
+            dist.all_reduce(t)
+            torch.cuda.synchronize()
+            assert t.sum().item() == 1024
+            print('rccl world', dist.get_world_size(), flush=True)
+            dist.destroy_process_group()
+        """))
+        ex = LocalExecutor(tmp_path / "root", detect_gpus=True)
+        try:
+            ex.pool_add({"pool_specification": {
+                "id": "gmp",
+                "nodes": [{"id": "g0", "host": "127.0.0.1",
+                           "gpus": {"dedicated": 1}}],
+                "node_configuration": {"rocm": {"verify": False}}}})
+            ex.start_local_agents("gmp")
+            ex.jobs_add({"job_specifications": [{
+                "id": "gang1",
+                "tasks": [{
+                    "id": "g",
+                    "command": f"python3 {prog}",
+                    "multi_instance": {
+                        "num_instances": 1,
+                        "gang": {"backend": "rccl", "gpus_per_rank": 1},
+                    },
+                }],
+            }]}, "gmp")
+            ex.run_until_idle(timeout=300)
+            t = ex.tasks_list("gang1")[0]
+            assert t["state"] == "completed", t
+        finally:
+            ex.stop_local_agents()
+            ex.store.close()
