@@ -1206,17 +1206,17 @@ class LocalExecutor:
         spec0 = self._build_spec(ps, js, ts, jid, tid, [], ranks)
         master_addr = None
         port = None
-        if ranks > 1:
+        if mi is not None:  # gangs of ANY size get the rendezvous env
             first = next(iter(by_node))
             master_addr = hosts.get(first, "127.0.0.1")
-            port = (mi.gang.master_port if mi and mi.gang.master_port
+            port = (mi.gang.master_port if mi.gang.master_port
                     else _free_port())
         rows = []
         rank_start = 0
         for node_id, node_slots in by_node.items():
             devs = [s["device_id"] for s in node_slots
                     if s["kind"] == "gpu"]
-            if ranks > 1:
+            if mi is not None:
                 ranks_here = (len(devs) // per_rank if per_rank
                               else len(node_slots))
             else:
@@ -1224,7 +1224,7 @@ class LocalExecutor:
             spec = _dc.replace(
                 spec0, device_ids=devs, num_instances=ranks_here,
                 rank_start=rank_start,
-                world_size=(ranks if ranks > 1 else None),
+                world_size=(ranks if mi is not None else None),
                 master_addr=master_addr or "127.0.0.1",
                 master_port=port if port else spec0.master_port)
             rows.append((ps.id, node_id, jid, tid, spec_to_json(spec),

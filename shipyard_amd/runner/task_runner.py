@@ -302,7 +302,9 @@ def launch(spec: LaunchSpec, pool_root: Path) -> TaskHandle:
     if spec.pre_execution_command:
         command = f"{spec.pre_execution_command}; {command}"
 
-    if world == 1:
+    if world == 1 and spec.world_size is None:
+        # plain task; an explicit world_size of 1 is a one-rank gang
+        # window and still gets the rendezvous env below
         paths = TaskPaths.create(pool_root, spec.job_id, spec.task_id)
         env = _base_env(spec, paths, pool_root)
         env.update(rt.gpu_env(spec.device_ids))
