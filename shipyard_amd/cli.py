@@ -684,6 +684,22 @@ def fed_jobs_add(ctx, federation_id, configdir, root, raw):
     ctx.emit({"queued": qid})
 
 
+@fed.command("jobs-term")
+@click.option("--federation-id", required=True)
+@click.option("--jobid", required=True)
+@_common
+@pass_ctx
+def fed_jobs_term(ctx, federation_id, jobid, configdir, root, raw):
+    """Queue a federation job cancellation (reference `fed jobs term`)."""
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.federation.scheduler import FederationProcessor
+
+    fp = FederationProcessor.from_config(
+        ctx.executor, ctx.conf(ConfigType.federation))
+    qid = fp.submit_cancel(federation_id, jobid)
+    ctx.emit({"queued": qid})
+
+
 @fed.command("process")
 @_common
 @pass_ctx

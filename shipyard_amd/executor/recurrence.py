@@ -25,10 +25,13 @@ def _parse_when(s: Optional[str]) -> Optional[float]:
 class JobScheduleRunner:
     """Tracks one recurring jobspec; spawns instance jobs on schedule."""
 
-    def __init__(self, executor, pool_id: str, jobspec: dict):
+    def __init__(self, executor, pool_id: str, jobspec: dict,
+                 sid: Optional[str] = None,
+                 state: Optional[dict] = None):
         self.ex = executor
         self.pool_id = pool_id
         self.jobspec = jobspec
+        self.sid = sid
         self.js = cfg.job_settings(jobspec)
         if self.js.recurrence is None:
             raise ValueError("jobspec has no recurrence")
@@ -38,6 +41,22 @@ class JobScheduleRunner:
         self.next_run = self.not_until or 0.0  # first run immediately
         self.instance = 0
         self.done = False
+        # durable progress: survive daemon restarts (the reference
+        # pickles the task map + relies on the Batch service's schedule
+        # state; here the kv record carries it)
+        if state:
+            self.instance = state.get("instance", 0)
+            self.next_run = state.get("next_run", self.next_run)
+            self.done = state.get("done", False)
+
+    def _persist(self) -> None:
+        if self.sid is None:
+            return
+        self.ex.store.kv_set(f"schedule:{self.sid}", json.dumps({
+            "pool": self.pool_id, "jobspec": self.jobspec,
+            "state": {"instance": self.instance,
+                      "next_run": self.next_run,
+                      "done": self.done}}))
 
     def maybe_spawn(self, now: Optional[float] = None) -> Optional[str]:
         now = now if now is not None else time.time()
@@ -45,6 +64,7 @@ class JobScheduleRunner:
             return None
         if self.not_after and now > self.not_after:
             self.done = True
+            self._persist()
             return None
         if self.js.recurrence.monitor_task_completion and self.instance:
             prev = f"{self.js.id}-{self.instance - 1:03d}"
@@ -61,4 +81,5 @@ class JobScheduleRunner:
         self.ex.jobs_add({"job_specifications": [spec]}, self.pool_id)
         self.instance += 1
         self.next_run = now + self.interval
+        self._persist()
         return inst_id

@@ -805,7 +805,8 @@ class LocalExecutor:
             if sid not in self._schedule_runners:
                 rec = json.loads(r["value"])
                 self._schedule_runners[sid] = JobScheduleRunner(
-                    self, rec["pool"], rec["jobspec"])
+                    self, rec["pool"], rec["jobspec"], sid=sid,
+                    state=rec.get("state"))
             inst = self._schedule_runners[sid].maybe_spawn(now)
             if inst:
                 spawned.append(inst)
@@ -845,10 +846,16 @@ class LocalExecutor:
             from shipyard_amd.executor.autoscale import AutoscaleController
 
             controllers = {}
+            last_sweep = 0.0
             while not self._sched_stop.is_set():
                 try:
                     self.schedule_once()
                     self.process_schedules()
+                    # retention sweeper (reference: task retention_time
+                    # is enforced by the Batch service; here the daemon)
+                    if time.time() - last_sweep > 60.0:
+                        last_sweep = time.time()
+                        self.clean_retained()
                     if autoscale:
                         now = time.time()
                         for p in self.pool_list():

@@ -74,6 +74,20 @@ class FederationProcessor:
              time.time()))
         return cur.lastrowid
 
+    def submit_cancel(self, federation_id: str, job_id: str) -> int:
+        """Queue a job cancellation (reference federation actions
+        beyond add: the fed proxy terminates the job at whichever
+        account/pool it landed on)."""
+        if federation_id not in self.federations:
+            raise KeyError(f"unknown federation {federation_id}")
+        cur = self.ex.store.execute(
+            "INSERT INTO fed_queue (federation_id, action, enqueued_at) "
+            "VALUES (?,?,?)",
+            (federation_id,
+             json.dumps({"kind": "cancel_job", "job_id": job_id}),
+             time.time()))
+        return cur.lastrowid
+
     # -- pool snapshots ------------------------------------------------
     def _snapshot(self, pool_id: str) -> Optional[PoolSnapshot]:
         row = self.ex.store.query_one(
@@ -185,6 +199,21 @@ class FederationProcessor:
             if fed is None:
                 self._mark(r["id"], "done")  # drop unknown federation
                 continue
+            if action["kind"] == "cancel_job":
+                jid = action["job_id"]
+                row = self.ex.store.query_one(
+                    "SELECT id, state FROM jobs WHERE id=?", (jid,))
+                if row is None:
+                    # may still be queued in a pending add_job action
+                    self._requeue(r)
+                    continue
+                if row["state"] in ("active", "disabled"):
+                    self.ex.job_terminate(jid)
+                self.ex.store.add_event(f"fed:{fid}", "job-cancelled",
+                                        {"job": jid})
+                self._mark(r["id"], "done")
+                n += 1
+                continue
             if action["kind"] != "add_job":
                 self._mark(r["id"], "done")
                 continue
@@ -197,6 +226,12 @@ class FederationProcessor:
                     break
                 target = self.find_target_pool_for_job(fed, jobspec)
                 if target is None:
+                    # no pool fits the job as written: try shrinking
+                    # int-sized gangs to each pool (reference
+                    # federation.py:2605 fixup_task_for_mismatch)
+                    target, jobspec = self._find_with_fixup(fed, fid,
+                                                            jobspec)
+                if target is None:
                     ok = False
                     break
                 placed.append((target, jobspec))
@@ -204,6 +239,7 @@ class FederationProcessor:
                 self._requeue(r)
                 continue
             for target, jobspec in placed:
+                jobspec = self._fixup_job_for_pool(fid, jobspec, target)
                 self.ex.jobs_add({"job_specifications": [jobspec]}, target)
                 self.ex.store.add_event(
                     f"fed:{fid}", "job-scheduled",
@@ -211,6 +247,46 @@ class FederationProcessor:
             self._mark(r["id"], "done")
             n += 1
         return n
+
+    def _find_with_fixup(self, fed: Federation, fid: str, jobspec: dict):
+        """Fallback placement: clamp the job to a candidate pool and
+        re-run the constraint match with the shrunken size."""
+        for pid in fed.pools:
+            cand = self._fixup_job_for_pool(fid, jobspec, pid, emit=False)
+            if cand == jobspec:
+                continue
+            target = self.find_target_pool_for_job(fed, cand)
+            if target is not None:
+                return target, self._fixup_job_for_pool(fid, jobspec,
+                                                        target)
+        return None, jobspec
+
+    def _fixup_job_for_pool(self, fid: str, jobspec: dict,
+                            pool_id: str, emit: bool = True) -> dict:
+        """Clamp gang sizes to the chosen pool's GPU slots (reference
+        federation.py:2605 fixup_task_for_mismatch: jobs written for a
+        bigger pool are rewritten for the one they landed on)."""
+        snap = self._snapshot(pool_id)
+        if snap is None:
+            return jobspec
+        total = snap.gpus_dedicated + snap.gpus_low_priority
+        out = json.loads(json.dumps(jobspec))
+        changed = False
+        for t in out.get("tasks", []):
+            mi = t.get("multi_instance")
+            if not mi:
+                continue
+            ni = mi.get("num_instances")
+            gpr = (mi.get("gang") or {}).get("gpus_per_rank", 1)
+            if isinstance(ni, int) and gpr and ni * gpr > total > 0:
+                mi["num_instances"] = max(total // gpr, 1)
+                changed = True
+        if changed and emit:
+            self.ex.store.add_event(
+                f"fed:{fid}", "job-fixup",
+                {"job": out.get("id"), "pool": pool_id,
+                 "gpu_slots": total})
+        return out
 
     def _mark(self, qid: int, state: str) -> None:
         self.ex.store.execute(

@@ -97,3 +97,59 @@ def test_unplaceable_job_backs_off(tmp_path):
     assert row["state"] == "blocked" and row["attempts"] == 1
     assert row["not_before"] > time.time()
     ex.store.close()
+
+
+def test_fed_cancel_job(tmp_path):
+    """cancel_job action terminates the job wherever it landed
+    (reference federation actions beyond add)."""
+    ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+    mkpool(ex, "qa", cpu=1)
+    fp = FederationProcessor(ex, {"f": Federation("f", ["qa"])})
+    fp.submit_job("f", {"job_specifications": [{
+        "id": "longjob", "tasks": [{"id": "t", "command": "sleep 600"}]}]})
+    assert fp.process_queue_once() == 1
+    ex.schedule_once()  # start the task
+    fp.submit_cancel("f", "longjob")
+    assert fp.process_queue_once() == 1
+    jobs = {j["id"]: j["state"] for j in ex.jobs_list()}
+    assert jobs["longjob"] == "terminated"
+    t = ex.tasks_list("longjob")[0]
+    assert t["state"] == "cancelled"
+    ex.store.close()
+
+
+def test_fed_cancel_unknown_job_backs_off(tmp_path):
+    ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+    mkpool(ex, "qa", cpu=1)
+    fp = FederationProcessor(ex, {"f": Federation("f", ["qa"])})
+    fp.submit_cancel("f", "nothere")
+    assert fp.process_queue_once() == 0
+    row = ex.store.query_one("SELECT * FROM fed_queue")
+    assert row["state"] == "blocked"
+    ex.store.close()
+
+
+def test_fed_fixup_gang_to_pool_size(tmp_path):
+    """A gang written for a bigger pool is rewritten to fit the pool it
+    lands on (reference federation.py:2605 fixup_task_for_mismatch)."""
+    ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+    mkpool(ex, "two", gpus=2)
+    fp = FederationProcessor(ex, {"f": Federation("f", ["two"])})
+    fp.submit_job("f", {"job_specifications": [{
+        "id": "gangjob", "tasks": [{
+            "id": "g", "command": "echo rank-ok",
+            "multi_instance": {
+                "num_instances": 4,
+                "gang": {"backend": "gloo", "gpus_per_rank": 1}},
+        }]}]})
+    assert fp.process_queue_once() == 1
+    ex.run_until_idle(timeout=60)
+    t = ex.tasks_list("gangjob")[0]
+    assert t["state"] == "completed"
+    base = ex.pool_root("two") / "jobs" / "gangjob" / "tasks" / "g"
+    ranks = sorted(p.name for p in base.glob("rank*"))
+    assert ranks == ["rank000", "rank001"]  # clamped 4 -> 2
+    ev = ex.store.query_one(
+        "SELECT payload FROM events WHERE category='job-fixup'")
+    assert ev is not None and "gpu_slots" in ev["payload"]
+    ex.store.close()

@@ -175,3 +175,33 @@ class TestCertsAndTLS:
         assert res.exit_code == 0 and "deleted" in res.output
         res = r.invoke(cli, ["cert", "list", *env_root])
         assert res.exit_code == 0 and "sha256" not in res.output
+
+
+def test_schedule_state_survives_restart(tmp_path):
+    """Recurrence progress (instance counter, next_run) is durable:
+    a restarted daemon continues numbering instead of re-spawning
+    instance 000 (the reference delegates this to the Batch service's
+    job-schedule state; here the kv record carries it)."""
+    import time as _time
+
+    ex = LocalExecutor(tmp_path / "rp", detect_gpus=False)
+    ex.pool_add({"pool_specification": {
+        "id": "prp", "gpus": {"dedicated": 0}, "cpu_slots": 1,
+        "node_configuration": {"rocm": {"verify": False}}}})
+    ex.jobs_add({"job_specifications": [{
+        "id": "sd", "tasks": [{"id": "t", "command": "true"}],
+        "recurrence": {"schedule": {"recurrence_interval": "01:00:00"}},
+    }]}, "prp")
+    assert ex.process_schedules() == ["sd-000"]
+    ex.run_until_idle(timeout=30)
+    ex.store.close()
+
+    # "restart": a fresh executor over the same root
+    ex2 = LocalExecutor(tmp_path / "rp", detect_gpus=False)
+    assert ex2.process_schedules() == []       # next_run is persisted
+    # one hour later the NEXT instance spawns (not a duplicate 000)
+    spawned = ex2.process_schedules(now=_time.time() + 3601)
+    assert spawned == ["sd-001"]
+    ex2.run_until_idle(timeout=30)
+    assert ex2.tasks_list("sd-001")[0]["state"] == "completed"
+    ex2.store.close()
