@@ -345,3 +345,45 @@ class TestMultiNodeGPU:
         finally:
             ex.stop_local_agents()
             ex.store.close()
+
+
+class TestMultiNodeStress:
+    def test_many_tasks_and_gangs_through_agents(self, mx, tmp_path):
+        """40 short tasks + 4 cpu gangs racing through 2 agents; every
+        task completes, no assignment rows leak, slots drain to idle."""
+        prog = tmp_path / "g.py"
+        prog.write_text(
+            "import torch, torch.distributed as dist\n"
+            "dist.init_process_group('gloo')\n"
+            "t = torch.ones(8)\n"
+            "dist.all_reduce(t)\n"
+            "assert t[0].item() == dist.get_world_size()\n"
+            "dist.destroy_process_group()\n")
+        _mk_pool(mx, cpu_per_node=2)
+        mx.start_local_agents("mp")
+        jobs = [{
+            "id": "bulk",
+            "tasks": [{"id": f"t{i}", "command": "true"}
+                      for i in range(40)],
+        }]
+        for g in range(4):
+            jobs.append({
+                "id": f"gang{g}",
+                "tasks": [{
+                    "id": "g",
+                    "command": f"python3 {prog}",
+                    "multi_instance": {
+                        "num_instances": 2,
+                        "gang": {"backend": "gloo", "gpus_per_rank": 0},
+                    },
+                }],
+            })
+        mx.jobs_add({"job_specifications": jobs}, "mp")
+        mx.run_until_idle(timeout=300)
+        for jid in ["bulk"] + [f"gang{g}" for g in range(4)]:
+            for t in mx.tasks_list(jid):
+                assert t["state"] == "completed", (jid, dict(t))
+        assert mx.store.query_one(
+            "SELECT COUNT(*) n FROM assignments")["n"] == 0
+        assert mx.store.query_one(
+            "SELECT COUNT(*) n FROM slots WHERE state!='idle'")["n"] == 0
