@@ -45,16 +45,16 @@ def corpus(target_mb: int = 512) -> bytes:
     return buf.getvalue()
 
 
-def main() -> None:
+def main(mb: int = 512, sizes=(4096, 8192, 16384, 32768)) -> None:
     assert torch.cuda.is_available()
     torch.zeros(1, device="cuda")
     torch.cuda.synchronize()
-    data = corpus()
+    data = corpus(mb)
     n = len(data)
     print(f"corpus {n/1e6:.0f} MB")
     td = Path(tempfile.mkdtemp(prefix="blk-bench-"))
     out = {}
-    for blk in (4096, 8192, 16384, 32768):
+    for blk in sizes:
         t0 = time.perf_counter()
         packed = shardfmt.pack(data, block_raw=blk)
         pack_s = time.perf_counter() - t0
@@ -86,4 +86,9 @@ def main() -> None:
 
 
 if __name__ == "__main__":
-    main()
+    import sys as _sys
+
+    mb = int(_sys.argv[1]) if len(_sys.argv) > 1 else 512
+    sizes = tuple(int(x) for x in _sys.argv[2].split(",")) \
+        if len(_sys.argv) > 2 else (4096, 8192, 16384, 32768)
+    main(mb, sizes)

@@ -53,12 +53,61 @@ class BlockEntry:
         return self.comp_len == self.raw_len
 
 
-@dataclass
+ENTRY_DT = None  # numpy structured dtype for the block table (lazy)
+
+
+def _entry_dt():
+    global ENTRY_DT
+    if ENTRY_DT is None:
+        import numpy as np
+
+        ENTRY_DT = np.dtype([("comp_off", "<u8"), ("comp_len", "<u4"),
+                             ("raw_len", "<u4"), ("crc", "<u4")])
+        assert ENTRY_DT.itemsize == ENTRY.size
+    return ENTRY_DT
+
+
 class ShardIndex:
-    block_raw: int
-    raw_size: int
-    blocks: List[BlockEntry]
-    payload_off: int  # byte offset of payload within the file
+    """Parsed shard index.  The block table is held as numpy column
+    arrays (comp_off/comp_len/raw_len/crc) so million-block shards
+    decode without per-block python; `.blocks` materializes the
+    BlockEntry view lazily for tests/CPU paths."""
+
+    def __init__(self, block_raw: int, raw_size: int, payload_off: int,
+                 table=None, blocks: List[BlockEntry] = None):
+        import numpy as np
+
+        self.block_raw = block_raw
+        self.raw_size = raw_size
+        self.payload_off = payload_off
+        if table is None:
+            blocks = blocks or []
+            table = np.zeros(len(blocks), dtype=_entry_dt())
+            for i, b in enumerate(blocks):
+                table[i] = (b.comp_off, b.comp_len, b.raw_len, b.crc32c)
+        self.table = table
+        self._blocks = blocks
+
+    @property
+    def n_blocks(self) -> int:
+        return len(self.table)
+
+    @property
+    def blocks(self) -> List[BlockEntry]:
+        if self._blocks is None:
+            self._blocks = [
+                BlockEntry(int(r["comp_off"]), int(r["comp_len"]),
+                           int(r["raw_len"]), int(r["crc"]))
+                for r in self.table]
+        return self._blocks
+
+    def raw_offs(self):
+        """Cumulative raw offset per block (numpy int64)."""
+        import numpy as np
+
+        out = np.zeros(len(self.table), dtype=np.int64)
+        np.cumsum(self.table["raw_len"][:-1], out=out[1:])
+        return out
 
 
 def _align16(n: int) -> int:
@@ -93,17 +142,16 @@ def pack(data: bytes, block_raw: int = DEFAULT_BLOCK_RAW,
 
 
 def read_index(buf: bytes) -> ShardIndex:
+    import numpy as np
+
     magic, flags, block_raw, raw_size, n_blocks = HEADER.unpack_from(buf, 0)
     if magic != MAGIC:
         raise ValueError("not a SYSHARD file")
-    blocks = []
-    pos = HEADER.size
-    for _ in range(n_blocks):
-        comp_off, comp_len, raw_len, crc = ENTRY.unpack_from(buf, pos)
-        pos += ENTRY.size
-        blocks.append(BlockEntry(comp_off, comp_len, raw_len, crc))
-    return ShardIndex(block_raw=block_raw, raw_size=raw_size, blocks=blocks,
-                      payload_off=pos)
+    table = np.frombuffer(buf, dtype=_entry_dt(), count=n_blocks,
+                          offset=HEADER.size)
+    return ShardIndex(block_raw=block_raw, raw_size=raw_size,
+                      payload_off=HEADER.size + n_blocks * ENTRY.size,
+                      table=table)
 
 
 def unpack_cpu(buf: bytes, verify: bool = True) -> bytes:
@@ -130,14 +178,88 @@ def _stored_contiguous(idx) -> bool:
     the decoded data and no per-block copies are needed.  This is the
     common case for incompressible shards packed with compress=False
     (block sizes that are 16 B multiples keep offsets aligned)."""
-    acc = 0
-    for b in idx.blocks:
-        if not b.stored or b.comp_off != acc:
-            return False
-        acc += b.raw_len + (-b.raw_len) % 16
-        if b.raw_len % 16 and b is not idx.blocks[-1]:
-            return False
-    return True
+    import numpy as np
+
+    t = idx.table
+    if len(t) == 0:
+        return True
+    if not (t["comp_len"] == t["raw_len"]).all():
+        return False
+    if (t["raw_len"][:-1] % 16).any():  # inner blocks must stay aligned
+        return False
+    acc = np.zeros(len(t), dtype=np.uint64)
+    np.cumsum(t["raw_len"][:-1], out=acc[1:])  # == aligned offs here
+    return bool((t["comp_off"] == acc).all())
+
+
+def _verify_crcs_device(out, idx, dev) -> None:
+    """GPU-side CRC compare: one kernel + one device equality, no
+    per-block python lists."""
+    import numpy as np
+    import torch
+
+    from shipyard_amd import ops
+
+    crcs = ops.crc32c_chunks(out[:idx.raw_size].contiguous(),
+                             chunk_size=idx.block_raw)
+    want = torch.from_numpy(
+        np.ascontiguousarray(idx.table["crc"]).view(np.int32)).to(dev)
+    eq = crcs.view(torch.int32) == want
+    if not bool(eq.all().item()):
+        bad = (~eq).nonzero().flatten()[:8].cpu().tolist()
+        raise ValueError(f"GPU CRC mismatch in blocks {bad}")
+
+
+def decode_device(d_comp, idx: ShardIndex, device, verify: bool = True):
+    """Decode an uploaded shard payload in HBM: batched gather-copy for
+    stored blocks + wave-cooperative LZ4 for compressed ones + chunked
+    CRC verify.  All host work is vectorized (numpy column arrays) —
+    O(1) python regardless of block count."""
+    import numpy as np
+    import torch
+
+    from shipyard_amd import ops
+
+    dev = device
+    if len(idx.table) == 0:
+        return torch.empty(0, dtype=torch.uint8, device=dev)
+    if _stored_contiguous(idx):
+        out = d_comp[:idx.raw_size]
+        if verify:
+            _verify_crcs_device(out, idx, dev)
+        return out
+
+    t = idx.table
+    raw_offs = idx.raw_offs()
+    stored = t["comp_len"] == t["raw_len"]
+    out = torch.empty(max(idx.raw_size, 1), dtype=torch.uint8, device=dev)
+
+    def to_dev(arr, dtype):
+        return torch.from_numpy(np.ascontiguousarray(arr)).to(dev).view(
+            dtype)
+
+    if stored.any():
+        ops.gather_copy(
+            d_comp, to_dev(t["comp_off"][stored].astype(np.int64),
+                           torch.int64),
+            out, to_dev(raw_offs[stored], torch.int64),
+            to_dev(t["raw_len"][stored].view(np.int32), torch.uint32))
+    lz4 = ~stored
+    if lz4.any():
+        status = ops.lz4_decode_blocks(
+            d_comp,
+            to_dev(t["comp_off"][lz4].astype(np.int64), torch.int64),
+            to_dev(t["comp_len"][lz4].view(np.int32), torch.uint32),
+            out,
+            to_dev(raw_offs[lz4], torch.int64),
+            to_dev(t["raw_len"][lz4].view(np.int32), torch.uint32),
+            raw_cap=idx.block_raw)
+        if not ops.lz4_all_ok(status):
+            raise ValueError(
+                f"GPU LZ4 decode failed: status={status.cpu().tolist()}")
+    if verify:
+        _verify_crcs_device(out, idx, dev)
+    return out[:idx.raw_size]
 
 
 def unpack_gpu(buf: bytes, device=None, verify: bool = True):
@@ -150,65 +272,12 @@ def unpack_gpu(buf: bytes, device=None, verify: bool = True):
     """
     import torch
 
-    from shipyard_amd import ops
-
     dev = device or torch.device("cuda", torch.cuda.current_device())
     idx = read_index(buf)
     payload = buf[idx.payload_off:]
     d_comp = torch.frombuffer(bytearray(payload), dtype=torch.uint8).to(
         dev, non_blocking=True)
-    out = torch.empty(max(idx.raw_size, 1), dtype=torch.uint8, device=dev)
-    if not idx.blocks:
-        return out[:0]
-
-    if _stored_contiguous(idx):
-        out = d_comp[:idx.raw_size]
-        if verify:
-            crcs = ops.crc32c_chunks(out.contiguous(),
-                                     chunk_size=idx.block_raw)
-            want = [b.crc32c for b in idx.blocks]
-            if [int(x) for x in crcs.tolist()] != want:
-                raise ValueError("GPU CRC mismatch in stored shard")
-        return out
-
-    lz4_blocks = [(i, b) for i, b in enumerate(idx.blocks) if not b.stored]
-    stored_blocks = [(i, b) for i, b in enumerate(idx.blocks) if b.stored]
-
-    # raw offsets are cumulative; precompute once
-    raw_offs = []
-    acc = 0
-    for b in idx.blocks:
-        raw_offs.append(acc)
-        acc += b.raw_len
-    for i, b in stored_blocks:
-        out[raw_offs[i]:raw_offs[i] + b.raw_len] = \
-            d_comp[b.comp_off:b.comp_off + b.comp_len]
-
-    if lz4_blocks:
-        mk64 = lambda v: torch.tensor(v, dtype=torch.int64, device=dev)
-        mk32 = lambda v: torch.tensor(v, dtype=torch.int64).to(
-            torch.uint32).to(dev)
-        status = ops.lz4_decode_blocks(
-            d_comp,
-            mk64([b.comp_off for _, b in lz4_blocks]),
-            mk32([b.comp_len for _, b in lz4_blocks]),
-            out,
-            mk64([raw_offs[i] for i, _ in lz4_blocks]),
-            mk32([b.raw_len for _, b in lz4_blocks]),
-            raw_cap=idx.block_raw)
-        if not ops.lz4_all_ok(status):
-            raise ValueError(
-                f"GPU LZ4 decode failed: status={status.cpu().tolist()}")
-
-    if verify:
-        # block_raw-aligned chunks == block boundaries (last may be short)
-        crcs = ops.crc32c_chunks(out[:idx.raw_size], chunk_size=idx.block_raw)
-        want = [b.crc32c for b in idx.blocks]
-        got = [int(x) for x in crcs.tolist()]
-        if got != want:
-            bad = [i for i, (a, c) in enumerate(zip(got, want)) if a != c]
-            raise ValueError(f"GPU CRC mismatch in blocks {bad[:8]}")
-    return out[:idx.raw_size]
+    return decode_device(d_comp, idx, dev, verify=verify)
 
 
 def pack_file(src: Path, dst: Path, block_raw: int = DEFAULT_BLOCK_RAW,

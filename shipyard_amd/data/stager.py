@@ -132,12 +132,13 @@ class ShardStager:
             else:
                 d_payload = self._upload(f, payload_total)
 
-        # parse table on host (tiny), decode + verify in HBM
+        # parse table on host (numpy, O(1) python), decode in HBM
+        import numpy as np
+
         idx = shardfmt.ShardIndex(
-            block_raw=block_raw, raw_size=raw_size,
-            blocks=[shardfmt.BlockEntry(*shardfmt.ENTRY.unpack_from(
-                table, i * shardfmt.ENTRY.size)) for i in range(n_blocks)],
-            payload_off=0)
+            block_raw=block_raw, raw_size=raw_size, payload_off=0,
+            table=np.frombuffer(table, dtype=shardfmt._entry_dt(),
+                                count=n_blocks))
         out = self._decode(d_payload, idx)
         sec = time.perf_counter() - t0
         res = StageResult(str(p), size, raw_size, sec, decoded=True,
@@ -149,62 +150,11 @@ class ShardStager:
     def _decode(self, d_payload, idx: shardfmt.ShardIndex):
         import torch
 
-        from shipyard_amd import ops
-
-        from shipyard_amd.data.shardfmt import _stored_contiguous
-
-        if _stored_contiguous(idx):
-            out = d_payload[:idx.raw_size]
-            if self.verify and idx.blocks:
-                with torch.cuda.stream(self.stream):
-                    crcs = ops.crc32c_chunks(out.contiguous(),
-                                             chunk_size=idx.block_raw)
-                self.stream.synchronize()
-                want = [b.crc32c for b in idx.blocks]
-                if [int(x) for x in crcs.tolist()] != want:
-                    raise ValueError("CRC mismatch in staged shard")
-            return out
-
-        out = torch.empty(max(idx.raw_size, 1), dtype=torch.uint8,
-                          device=self.device)
-        raw_offs = []
-        acc = 0
-        for b in idx.blocks:
-            raw_offs.append(acc)
-            acc += b.raw_len
-        lz4_blocks = [(i, b) for i, b in enumerate(idx.blocks)
-                      if not b.stored]
-        stored = [(i, b) for i, b in enumerate(idx.blocks) if b.stored]
         with torch.cuda.stream(self.stream):
-            for i, b in stored:
-                out[raw_offs[i]:raw_offs[i] + b.raw_len] = \
-                    d_payload[b.comp_off:b.comp_off + b.comp_len]
+            out = shardfmt.decode_device(d_payload, idx, self.device,
+                                         verify=self.verify)
         self.stream.synchronize()
-        if lz4_blocks:
-            mk64 = lambda v: torch.tensor(v, dtype=torch.int64,
-                                          device=self.device)
-            mk32 = lambda v: torch.tensor(v, dtype=torch.int64).to(
-                torch.uint32).to(self.device)
-            with torch.cuda.stream(self.stream):
-                status = ops.lz4_decode_blocks(
-                    d_payload, mk64([b.comp_off for _, b in lz4_blocks]),
-                    mk32([b.comp_len for _, b in lz4_blocks]), out,
-                    mk64([raw_offs[i] for i, _ in lz4_blocks]),
-                    mk32([b.raw_len for _, b in lz4_blocks]),
-                    raw_cap=idx.block_raw)
-            self.stream.synchronize()
-            if not ops.lz4_all_ok(status):
-                raise ValueError("GPU decode failed in stager")
-        if self.verify and idx.blocks:
-            with torch.cuda.stream(self.stream):
-                crcs = ops.crc32c_chunks(out[:idx.raw_size],
-                                         chunk_size=idx.block_raw)
-            self.stream.synchronize()
-            want = [b.crc32c for b in idx.blocks]
-            got = [int(x) for x in crcs.tolist()]
-            if got != want:
-                raise ValueError("CRC mismatch in staged shard")
-        return out[:idx.raw_size]
+        return out
 
     def stage_many(self, paths: List) -> Dict[str, StageResult]:
         out = {}

@@ -66,6 +66,11 @@ def _load() -> ctypes.CDLL:
     ]
     lib.sy_lz4_decode_blocks_pc.restype = ctypes.c_int
     lib.sy_lz4_decode_blocks_pc.argtypes = lib.sy_lz4_decode_blocks.argtypes
+    lib.sy_gather_copy.restype = ctypes.c_int
+    lib.sy_gather_copy.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_uint32, ctypes.c_void_p,
+    ]
     lib.sy_stage_file.restype = ctypes.c_int
     lib.sy_stage_file.argtypes = [
         ctypes.c_char_p, ctypes.c_void_p, ctypes.c_uint64, ctypes.c_uint64,
@@ -141,6 +146,23 @@ def crc32c_file_digest(data, chunk_size: int = 256 * 1024) -> int:
         acc = gf2.combine_raw(acc, int(r), clen)
         pos += clen
     return gf2.finish(acc, n)
+
+
+def gather_copy(src, src_off, dst, dst_off, lens):
+    """Batched device gather-copy (stored shard blocks): one workgroup
+    per block, uint4 body + byte tail.  src_off/dst_off int64 byte
+    offsets (src 16 B aligned by the format), lens uint32."""
+    lib = _load()
+    n = src_off.numel()
+    if n == 0:
+        return
+    rc = lib.sy_gather_copy(
+        ctypes.c_void_p(src.data_ptr()),
+        ctypes.c_void_p(src_off.data_ptr()),
+        ctypes.c_void_p(dst.data_ptr()),
+        ctypes.c_void_p(dst_off.data_ptr()),
+        ctypes.c_void_p(lens.data_ptr()), ctypes.c_uint32(n), _stream())
+    _check(rc, "sy_gather_copy")
 
 
 def lz4_decode_blocks(comp, in_off, in_len, out, out_off, out_len,

@@ -17,7 +17,8 @@ CSRC = HERE / "csrc"
 LIB = HERE / "libshipyardops.so"
 ARCH = os.environ.get("SHIPYARD_GPU_ARCH", "gfx950")
 
-SOURCES = ["crc32c.hip", "lz4_decode.hip", "sha256.hip", "stager.cpp"]
+SOURCES = ["crc32c.hip", "lz4_decode.hip", "sha256.hip",
+           "gather_copy.hip", "stager.cpp"]
 
 
 def hipcc() -> str:
