@@ -202,11 +202,13 @@ def _verify_crcs_device(out, idx, dev) -> None:
 
     crcs = ops.crc32c_chunks(out[:idx.raw_size].contiguous(),
                              chunk_size=idx.block_raw)
+    # crc32c_chunks finishes the GF(2) combine on the host and returns
+    # a CPU tensor; compare there (one vectorized op, no python lists)
     want = torch.from_numpy(
-        np.ascontiguousarray(idx.table["crc"]).view(np.int32)).to(dev)
-    eq = crcs.view(torch.int32) == want
+        np.ascontiguousarray(idx.table["crc"]).view(np.int32).copy())
+    eq = crcs.view(torch.int32).cpu() == want
     if not bool(eq.all().item()):
-        bad = (~eq).nonzero().flatten()[:8].cpu().tolist()
+        bad = (~eq).nonzero().flatten()[:8].tolist()
         raise ValueError(f"GPU CRC mismatch in blocks {bad}")
 
 
