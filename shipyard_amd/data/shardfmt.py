@@ -277,6 +277,8 @@ def unpack_gpu(buf: bytes, device=None, verify: bool = True):
     dev = device or torch.device("cuda", torch.cuda.current_device())
     idx = read_index(buf)
     payload = buf[idx.payload_off:]
+    if not payload:  # empty shard: frombuffer rejects 0-length buffers
+        return torch.empty(0, dtype=torch.uint8, device=dev)
     d_comp = torch.frombuffer(bytearray(payload), dtype=torch.uint8).to(
         dev, non_blocking=True)
     return decode_device(d_comp, idx, dev, verify=verify)
