@@ -211,6 +211,77 @@ def pool_stats(ctx, poolid, configdir, root, raw):
     ctx.emit(ctx.executor.pool_stats(pid))
 
 
+@pool.command("exists")
+@click.option("--poolid", required=True)
+@_common
+@pass_ctx
+def pool_exists(ctx, poolid, configdir, root, raw):
+    """Exit 0 with {"exists": true} iff the pool exists (reference
+    `pool exists`)."""
+    _apply(ctx, configdir, root, raw)
+    row = ctx.executor.store.query_one(
+        "SELECT id FROM pools WHERE id=?", (poolid,))
+    ctx.emit({"exists": row is not None})
+    if row is None:
+        raise SystemExit(1)
+
+
+@pool.command("ssh")
+@click.option("--poolid", required=True)
+@click.option("--node", "node_id", required=True)
+@click.option("--command", "cmd", default=None,
+              help="print the ssh invocation; with --command, include it")
+@_common
+@pass_ctx
+def pool_ssh(ctx, poolid, node_id, cmd, configdir, root, raw):
+    """SSH command for a multi-node pool's node (reference `pool ssh`)."""
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.utils import crypto
+
+    ps = ctx.executor.pool_settings_of(poolid)
+    node = next((n for n in ps.nodes if n.id == node_id), None)
+    if node is None:
+        raise click.ClickException(f"no node {node_id} in pool {poolid}")
+    out = crypto.ssh_command(node.host, cmd or "", username=node.ssh_user,
+                             private_key=node.ssh_private_key)
+    if not cmd:
+        out = out[:-1]
+    ctx.emit({"ssh": " ".join(out)})
+
+
+@pool.command("autoscale-enable")
+@click.option("--poolid", required=True)
+@_common
+@pass_ctx
+def pool_autoscale_enable(ctx, poolid, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    ctx.executor.store.execute("DELETE FROM kv WHERE key=?",
+                               (f"autoscale_disabled:{poolid}",))
+    ctx.emit({"autoscale": "enabled"})
+
+
+@pool.command("autoscale-disable")
+@click.option("--poolid", required=True)
+@_common
+@pass_ctx
+def pool_autoscale_disable(ctx, poolid, configdir, root, raw):
+    """Runtime kill switch for the daemon's autoscale loop (reference
+    `pool autoscale disable`)."""
+    _apply(ctx, configdir, root, raw)
+    ctx.executor.store.kv_set(f"autoscale_disabled:{poolid}", "1")
+    ctx.emit({"autoscale": "disabled"})
+
+
+@pool.command("autoscale-lastexec")
+@click.option("--poolid", required=True)
+@_common
+@pass_ctx
+def pool_autoscale_lastexec(ctx, poolid, configdir, root, raw):
+    _apply(ctx, configdir, root, raw)
+    v = ctx.executor.store.kv_get(f"autoscale_lastexec:{poolid}")
+    ctx.emit({"lastexec": float(v) if v else None})
+
+
 @pool.command("autoscale-evaluate")
 @click.option("--poolid")
 @_common
@@ -416,6 +487,23 @@ def tasks_term(ctx, jobid, taskid, configdir, root, raw):
 def tasks_list(ctx, jobid, configdir, root, raw):
     _apply(ctx, configdir, root, raw)
     ctx.emit(ctx.executor.tasks_list(jobid))
+
+
+@jobs_tasks.command("count")
+@click.option("--jobid", default=None, help="default: all jobs")
+@_common
+@pass_ctx
+def tasks_count(ctx, jobid, configdir, root, raw):
+    """Task counts by state (reference `jobs tasks count`)."""
+    _apply(ctx, configdir, root, raw)
+    q = "SELECT state, COUNT(*) n FROM tasks"
+    args = []
+    if jobid:
+        q += " WHERE job_id=?"
+        args.append(jobid)
+    q += " GROUP BY state"
+    ctx.emit({r["state"]: r["n"]
+              for r in ctx.executor.store.query(q, args)})
 
 
 # ---------------------------------------------------------------- data
@@ -713,6 +801,37 @@ def fed_process(ctx, configdir, root, raw):
     ctx.emit({"processed": n})
 
 
+@fed.command("list")
+@_common
+@pass_ctx
+def fed_list(ctx, configdir, root, raw):
+    """Federations + member pools (reference `fed list`)."""
+    _apply(ctx, configdir, root, raw)
+    from shipyard_amd.federation.scheduler import FederationProcessor
+
+    fp = FederationProcessor.from_config(
+        ctx.executor, ctx.conf(ConfigType.federation))
+    ctx.emit({fid: {"pools": f.pools,
+                    "force_unique_job_ids": f.force_unique_job_ids}
+              for fid, f in fp.federations.items()})
+
+
+@fed.command("jobs-list")
+@click.option("--federation-id", default=None)
+@_common
+@pass_ctx
+def fed_jobs_list(ctx, federation_id, configdir, root, raw):
+    """Queued/blocked federation actions (reference `fed jobs list`)."""
+    _apply(ctx, configdir, root, raw)
+    q = ("SELECT id, federation_id, action, state, attempts, "
+         "enqueued_at FROM fed_queue")
+    args = []
+    if federation_id:
+        q += " WHERE federation_id=?"
+        args.append(federation_id)
+    ctx.emit([dict(r) for r in ctx.executor.store.query(q, args)])
+
+
 @jobs.command("migrate")
 @click.option("--jobid", required=True)
 @click.option("--poolid", required=True, help="destination pool")
@@ -751,6 +870,33 @@ def nodes_list(ctx, poolid, configdir, root, raw):
         "SELECT slot_id, kind, device_id, dedicated, state FROM slots "
         "WHERE pool_id=? ORDER BY slot_id", (poolid,))
     ctx.emit([dict(r) for r in rows])
+
+
+@pool_nodes.command("ps")
+@click.option("--poolid", required=True)
+@_common
+@pass_ctx
+def nodes_ps(ctx, poolid, configdir, root, raw):
+    """Running tasks per slot/node (reference `pool nodes ps`)."""
+    _apply(ctx, configdir, root, raw)
+    import json as _json
+
+    rows = ctx.executor.store.query(
+        "SELECT t.job_id, t.id, t.slots_json, t.start_time FROM tasks t "
+        "JOIN jobs j ON t.job_id=j.id WHERE j.pool_id=? AND "
+        "t.state='running'", (poolid,))
+    slot_node = {r["slot_id"]: r["node_id"] for r in
+                 ctx.executor.store.query(
+                     "SELECT slot_id, node_id FROM slots WHERE pool_id=?",
+                     (poolid,))}
+    out = []
+    for r in rows:
+        slots = _json.loads(r["slots_json"] or "[]")
+        out.append({"job": r["job_id"], "task": r["id"], "slots": slots,
+                    "nodes": sorted({slot_node.get(s, "?")
+                                     for s in slots}),
+                    "start_time": r["start_time"]})
+    ctx.emit(out)
 
 
 @pool_nodes.command("hosts")

@@ -109,7 +109,12 @@ class AutoscaleController:
         if now_ts - self.last_eval < \
                 self.settings.evaluation_interval.total_seconds():
             return None
+        # runtime kill switch (reference `pool autoscale disable`)
+        if self.ex.store.kv_get(f"autoscale_disabled:{self.pool_id}"):
+            return None
         self.last_eval = now_ts
+        self.ex.store.kv_set(f"autoscale_lastexec:{self.pool_id}",
+                             str(now_ts))
         row = self.ex.store.query_one(
             "SELECT COUNT(*) n FROM tasks t JOIN jobs j ON t.job_id=j.id "
             "WHERE j.pool_id=? AND t.state IN ('ready','running')",

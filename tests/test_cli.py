@@ -199,3 +199,105 @@ def test_pool_nodes_list(configdir, tmp_path):
     assert r.exit_code == 0
     rows = json.loads(r.output)
     assert len(rows) == 2 and rows[0]["state"] == "idle"
+
+
+class TestParityVerbs:
+    def _root(self, tmp_path):
+        return ["--root", str(tmp_path / "root")]
+
+    def test_pool_exists_and_count_and_ps(self, tmp_path):
+        from click.testing import CliRunner
+
+        from shipyard_amd.cli import cli
+
+        r = CliRunner()
+        rt = self._root(tmp_path)
+        res = r.invoke(cli, ["pool", "exists", "--poolid", "nope", *rt])
+        assert res.exit_code == 1 and '"exists": false' in res.output
+        # make a pool + long task, then ps shows it
+        import yaml
+
+        (tmp_path / "cfg").mkdir()
+        (tmp_path / "cfg" / "pool.yaml").write_text(yaml.safe_dump(
+            {"pool_specification": {
+                "id": "pv", "cpu_slots": 1, "gpus": {"dedicated": 0},
+                "node_configuration": {"rocm": {"verify": False}}}}))
+        (tmp_path / "cfg" / "jobs.yaml").write_text(yaml.safe_dump(
+            {"job_specifications": [
+                {"id": "jv", "tasks": [{"id": "t",
+                                        "command": "sleep 30"}]}]}))
+        cfg = ["--configdir", str(tmp_path / "cfg")]
+        assert r.invoke(cli, ["pool", "add", *cfg, *rt]).exit_code == 0
+        res = r.invoke(cli, ["pool", "exists", "--poolid", "pv",
+                             *cfg, *rt])
+        assert res.exit_code == 0 and '"exists": true' in res.output
+        assert r.invoke(cli, ["jobs", "add", *cfg, *rt]).exit_code == 0
+        res = r.invoke(cli, ["jobs", "tasks", "count", *cfg, *rt])
+        assert res.exit_code == 0 and "pending" in res.output
+        # drive the task to running in-process, then ps sees it
+        import time
+
+        from shipyard_amd.executor import LocalExecutor
+
+        ex = LocalExecutor(tmp_path / "root", detect_gpus=False)
+        deadline = time.monotonic() + 30
+        while time.monotonic() < deadline:
+            ex.schedule_once()
+            if any(t["state"] == "running"
+                   for t in ex.tasks_list("jv")):
+                break
+            time.sleep(0.05)
+        res = r.invoke(cli, ["jobs", "tasks", "count", *cfg, *rt])
+        assert res.exit_code == 0 and "running" in res.output
+        res = r.invoke(cli, ["pool", "nodes", "ps", "--poolid", "pv",
+                             *cfg, *rt])
+        assert res.exit_code == 0 and '"task": "t"' in res.output
+        ex.job_terminate("jv")
+        ex.store.close()
+
+    def test_pool_ssh_synthesis(self, tmp_path):
+        import yaml
+        from click.testing import CliRunner
+
+        from shipyard_amd.cli import cli
+
+        r = CliRunner()
+        (tmp_path / "cfg").mkdir()
+        (tmp_path / "cfg" / "pool.yaml").write_text(yaml.safe_dump(
+            {"pool_specification": {
+                "id": "mn",
+                "nodes": [{"id": "a", "host": "10.1.2.3",
+                           "ssh": {"username": "u",
+                                   "private_key": "/k"}}],
+                "node_configuration": {"rocm": {"verify": False}}}}))
+        rt = self._root(tmp_path)
+        cfg = ["--configdir", str(tmp_path / "cfg")]
+        assert r.invoke(cli, ["pool", "add", *cfg, *rt]).exit_code == 0
+        res = r.invoke(cli, ["pool", "ssh", "--poolid", "mn", "--node",
+                             "a", *cfg, *rt])
+        assert res.exit_code == 0 and "u@10.1.2.3" in res.output
+
+    def test_autoscale_enable_disable_lastexec(self, tmp_path):
+        import yaml
+        from click.testing import CliRunner
+
+        from shipyard_amd.cli import cli
+
+        r = CliRunner()
+        (tmp_path / "cfg").mkdir()
+        (tmp_path / "cfg" / "pool.yaml").write_text(yaml.safe_dump(
+            {"pool_specification": {
+                "id": "au", "cpu_slots": 1, "gpus": {"dedicated": 0},
+                "node_configuration": {"rocm": {"verify": False}}}}))
+        rt = self._root(tmp_path)
+        cfg = ["--configdir", str(tmp_path / "cfg")]
+        assert r.invoke(cli, ["pool", "add", *cfg, *rt]).exit_code == 0
+        res = r.invoke(cli, ["pool", "autoscale-disable", "--poolid",
+                             "au", *cfg, *rt])
+        assert res.exit_code == 0 and "disabled" in res.output
+        res = r.invoke(cli, ["pool", "autoscale-lastexec", "--poolid",
+                             "au", *cfg, *rt])
+        assert res.exit_code == 0 and "null" in res.output
+        res = r.invoke(cli, ["pool", "autoscale-enable", "--poolid",
+                             "au", *cfg, *rt])
+        assert res.exit_code == 0 and "enabled" in res.output
