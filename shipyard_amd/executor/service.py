@@ -278,14 +278,20 @@ class LocalExecutor:
             raise ExecutorError(f"no pool {pool_id}")
         conf = json.loads(row["spec_json"])
         ps = cfg.pool_settings(conf)
+        if ps.nodes:
+            raise ExecutorError(
+                "multi-node pools size by their node list; edit nodes: "
+                "and re-add the pool (per-node resize is a planned "
+                "follow-up)")
         new_ded = dedicated if dedicated is not None else ps.gpus_dedicated
         new_low = (low_priority if low_priority is not None
                    else ps.gpus_low_priority)
         host = self.host_gpu_count()
         if host and new_ded + new_low > host * ps.max_tasks_per_gpu:
             raise ExecutorError("resize exceeds host GPU capacity")
-        conf["pool_specification"]["gpus"]["dedicated"] = new_ded
-        conf["pool_specification"]["gpus"]["low_priority"] = new_low
+        gpus = conf["pool_specification"].setdefault("gpus", {})
+        gpus["dedicated"] = new_ded
+        gpus["low_priority"] = new_low
         with self.store.transaction() as conn:
             conn.execute("DELETE FROM slots WHERE pool_id=? "
                          "AND state='idle'", (pool_id,))

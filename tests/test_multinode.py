@@ -387,3 +387,24 @@ class TestMultiNodeStress:
             "SELECT COUNT(*) n FROM assignments")["n"] == 0
         assert mx.store.query_one(
             "SELECT COUNT(*) n FROM slots WHERE state!='idle'")["n"] == 0
+
+
+def test_multinode_pool_resize_rejected(tmp_path):
+    ex = LocalExecutor(tmp_path / "root", detect_gpus=False)
+    _mk_pool(ex)
+    from shipyard_amd.executor import ExecutorError
+
+    with pytest.raises(ExecutorError, match="node list"):
+        ex.pool_resize("mp", dedicated=4)
+    ex.store.close()
+
+
+def test_gpusless_pool_resize(tmp_path):
+    """Pools created without a gpus: key (legal since nodes: landed)
+    still resize."""
+    ex = LocalExecutor(tmp_path / "root", detect_gpus=False)
+    ex.pool_add({"pool_specification": {
+        "id": "c", "cpu_slots": 1,
+        "node_configuration": {"rocm": {"verify": False}}}})
+    ex.pool_resize("c", dedicated=0)  # no KeyError
+    ex.store.close()
