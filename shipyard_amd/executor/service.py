@@ -612,6 +612,16 @@ class LocalExecutor:
                 self.store.execute(
                     "UPDATE tasks SET state='ready', slots_json=NULL "
                     "WHERE job_id=? AND id=?", (jid, tid))
+        # remote (agent-executed) tasks: cancel their windows, free the
+        # slots, and return them to ready the same way
+        remote = self.store.query(
+            "SELECT id FROM tasks WHERE job_id=? AND state='running'",
+            (job_id,))
+        if remote:
+            self._cancel_assignments(job_id)
+            self.store.execute(
+                "UPDATE tasks SET state='ready', slots_json=NULL "
+                "WHERE job_id=? AND state='running'", (job_id,))
         self.job_disable(job_id)
 
     def job_migrate(self, job_id: str, dest_pool: str) -> None:

@@ -408,3 +408,51 @@ def test_gpusless_pool_resize(tmp_path):
         "node_configuration": {"rocm": {"verify": False}}}})
     ex.pool_resize("c", dedicated=0)  # no KeyError
     ex.store.close()
+
+
+def test_migrate_remote_job_between_pools(tmp_path):
+    """disable -> requeue -> patch pool works when the running task is
+    on a node agent (reference convoy/batch.py:1855 job migration)."""
+    import time as _time
+
+    ex = LocalExecutor(tmp_path / "root", detect_gpus=False)
+    try:
+        _mk_pool(ex, cpu_per_node=1, n_nodes=1)
+        ex.pool_add({"pool_specification": {
+            "id": "dest", "cpu_slots": 1, "gpus": {"dedicated": 0},
+            "node_configuration": {"rocm": {"verify": False}}}})
+        ex.start_local_agents("mp")
+        ex.jobs_add({"job_specifications": [{
+            "id": "mig", "tasks": [{"id": "t",
+                                    "command": "echo done; sleep 600"}],
+        }]}, "mp")
+        deadline = _time.monotonic() + 30
+        while _time.monotonic() < deadline:
+            ex.schedule_once()
+            if ex.store.query_one(
+                    "SELECT 1 FROM assignments WHERE state='running'"):
+                break
+            _time.sleep(0.05)
+        ex.job_migrate("mig", "dest")
+        # after migration the task restarts on the local dest pool; use
+        # a short command this time by replacing via requeue semantics:
+        # the same spec reruns, so just terminate after it starts
+        deadline = _time.monotonic() + 30
+        started = False
+        while _time.monotonic() < deadline:
+            ex.schedule_once()
+            t = ex.tasks_list("mig")[0]
+            if t["state"] == "running" and ("mig", "t") in ex._handles:
+                started = True
+                break
+            _time.sleep(0.05)
+        assert started, "task did not restart on dest pool"
+        assert ex.job_pool("mig") == "dest"
+        ex.job_terminate("mig")
+        # source pool slots drained
+        assert ex.store.query_one(
+            "SELECT COUNT(*) n FROM slots WHERE pool_id='mp' AND "
+            "state!='idle'")["n"] == 0
+    finally:
+        ex.stop_local_agents()
+        ex.store.close()
