@@ -456,3 +456,34 @@ def test_migrate_remote_job_between_pools(tmp_path):
     finally:
         ex.stop_local_agents()
         ex.store.close()
+
+
+def test_data_flow_through_agents(tmp_path):
+    """input_data/output_data across the shared root when the task runs
+    on a node agent, with the daemon scheduler thread driving."""
+    ex = LocalExecutor(tmp_path / "root", detect_gpus=False)
+    try:
+        _mk_pool(ex, cpu_per_node=1, n_nodes=2)
+        ex.start_local_agents("mp")
+        ex.stores["default"].upload_bytes("seed/in.txt", b"agent-data\n")
+        ex.start_scheduler(poll=0.02)
+        ex.jobs_add({"job_specifications": [{
+            "id": "dj",
+            "tasks": [{
+                "id": "t",
+                "command": "cat in.txt > out.txt; echo via-agent >> out.txt",
+                "input_data": {"local_storage": [
+                    {"remote_path": "seed"}]},
+                "output_data": {"local_storage": [
+                    {"remote_path": "res", "include": ["out.txt"]}]},
+            }],
+        }]}, "mp")
+        ex.wait_for_job("dj", timeout=60)
+        t = ex.tasks_list("dj")[0]
+        assert t["state"] == "completed", dict(t)
+        got = ex.stores["default"].download_bytes("res/out.txt")
+        assert b"agent-data" in got and b"via-agent" in got
+    finally:
+        ex.stop_scheduler()
+        ex.stop_local_agents()
+        ex.store.close()
