@@ -54,3 +54,34 @@ def dump(store: Store, source_prefix: str = "") -> str:
         lines.append(f"{e['ts']:.3f} {e['source']} {e['event']}{delta} "
                      f"{json.dumps(extra) if extra else ''}".rstrip())
     return "\n".join(lines)
+
+
+def chart(store: Store, width: int = 72) -> str:
+    """ASCII Gantt of per-source spans (the cascade/graph.py gnuplot
+    chart, terminal edition): one bar per source from its first to its
+    last perf event, markers at each event."""
+    evs = events(store)
+    if not evs:
+        return "(no perf events)"
+    t0 = min(e["ts"] for e in evs)
+    t1 = max(e["ts"] for e in evs)
+    span = max(t1 - t0, 1e-9)
+    per_src = {}
+    for e in evs:
+        per_src.setdefault(e["source"], []).append(e["ts"])
+    name_w = min(max(len(s) for s in per_src), 28)
+    lines = [f"{'source':<{name_w}} |{'-' * width}| "
+             f"span {span:.3f}s"]
+    for src in sorted(per_src, key=lambda s: min(per_src[s])):
+        ts = per_src[src]
+        lo = int((min(ts) - t0) / span * (width - 1))
+        hi = int((max(ts) - t0) / span * (width - 1))
+        row = [" "] * width
+        for i in range(lo, hi + 1):
+            row[i] = "="
+        for t in ts:
+            row[int((t - t0) / span * (width - 1))] = "*"
+        label = src if len(src) <= name_w else src[:name_w - 1] + "~"
+        lines.append(f"{label:<{name_w}} |{''.join(row)}| "
+                     f"{max(ts) - min(ts):.3f}s")
+    return "\n".join(lines)

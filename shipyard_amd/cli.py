@@ -700,13 +700,18 @@ def diag_prune(ctx, older_than_hours, configdir, root, raw):
 
 
 @diag.command("timeline")
+@click.option("--chart", is_flag=True,
+              help="ASCII Gantt (reference cascade/graph.py analogue)")
 @_common
 @pass_ctx
-def diag_timeline(ctx, configdir, root, raw):
+def diag_timeline(ctx, chart, configdir, root, raw):
     _apply(ctx, configdir, root, raw)
     from shipyard_amd.cascade import perf
 
-    ctx.emit(perf.timeline(ctx.executor.store))
+    if chart:
+        click.echo(perf.chart(ctx.executor.store))
+    else:
+        ctx.emit(perf.timeline(ctx.executor.store))
 
 
 # ---------------------------------------------------------------- monitor

@@ -127,3 +127,27 @@ def test_on_demand_image_staging(tmp_path):
     assert (ex.pool_root("odp") / "images" / "lazy-img" /
             ".complete").exists()
     ex.store.close()
+
+
+def test_perf_ascii_chart(tmp_path):
+    """The cascade/graph.py analogue renders a Gantt with one bar per
+    source and star markers at events."""
+    from shipyard_amd.cascade import perf
+    from shipyard_amd.executor.store import Store
+
+    st = Store(tmp_path / "s.db")
+    assert "(no perf events)" in perf.chart(st)
+    base = 1000.0
+    for src, offs in (("pool:a", [0.0, 1.0, 4.0]),
+                      ("pool:b", [2.0, 3.0])):
+        for o in offs:
+            st.execute("INSERT INTO perf (ts, source, event) "
+                       "VALUES (?,?,?)", (base + o, src, "e"))
+    out = perf.chart(st, width=40)
+    lines = out.splitlines()
+    assert lines[0].startswith("source")
+    assert lines[1].startswith("pool:a") and lines[1].count("*") == 3
+    assert lines[2].startswith("pool:b") and lines[2].count("*") == 2
+    # pool:a spans the whole window, pool:b sits inside it
+    assert lines[1].index("*") < lines[2].index("*")
+    st.close()
