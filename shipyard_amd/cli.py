@@ -699,6 +699,24 @@ def diag_prune(ctx, older_than_hours, configdir, root, raw):
     ctx.emit({"pruned": n})
 
 
+@diag.command("config")
+@_common
+@pass_ctx
+def diag_config(ctx, configdir, root, raw):
+    """Print every resolved + validated config family (the reference's
+    global `--show-config` flag as a verb)."""
+    _apply(ctx, configdir, root, raw)
+    out = {}
+    for ct in ConfigType:
+        try:
+            doc = ctx.conf(ct)
+        except Exception:
+            continue
+        if doc:
+            out[ct.name] = doc
+    ctx.emit(out)
+
+
 @diag.command("timeline")
 @click.option("--chart", is_flag=True,
               help="ASCII Gantt (reference cascade/graph.py analogue)")
