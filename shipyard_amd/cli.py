@@ -895,6 +895,37 @@ def nodes_list(ctx, poolid, configdir, root, raw):
     ctx.emit([dict(r) for r in rows])
 
 
+@pool_nodes.command("add")
+@click.option("--poolid", required=True)
+@click.option("--spec", required=True,
+              help="node spec as JSON, e.g. "
+                   "'{\"id\":\"c\",\"host\":\"10.0.0.6\","
+                   "\"gpus\":{\"dedicated\":8}}'")
+@_common
+@pass_ctx
+def nodes_add(ctx, poolid, spec, configdir, root, raw):
+    """Grow a multi-node pool by one node."""
+    _apply(ctx, configdir, root, raw)
+    import json as _json
+
+    ctx.executor.node_add(poolid, _json.loads(spec))
+    ctx.emit(ctx.executor.nodes_list(poolid))
+
+
+@pool_nodes.command("del")
+@click.option("--poolid", required=True)
+@click.option("--node", "node_id", required=True)
+@click.option("--force", is_flag=True)
+@_common
+@pass_ctx
+def nodes_del(ctx, poolid, node_id, force, configdir, root, raw):
+    """Remove a node from a multi-node pool (its agent exits on the
+    next heartbeat)."""
+    _apply(ctx, configdir, root, raw)
+    ctx.executor.node_remove(poolid, node_id, force=force)
+    ctx.emit(ctx.executor.nodes_list(poolid))
+
+
 @pool_nodes.command("ps")
 @click.option("--poolid", required=True)
 @_common

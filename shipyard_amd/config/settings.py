@@ -101,19 +101,23 @@ class PoolSettings:
     nodes: Tuple["NodeSettings", ...] = ()
 
 
+def node_settings(n: Dict[str, Any]) -> NodeSettings:
+    """One `nodes:` entry of a multi-node pool."""
+    return NodeSettings(
+        id=n["id"],
+        host=n.get("host", "127.0.0.1"),
+        gpus_dedicated=_get(n, "gpus", "dedicated", default=0),
+        cpu_slots=n.get("cpu_slots", 0),
+        device_ids=_get(n, "gpus", "device_ids"),
+        ssh_user=_get(n, "ssh", "username"),
+        ssh_private_key=_get(n, "ssh", "private_key"),
+    )
+
+
 def pool_settings(conf: Dict[str, Any]) -> PoolSettings:
     """reference convoy/settings.py:1277 `pool_settings`"""
     p = conf["pool_specification"]
-    nodes = tuple(
-        NodeSettings(
-            id=n["id"],
-            host=n.get("host", "127.0.0.1"),
-            gpus_dedicated=_get(n, "gpus", "dedicated", default=0),
-            cpu_slots=n.get("cpu_slots", 0),
-            device_ids=_get(n, "gpus", "device_ids"),
-            ssh_user=_get(n, "ssh", "username"),
-            ssh_private_key=_get(n, "ssh", "private_key"),
-        ) for n in (p.get("nodes") or []))
+    nodes = tuple(node_settings(n) for n in (p.get("nodes") or []))
     scen = _get(p, "autoscale", "scenario")
     scenario = None
     if scen:

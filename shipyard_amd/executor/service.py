@@ -280,9 +280,8 @@ class LocalExecutor:
         ps = cfg.pool_settings(conf)
         if ps.nodes:
             raise ExecutorError(
-                "multi-node pools size by their node list; edit nodes: "
-                "and re-add the pool (per-node resize is a planned "
-                "follow-up)")
+                "multi-node pools size by their node list: use "
+                "`pool nodes add/del` (node_add/node_remove)")
         new_ded = dedicated if dedicated is not None else ps.gpus_dedicated
         new_low = (low_priority if low_priority is not None
                    else ps.gpus_low_priority)
@@ -658,6 +657,95 @@ class LocalExecutor:
         return [dict(r) for r in self.store.query(
             "SELECT node_id, host, state, heartbeat, agent_pid FROM nodes "
             "WHERE pool_id=? ORDER BY node_id", (pool_id,))]
+
+    def node_add(self, pool_id: str, node_spec: Dict[str, Any]) -> None:
+        """Grow a multi-node pool by one node (the multi-node answer to
+        `pool resize` up; reference adds VMs via resize_pool,
+        convoy/batch.py:1372)."""
+        row = self.store.query_one("SELECT spec_json FROM pools WHERE "
+                                   "id=?", (pool_id,))
+        if row is None:
+            raise ExecutorError(f"no pool {pool_id}")
+        conf = json.loads(row["spec_json"])
+        p = conf["pool_specification"]
+        if not p.get("nodes"):
+            raise ExecutorError(f"pool {pool_id} is not multi-node")
+        node = cfg.node_settings(node_spec)
+        if any(n["id"] == node.id for n in p["nodes"]):
+            raise ExecutorError(f"node {node.id} already in {pool_id}")
+        p["nodes"].append(node_spec)
+        ps = cfg.pool_settings(conf)
+        top = self.store.query_one(
+            "SELECT COALESCE(MAX(slot_id), -1) m FROM slots WHERE "
+            "pool_id=?", (pool_id,))["m"]
+        rows = []
+        slot = top + 1
+        devs = node.device_ids or list(range(node.gpus_dedicated))
+        for i in range(node.gpus_dedicated):
+            for _ in range(ps.max_tasks_per_gpu):
+                rows.append((pool_id, slot, "gpu", devs[i], 1, "idle",
+                             node.id))
+                slot += 1
+        for _ in range(node.cpu_slots):
+            rows.append((pool_id, slot, "cpu", None, 1, "idle", node.id))
+            slot += 1
+        with self.store.transaction() as conn:
+            conn.execute(
+                "UPDATE pools SET spec_json=?, gpus_dedicated=?, "
+                "cpu_slots=? WHERE id=?",
+                (json.dumps(conf), ps.gpus_dedicated, ps.cpu_slots,
+                 pool_id))
+            conn.execute("INSERT INTO nodes (pool_id, node_id, host) "
+                         "VALUES (?,?,?)", (pool_id, node.id, node.host))
+            conn.executemany(
+                "INSERT INTO slots (pool_id, slot_id, kind, device_id, "
+                "dedicated, state, node_id) VALUES (?,?,?,?,?,?,?)", rows)
+        self.store.add_event(f"pool:{pool_id}", "node-added",
+                             {"node": node.id, "host": node.host})
+
+    def node_remove(self, pool_id: str, node_id: str,
+                    force: bool = False) -> None:
+        """Shrink a multi-node pool: refuse while the node runs work
+        unless force (then its tasks are terminated).  The node's agent
+        exits on its next heartbeat (row gone)."""
+        busy = self.store.query(
+            "SELECT DISTINCT t.job_id, t.id FROM tasks t JOIN jobs j ON "
+            "t.job_id=j.id WHERE j.pool_id=? AND t.state='running' AND "
+            "EXISTS (SELECT 1 FROM slots s WHERE s.pool_id=? AND "
+            "s.node_id=? AND s.state='busy' AND s.slot_id IN (SELECT "
+            "value FROM json_each(t.slots_json)))",
+            (pool_id, pool_id, node_id))
+        if busy and not force:
+            raise ExecutorError(
+                f"node {node_id} runs {len(busy)} task(s); use force")
+        for b in busy:
+            self.task_terminate(b["job_id"], b["id"])
+        row = self.store.query_one("SELECT spec_json FROM pools WHERE "
+                                   "id=?", (pool_id,))
+        if row is None:
+            raise ExecutorError(f"no pool {pool_id}")
+        conf = json.loads(row["spec_json"])
+        p = conf["pool_specification"]
+        p["nodes"] = [n for n in (p.get("nodes") or [])
+                      if n["id"] != node_id]
+        ps = cfg.pool_settings(conf)
+        with self.store.transaction() as conn:
+            conn.execute("DELETE FROM slots WHERE pool_id=? AND "
+                         "node_id=?", (pool_id, node_id))
+            conn.execute("DELETE FROM nodes WHERE pool_id=? AND "
+                         "node_id=?", (pool_id, node_id))
+            conn.execute(
+                "UPDATE assignments SET state='done', rc=-15, "
+                "updated_at=? WHERE pool_id=? AND node_id=? AND "
+                "state IN ('queued','running','cancelling')",
+                (time.time(), pool_id, node_id))
+            conn.execute(
+                "UPDATE pools SET spec_json=?, gpus_dedicated=?, "
+                "cpu_slots=? WHERE id=?",
+                (json.dumps(conf), ps.gpus_dedicated, ps.cpu_slots,
+                 pool_id))
+        self.store.add_event(f"pool:{pool_id}", "node-removed",
+                             {"node": node_id})
 
     def agent_command(self, pool_id: str,
                       node: cfg.NodeSettings) -> List[str]:

@@ -487,3 +487,55 @@ def test_data_flow_through_agents(tmp_path):
         ex.stop_scheduler()
         ex.stop_local_agents()
         ex.store.close()
+
+
+class TestNodeMembership:
+    def test_node_add_and_remove(self, mx):
+        _mk_pool(mx, cpu_per_node=1, n_nodes=1)
+        mx.node_add("mp", {"id": "n9", "host": "127.0.0.1",
+                           "cpu_slots": 2})
+        assert [n["node_id"] for n in mx.nodes_list("mp")] == ["n0", "n9"]
+        assert mx.store.query_one(
+            "SELECT COUNT(*) n FROM slots WHERE pool_id='mp'")["n"] == 3
+        assert mx.pool_settings_of("mp").cpu_slots == 3
+        # duplicate rejected
+        from shipyard_amd.executor import ExecutorError
+
+        with pytest.raises(ExecutorError, match="already"):
+            mx.node_add("mp", {"id": "n9"})
+        mx.node_remove("mp", "n9")
+        assert [n["node_id"] for n in mx.nodes_list("mp")] == ["n0"]
+        assert mx.store.query_one(
+            "SELECT COUNT(*) n FROM slots WHERE pool_id='mp'")["n"] == 1
+
+    def test_node_remove_busy_needs_force_and_agent_exits(self, mx):
+        _mk_pool(mx, cpu_per_node=1, n_nodes=2)
+        procs = mx.start_local_agents("mp")
+        mx.jobs_add({"job_specifications": [{
+            "id": "nj", "tasks": [
+                {"id": f"t{i}", "command": "sleep 600"} for i in range(2)],
+        }]}, "mp")
+        deadline = time.monotonic() + 30
+        while time.monotonic() < deadline:
+            mx.schedule_once()
+            n = mx.store.query_one(
+                "SELECT COUNT(*) c FROM assignments WHERE "
+                "state='running'")["c"]
+            if n == 2:
+                break
+            time.sleep(0.05)
+        from shipyard_amd.executor import ExecutorError
+
+        with pytest.raises(ExecutorError, match="force"):
+            mx.node_remove("mp", "n1")
+        mx.node_remove("mp", "n1", force=True)
+        assert [n["node_id"] for n in mx.nodes_list("mp")] == ["n0"]
+        # n1's agent exits once its row is gone
+        deadline = time.monotonic() + 30
+        while time.monotonic() < deadline:
+            if procs[1].poll() is not None:
+                break
+            time.sleep(0.1)
+        assert procs[1].poll() is not None, "removed node's agent lived on"
+        # n0's task is still running; terminate to clean up
+        mx.job_terminate("nj")
