@@ -90,6 +90,23 @@ class SlurmAdapter:
                 if row is None:
                     continue
                 maxn = pconf.get("max_compute_nodes", 8)
+                if self.ex.pool_settings_of(pool_id).nodes:
+                    # multi-node pool: the Slurm host JOINS as a pool
+                    # node (the analogue of an elastic VM joining the
+                    # Batch pool, reference slurm/slurm.py:721)
+                    if len(self.ex.nodes_list(pool_id)) >= maxn:
+                        continue
+                    self.ex.node_add(pool_id, {
+                        "id": host, "host": host,
+                        "gpus": {"dedicated":
+                                 pconf.get("gpus_per_node", 8)}})
+                    assigned[host] = {"pool": pool_id, "type": "node",
+                                      "at": time.time()}
+                    done.append(host)
+                    self.ex.store.add_event(f"slurm:{host}", "resumed",
+                                            {"pool": pool_id,
+                                             "as": "node"})
+                    break
                 node_type = pconf.get("compute_node_type", "dedicated")
                 cur = row["gpus_dedicated"] if node_type == "dedicated" \
                     else row["gpus_low_priority"]
@@ -121,7 +138,9 @@ class SlurmAdapter:
                 "WHERE id=?", (rec["pool"],))
             if row is None:
                 continue
-            if rec["type"] == "dedicated":
+            if rec["type"] == "node":
+                self.ex.node_remove(rec["pool"], host, force=True)
+            elif rec["type"] == "dedicated":
                 self.ex.pool_resize(rec["pool"], dedicated=max(
                     row["gpus_dedicated"] - 1, 0))
             else:

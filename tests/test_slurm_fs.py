@@ -116,3 +116,38 @@ def test_shared_volume_storage_cluster_in_task(tmp_path):
     assert ex.tasks_list("jv")[0]["state"] == "completed"
     assert (mnt / "out.txt").read_text().strip() == "vol-ok"
     ex.store.close()
+
+
+def test_slurm_resume_joins_multinode_pool(tmp_path):
+    """Elastic Slurm over a multi-node pool: a resumed host joins as a
+    pool node; suspend removes it (reference slurm/slurm.py:721
+    add_nodes_to_pool analogue)."""
+    from shipyard_amd.executor import LocalExecutor
+    from shipyard_amd.slurm_elastic.adapter import SlurmAdapter
+
+    ex = LocalExecutor(tmp_path / "root", detect_gpus=False)
+    ex.pool_add({"pool_specification": {
+        "id": "mnp",
+        "nodes": [{"id": "seed", "host": "127.0.0.1", "cpu_slots": 1}],
+        "node_configuration": {"rocm": {"verify": False}}}})
+    conf = {"slurm": {
+        "cluster_id": "cl",
+        "elastic_partitions": {
+            "gpu": {"batch_pools": {"mnp": {
+                "max_compute_nodes": 3, "gpus_per_node": 2}}},
+        }}}
+    ad = SlurmAdapter(ex, conf)
+    done = ad.resume("cl-gpu-[1-2]")
+    assert done == ["cl-gpu-1", "cl-gpu-2"]
+    names = [n["node_id"] for n in ex.nodes_list("mnp")]
+    assert set(names) == {"seed", "cl-gpu-1", "cl-gpu-2"}
+    # each joined node contributed 2 gpu slots
+    assert ex.store.query_one(
+        "SELECT COUNT(*) n FROM slots WHERE pool_id='mnp' AND "
+        "kind='gpu'")["n"] == 4
+    # cap respected
+    assert ad.resume("cl-gpu-3") == []
+    assert ad.suspend("cl-gpu-1") == ["cl-gpu-1"]
+    names = [n["node_id"] for n in ex.nodes_list("mnp")]
+    assert "cl-gpu-1" not in names and "cl-gpu-2" in names
+    ex.store.close()
