@@ -134,6 +134,15 @@ def collect_executor_metrics(store) -> Dict[str, float]:
         "SELECT COUNT(*) n FROM fed_queue WHERE state IN "
         "('queued','blocked')")
     out["fed_queue_depth"] = row["n"]
+    # multi-node pools: agent liveness + assignment backlog
+    for r in store.query(
+            "SELECT state, COUNT(*) n FROM nodes GROUP BY state"):
+        out[f"nodes_{r['state']}"] = r["n"]
+    row = store.query_one(
+        "SELECT COUNT(*) n FROM assignments WHERE state IN "
+        "('queued','running')")
+    if row["n"]:
+        out["assignments_active"] = row["n"]
     return out
 
 
