@@ -39,7 +39,10 @@ class ObjectStore:
         dst.parent.mkdir(parents=True, exist_ok=True)
         if pack:
             dst = dst.with_suffix(dst.suffix + ".syshard")
-            dst.write_bytes(shardfmt.pack(data))
+            # parallel block compression above 4 MiB (pure-python LZ4
+            # is ~12 MB/s per core)
+            workers = 0 if len(data) > (4 << 20) else None
+            dst.write_bytes(shardfmt.pack(data, workers=workers))
         else:
             dst.write_bytes(data)
         if manifest:

@@ -195,3 +195,21 @@ class TestObjectStoreOps:
 
         with _pytest.raises(FileNotFoundError):
             st.download_bytes("nope.bin")
+
+
+class TestParallelPack:
+    def test_parallel_pack_identical_to_serial(self):
+        import os
+        import random
+
+        random.seed(9)
+        data = (b"layer content " * 40000 + os.urandom(200_000)
+                + bytes(random.choices(b"qrstuv", k=150_000)))
+        serial = shardfmt.pack(data)
+        parallel = shardfmt.pack(data, workers=2)
+        assert serial == parallel
+        assert shardfmt.unpack_cpu(parallel) == data
+
+    def test_parallel_pack_empty_and_tiny(self):
+        assert shardfmt.pack(b"", workers=2) == shardfmt.pack(b"")
+        assert shardfmt.pack(b"x", workers=2) == shardfmt.pack(b"x")
