@@ -137,7 +137,21 @@ def pack(data: bytes, block_raw: int = DEFAULT_BLOCK_RAW,
     (the pure-python compressor does ~12 MB/s per core)."""
     crcs = gf2.crc32c_chunks_numpy(data, block_raw)
     comps: List[bytes] = []
-    if compress and data and workers is not None:
+    if compress and data:
+        # native multi-threaded compressor when the ops library is
+        # built (CPU-only entry point — no GPU needed); python fallback
+        try:
+            from shipyard_amd import ops
+
+            nat = ops.lz4_compress_blocks(data, block_raw)
+            comps = [c if c is not None else
+                     data[i * block_raw:(i + 1) * block_raw]
+                     for i, c in enumerate(nat)]
+        except Exception:
+            comps = []
+    if comps:
+        pass
+    elif compress and data and workers is not None:
         import concurrent.futures as _cf
         import os as _os
 

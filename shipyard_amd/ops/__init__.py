@@ -71,6 +71,11 @@ def _load() -> ctypes.CDLL:
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
         ctypes.c_void_p, ctypes.c_uint32, ctypes.c_void_p,
     ]
+    lib.sy_lz4_compress_blocks.restype = ctypes.c_int
+    lib.sy_lz4_compress_blocks.argtypes = [
+        ctypes.c_void_p, ctypes.c_uint64, ctypes.c_uint32, ctypes.c_void_p,
+        ctypes.c_uint64, ctypes.c_void_p, ctypes.c_uint32, ctypes.c_int,
+    ]
     lib.sy_stage_file.restype = ctypes.c_int
     lib.sy_stage_file.argtypes = [
         ctypes.c_char_p, ctypes.c_void_p, ctypes.c_uint64, ctypes.c_uint64,
@@ -146,6 +151,43 @@ def crc32c_file_digest(data, chunk_size: int = 256 * 1024) -> int:
         acc = gf2.combine_raw(acc, int(r), clen)
         pos += clen
     return gf2.finish(acc, n)
+
+
+def lz4_compress_blocks(data: bytes, block_raw: int,
+                        threads: int = 0):
+    """CPU multi-threaded LZ4 block compression (native authoring path
+    for SYSHARD).  Returns a list with one entry per block: compressed
+    bytes, or None where the block is incompressible (store raw).
+    Host-only — needs no GPU, just the built library."""
+    import numpy as np
+
+    lib = _load()
+    n_blocks = (len(data) + block_raw - 1) // block_raw
+    if n_blocks == 0:
+        return []
+    stride = block_raw + block_raw // 255 + 32
+    dst = np.empty(n_blocks * stride, dtype=np.uint8)
+    lens = np.zeros(n_blocks, dtype=np.uint32)
+    rc = lib.sy_lz4_compress_blocks(
+        ctypes.c_char_p(data), ctypes.c_uint64(len(data)),
+        ctypes.c_uint32(block_raw),
+        ctypes.c_void_p(dst.ctypes.data), ctypes.c_uint64(stride),
+        ctypes.c_void_p(lens.ctypes.data), ctypes.c_uint32(n_blocks),
+        ctypes.c_int(threads))
+    _check(rc, "sy_lz4_compress_blocks")
+    out = []
+    for b in range(n_blocks):
+        ln = int(lens[b])
+        out.append(bytes(dst[b * stride:b * stride + ln].tobytes())
+                   if ln else None)
+    return out
+
+
+def native_compress_available() -> bool:
+    try:
+        return hasattr(_load(), "sy_lz4_compress_blocks")
+    except Exception:
+        return False
 
 
 def gather_copy(src, src_off, dst, dst_off, lens):
