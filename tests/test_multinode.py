@@ -582,3 +582,47 @@ def test_membership_churn_under_load(mx, tmp_path):
             proc.wait(timeout=15)
     finally:
         mx.stop_scheduler()
+
+
+def test_coordinator_restart_recovers_remote_tasks(tmp_path):
+    """Remote execution is durable across coordinator restarts: the
+    agents run on, and a fresh LocalExecutor over the same root
+    collects their results (the store carries all state — the Azure
+    service's durability analogue)."""
+    import time as _time
+
+    ex = LocalExecutor(tmp_path / "root", detect_gpus=False)
+    ex.pool_add({"pool_specification": {
+        "id": "mp",
+        "nodes": [{"id": "n0", "host": "127.0.0.1", "cpu_slots": 1}],
+        "node_configuration": {"rocm": {"verify": False}}}})
+    procs = ex.start_local_agents("mp")
+    ex.jobs_add({"job_specifications": [{
+        "id": "rj",
+        "tasks": [{"id": "t", "command": "sleep 1; echo survived"}],
+    }]}, "mp")
+    deadline = _time.monotonic() + 30
+    while _time.monotonic() < deadline:
+        ex.schedule_once()
+        if ex.store.query_one(
+                "SELECT 1 FROM assignments WHERE state='running'"):
+            break
+        _time.sleep(0.05)
+    # coordinator "crashes" (close the store; agents are separate
+    # processes and keep running)
+    ex._agents.clear()  # simulate losing track of the Popen objects
+    ex.store.close()
+
+    ex2 = LocalExecutor(tmp_path / "root", detect_gpus=False)
+    try:
+        ex2.run_until_idle(timeout=60)
+        t = ex2.tasks_list("rj")[0]
+        assert t["state"] == "completed", dict(t)
+        out = ex2.task_file("mp", "rj", "t").read_text()
+        assert "survived" in out
+    finally:
+        # stop the orphaned agents via pool delete (row-gone exit)
+        ex2.pool_del("mp", force=True)
+        for p in procs:
+            p.wait(timeout=15)
+        ex2.store.close()
