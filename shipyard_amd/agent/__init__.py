@@ -120,8 +120,11 @@ class NodeAgent:
         (None = run until stopped)."""
         last_work = time.monotonic()
         while not self._stop:
-            if self.run_once():
-                last_work = time.monotonic()
+            try:
+                if self.run_once():
+                    last_work = time.monotonic()
+            except Exception as exc:  # keep the agent alive through
+                logger.error("agent loop error: %s", exc)  # store blips
             if (idle_exit_s is not None and not self._handles
                     and time.monotonic() - last_work > idle_exit_s):
                 break
