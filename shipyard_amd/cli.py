@@ -149,6 +149,44 @@ def account_info(ctx, configdir, root, raw):
     ctx.emit(info)
 
 
+@account.command("quota")
+@_common
+@pass_ctx
+def account_quota(ctx, configdir, root, raw):
+    """Capacity vs allocation (the `account quota` analogue: the quota
+    authority here is the host/node inventory, not an Azure account)."""
+    _apply(ctx, configdir, root, raw)
+    import os as _os
+
+    rows = ctx.executor.store.query(
+        "SELECT kind, state, COUNT(*) n FROM slots GROUP BY kind, state")
+    slots = {}
+    for r in rows:
+        slots.setdefault(r["kind"], {})[r["state"]] = r["n"]
+    ctx.emit({
+        "host_gpus": ctx.executor.host_gpu_count(),
+        "host_cpus": _os.cpu_count(),
+        "pools": len(ctx.executor.pool_list()),
+        "slots": slots,
+    })
+
+
+@account.command("images")
+@_common
+@pass_ctx
+def account_images(ctx, configdir, root, raw):
+    """Images cached across pool replicator caches (the `account
+    images` analogue)."""
+    _apply(ctx, configdir, root, raw)
+    out = {}
+    for p in ctx.executor.pool_list():
+        cache = ctx.executor.pool_root(p["id"]) / "images"
+        if cache.is_dir():
+            out[p["id"]] = sorted(d.name for d in cache.iterdir()
+                                  if d.is_dir())
+    ctx.emit(out)
+
+
 # ---------------------------------------------------------------- pool
 @cli.group()
 def pool():
@@ -632,6 +670,17 @@ def storage_clear(ctx, prefix, configdir, root, raw):
         st.delete(name)
         n += 1
     ctx.emit({"deleted": n})
+
+
+@storage.command("del")
+@click.option("--path", required=True,
+              help="object path or directory prefix to delete")
+@_common
+@pass_ctx
+def storage_del(ctx, path, configdir, root, raw):
+    """Delete one object or a whole prefix (reference `storage del`)."""
+    _apply(ctx, configdir, root, raw)
+    ctx.emit({"deleted": ctx.executor.stores["default"].delete(path)})
 
 
 @storage.command("list")

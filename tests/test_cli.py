@@ -301,3 +301,32 @@ class TestParityVerbs:
         res = r.invoke(cli, ["pool", "autoscale-enable", "--poolid",
                              "au", *cfg, *rt])
         assert res.exit_code == 0 and "enabled" in res.output
+
+    def test_account_quota_images_and_storage_del(self, tmp_path):
+        import yaml
+        from click.testing import CliRunner
+
+        from shipyard_amd.cli import cli
+        from shipyard_amd.executor import LocalExecutor
+
+        r = CliRunner()
+        rt = self._root(tmp_path)
+        (tmp_path / "cfg").mkdir()
+        (tmp_path / "cfg" / "pool.yaml").write_text(yaml.safe_dump(
+            {"pool_specification": {
+                "id": "q", "cpu_slots": 2, "gpus": {"dedicated": 0},
+                "node_configuration": {"rocm": {"verify": False}}}}))
+        cfg = ["--configdir", str(tmp_path / "cfg")]
+        assert r.invoke(cli, ["pool", "add", *cfg, *rt]).exit_code == 0
+        res = r.invoke(cli, ["account", "quota", *cfg, *rt])
+        assert res.exit_code == 0 and '"pools": 1' in res.output \
+            and '"idle": 2' in res.output
+        res = r.invoke(cli, ["account", "images", *cfg, *rt])
+        assert res.exit_code == 0
+        # storage del: seed an object then delete its prefix
+        ex = LocalExecutor(tmp_path / "root", detect_gpus=False)
+        ex.stores["default"].upload_bytes("pfx/a.bin", b"1")
+        ex.store.close()
+        res = r.invoke(cli, ["storage", "del", "--path", "pfx",
+                             *cfg, *rt])
+        assert res.exit_code == 0 and '"deleted": true' in res.output
