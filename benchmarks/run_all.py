@@ -38,6 +38,15 @@ def main():
         print(json.dumps(bench_block_size(data, br, 256 << 20)),
               flush=True)
 
+    section("compress_native")
+    subprocess.run([sys.executable,
+                    str(REPO / "benchmarks" / "compress_bench.py"), "128"])
+
+    section("block_size_end_to_end")
+    subprocess.run([sys.executable,
+                    str(REPO / "benchmarks" / "block_size_bench.py"),
+                    "128", "8192"])
+
     section("stager")
     subprocess.run([sys.executable,
                     str(REPO / "benchmarks" / "stager_bench.py"), "1.0"])
