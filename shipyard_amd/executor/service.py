@@ -195,16 +195,19 @@ class LocalExecutor:
             "dedicated, state, node_id) VALUES (?,?,?,?,?,?,?)", rows)
 
     def wait_for_pool_ready(self, pool_id: str,
-                            timeout: float = 300.0) -> None:
+                            timeout: Optional[float] = None) -> None:
         """Node-ready state machine, local edition (reference
         convoy/batch.py:625 _block_for_nodes_ready): verify ROCm when
-        requested, run start-task commands, mark active."""
+        requested, run start-task commands, mark active.  Default
+        timeout = the pool's resize_timeout."""
         row = self.store.query_one("SELECT * FROM pools WHERE id=?",
                                    (pool_id,))
         if row is None:
             raise ExecutorError(f"no pool {pool_id}")
         conf = json.loads(row["spec_json"])
         ps = cfg.pool_settings(conf)
+        if timeout is None:
+            timeout = ps.resize_timeout.total_seconds()
         t0 = time.time()
         try:
             # multi-node pools: GPUs live on the agents' hosts, not here
@@ -444,6 +447,14 @@ class LocalExecutor:
         def compile_one(taskspec: dict) -> Tuple[str, dict, List[str]]:
             nonlocal seq
             ts = cfg.task_settings(taskspec, js, ps)
+            if (ts.multi_instance is not None and len(ps.nodes) > 1
+                    and not ps.inter_node_communication_enabled):
+                # reference batch.py requires inter-node comm for
+                # multi-instance tasks; here it gates gangs SPANNING
+                # nodes (intra-node gangs ride xGMI and need no flag)
+                raise ExecutorError(
+                    f"multi_instance task on multi-node pool {ps.id} "
+                    "requires inter_node_communication_enabled: true")
             tid = ts.id or self._autogen_id(js, seq)
             deps = list(ts.depends_on)
             if ts.depends_on_range:
@@ -1096,9 +1107,24 @@ class LocalExecutor:
                 "('queued','running','cancelling')",
                 (time.time(), d["pool_id"], d["node_id"]))
             n += cur.rowcount
+            # policy (reference attempt_recovery_on_unusable): with
+            # recovery, slots stay schedulable so a restarted agent
+            # resumes service; without, the node's slots go offline
+            # until manual remediation (`pool nodes online`)
+            try:
+                recover = self._pool_settings(
+                    d["pool_id"]).attempt_recovery_on_unusable
+            except ExecutorError:
+                recover = False
+            if not recover:
+                self.store.execute(
+                    "UPDATE slots SET state='offline' WHERE pool_id=? "
+                    "AND node_id=? AND state IN ('idle','busy')",
+                    (d["pool_id"], d["node_id"]))
             self.store.add_event(f"node:{d['pool_id']}/{d['node_id']}",
                                  "agent_dead", {"failed_assignments":
-                                                cur.rowcount})
+                                                cur.rowcount,
+                                                "recovery": recover})
         return n
 
     def _finish_task(self, jid: str, tid: str, rc: int,
@@ -1114,7 +1140,7 @@ class LocalExecutor:
                 conn.execute(
                     "UPDATE slots SET state='idle', task_ref=NULL WHERE "
                     "pool_id=(SELECT pool_id FROM jobs WHERE id=?) AND "
-                    "slot_id=?", (jid, slot_id))
+                    "slot_id=? AND state='busy'", (jid, slot_id))
         js = self._job_settings(jid)
         ps = self._pool_settings(self._job_pool(jid))
         ts = cfg.task_settings(spec, js, ps)
@@ -1412,10 +1438,13 @@ class LocalExecutor:
             return [dict(row) for row in rows]
 
     def _release_slots(self, pool_id: str, slot_ids: List[int]) -> None:
+        # only busy -> idle: a slot offlined meanwhile (dead agent,
+        # manual remediation) must stay offline
         for sid in slot_ids:
             self.store.execute(
                 "UPDATE slots SET state='idle', task_ref=NULL WHERE "
-                "pool_id=? AND slot_id=?", (pool_id, sid))
+                "pool_id=? AND slot_id=? AND state='busy'",
+                (pool_id, sid))
 
     def _task_wd(self, pool_id: str, jid: str, tid: str) -> Path:
         return self.pool_root(pool_id) / "jobs" / jid / "tasks" / tid / "wd"

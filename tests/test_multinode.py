@@ -14,6 +14,7 @@ from shipyard_amd.executor import LocalExecutor
 def _mk_pool(ex, cpu_per_node=2, n_nodes=2):
     ex.pool_add({"pool_specification": {
         "id": "mp",
+        "inter_node_communication_enabled": True,
         "nodes": [{"id": f"n{i}", "host": "127.0.0.1",
                    "cpu_slots": cpu_per_node} for i in range(n_nodes)],
         "node_configuration": {"rocm": {"verify": False}}}})
@@ -626,3 +627,64 @@ def test_coordinator_restart_recovers_remote_tasks(tmp_path):
         for p in procs:
             p.wait(timeout=15)
         ex2.store.close()
+
+
+def test_gang_requires_inter_node_comm_flag(tmp_path):
+    """Reference parity: multi-instance tasks on multi-VM pools require
+    inter_node_communication_enabled (convoy/batch.py submission
+    check); here it gates gangs on multi-node pools."""
+    from shipyard_amd.executor import ExecutorError
+
+    ex = LocalExecutor(tmp_path / "root", detect_gpus=False)
+    ex.pool_add({"pool_specification": {
+        "id": "noinc",
+        "nodes": [{"id": "a", "cpu_slots": 1},
+                  {"id": "b", "cpu_slots": 1}],
+        "node_configuration": {"rocm": {"verify": False}}}})
+    with pytest.raises(ExecutorError, match="inter_node_communication"):
+        ex.jobs_add({"job_specifications": [{
+            "id": "g", "tasks": [{
+                "id": "t", "command": "true",
+                "multi_instance": {
+                    "num_instances": 2,
+                    "gang": {"backend": "gloo", "gpus_per_rank": 0}},
+            }]}]}, "noinc")
+    ex.store.close()
+
+
+def test_dead_agent_slot_policy(tmp_path):
+    """attempt_recovery_on_unusable=False offlines a dead node's slots
+    (manual remediation); True leaves them schedulable for a restarted
+    agent."""
+    import time as _time
+
+    for recover in (False, True):
+        ex = LocalExecutor(tmp_path / f"r{recover}", detect_gpus=False)
+        ex.pool_add({"pool_specification": {
+            "id": "p",
+            "attempt_recovery_on_unusable": recover,
+            "nodes": [{"id": "n0", "cpu_slots": 2}],
+            "node_configuration": {"rocm": {"verify": False}}}})
+        procs = ex.start_local_agents("p")
+        ex.jobs_add({"job_specifications": [{
+            "id": "j", "tasks": [{"id": "t", "command": "sleep 600",
+                                  "max_task_retries": 0}]}]}, "p")
+        deadline = _time.monotonic() + 30
+        while _time.monotonic() < deadline:
+            ex.schedule_once()
+            if ex.store.query_one(
+                    "SELECT 1 FROM assignments WHERE state='running'"):
+                break
+            _time.sleep(0.05)
+        procs[0].kill()
+        procs[0].wait(timeout=10)
+        ex.reap_dead_agents(max_age_s=0.0)
+        ex.run_until_idle(timeout=30)
+        states = {r["state"] for r in ex.store.query(
+            "SELECT state FROM slots WHERE pool_id='p'")}
+        if recover:
+            assert states == {"idle"}, states
+        else:
+            assert states == {"offline"}, states
+        ex.stop_local_agents()
+        ex.store.close()
