@@ -1292,7 +1292,8 @@ class LocalExecutor:
                 pool_id, gpus_needed, cpu_ok=(gpus_needed == 0),
                 ranks=ranks, span_ok=(ts.multi_instance is not None),
                 chunk=(ts.multi_instance.gang.gpus_per_rank
-                       if ts.multi_instance else 1))
+                       if ts.multi_instance else 1),
+                fill=ps.node_fill_type)
             if slots is None:
                 continue
             slot_ids = [s["slot_id"] for s in slots]
@@ -1377,21 +1378,39 @@ class LocalExecutor:
 
     def _try_allocate(self, pool_id: str, gpus: int, cpu_ok: bool,
                       ranks: int = 1, span_ok: bool = False,
-                      chunk: int = 1) -> Optional[List[dict]]:
+                      chunk: int = 1,
+                      fill: str = "pack") -> Optional[List[dict]]:
         """Claim slots for a task.  Single tasks stay on one node;
         gangs (span_ok) may span nodes, each node contributing a
         multiple of `chunk` (= gpus_per_rank) devices so every rank's
-        devices are co-resident."""
+        devices are co-resident.  fill (pool node_fill_type): "pack"
+        fills nodes/devices in order; "spread" prefers the least-loaded
+        node (and, with max_tasks_per_gpu > 1, device)."""
         with self.store.transaction() as conn:
             multi = conn.execute(
                 "SELECT COUNT(*) FROM nodes WHERE pool_id=?",
                 (pool_id,)).fetchone()[0] > 0
+            busy_node: Dict[str, int] = {}
+            busy_dev: Dict[int, int] = {}
+            if fill == "spread":
+                for r in conn.execute(
+                        "SELECT node_id, device_id FROM slots WHERE "
+                        "pool_id=? AND state='busy'", (pool_id,)):
+                    busy_node[r["node_id"]] = \
+                        busy_node.get(r["node_id"], 0) + 1
+                    if r["device_id"] is not None:
+                        busy_dev[r["device_id"]] = \
+                            busy_dev.get(r["device_id"], 0) + 1
             if gpus > 0:
                 idle = list(conn.execute(
                     "SELECT slot_id, kind, device_id, node_id FROM slots "
                     "WHERE pool_id=? AND state='idle' AND kind='gpu' "
                     "ORDER BY node_id, dedicated DESC, device_id ASC",
                     (pool_id,)))
+                if fill == "spread":
+                    idle.sort(key=lambda r: (
+                        busy_node.get(r["node_id"], 0),
+                        busy_dev.get(r["device_id"], 0)))
                 # distinct devices per node, preserving order
                 per_node: Dict[str, List] = {}
                 for row in idle:
@@ -1427,8 +1446,12 @@ class LocalExecutor:
                 rows = list(conn.execute(
                     "SELECT slot_id, kind, device_id, node_id FROM slots "
                     "WHERE pool_id=? AND state='idle' "
-                    "ORDER BY kind='gpu', node_id, slot_id LIMIT ?",
-                    (pool_id, want)))
+                    "ORDER BY kind='gpu', node_id, slot_id", (pool_id,)))
+                if fill == "spread":
+                    rows.sort(key=lambda r: (r["kind"] == "gpu",
+                                             busy_node.get(r["node_id"],
+                                                           0)))
+                rows = rows[:want]
                 if len(rows) < want:
                     return None
             for row in rows:

@@ -688,3 +688,27 @@ def test_dead_agent_slot_policy(tmp_path):
             assert states == {"offline"}, states
         ex.stop_local_agents()
         ex.store.close()
+
+
+def test_node_fill_type_spread_balances(tmp_path):
+    """node_fill_type spread places consecutive tasks on the
+    least-loaded node; pack fills the first node's slots first."""
+    import json as _json
+
+    for fill, expect_nodes in (("pack", {"a"}), ("spread", {"a", "b"})):
+        ex = LocalExecutor(tmp_path / fill, detect_gpus=False)
+        ex.pool_add({"pool_specification": {
+            "id": "p", "node_fill_type": fill,
+            "nodes": [{"id": "a", "cpu_slots": 2},
+                      {"id": "b", "cpu_slots": 2}],
+            "node_configuration": {"rocm": {"verify": False}}}})
+        ex.jobs_add({"job_specifications": [{
+            "id": "j", "tasks": [
+                {"id": "t0", "command": "sleep 600"},
+                {"id": "t1", "command": "sleep 600"}]}]}, "p")
+        ex.schedule_once()
+        nodes = {r["node_id"] for r in ex.store.query(
+            "SELECT node_id FROM slots WHERE state='busy'")}
+        assert nodes == expect_nodes, (fill, nodes)
+        ex.job_terminate("j")
+        ex.store.close()
