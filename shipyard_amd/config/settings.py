@@ -442,7 +442,7 @@ def task_settings(taskspec: Dict[str, Any], job: JobSettings,
         entrypoint=taskspec.get("entrypoint"),
         environment_variables=env,
         gpus=gpus,
-        exclusive_gpus=taskspec.get("exclusive_gpus", gpus > 0),
+        exclusive_gpus=taskspec.get("exclusive_gpus", False),
         shm_size=utils.parse_size(taskspec.get("shm_size")),
         data_volumes=list(taskspec.get("data_volumes") or []),
         shared_data_volumes=list(taskspec.get("shared_data_volumes") or []),

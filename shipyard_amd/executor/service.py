@@ -447,6 +447,10 @@ class LocalExecutor:
         def compile_one(taskspec: dict) -> Tuple[str, dict, List[str]]:
             nonlocal seq
             ts = cfg.task_settings(taskspec, js, ps)
+            if ts.multi_instance is not None and ts.exclusive_gpus:
+                raise ExecutorError(
+                    "exclusive_gpus applies to single tasks; gangs own "
+                    "their ranks' devices via gpus_per_rank")
             if (ts.multi_instance is not None and len(ps.nodes) > 1
                     and not ps.inter_node_communication_enabled):
                 # reference batch.py requires inter-node comm for
@@ -560,6 +564,7 @@ class LocalExecutor:
             conn.execute(
                 "UPDATE jobs SET state='terminated', completed_at=? "
                 "WHERE id=?", (time.time(), job_id))
+        self._run_job_release(job_id)
 
     def _cancel_assignments(self, job_id: str,
                             task_id: Optional[str] = None) -> None:
@@ -1293,12 +1298,18 @@ class LocalExecutor:
                 ranks=ranks, span_ok=(ts.multi_instance is not None),
                 chunk=(ts.multi_instance.gang.gpus_per_rank
                        if ts.multi_instance else 1),
-                fill=ps.node_fill_type)
+                fill=ps.node_fill_type, exclusive=ts.exclusive_gpus)
             if slots is None:
                 continue
             slot_ids = [s["slot_id"] for s in slots]
             device_ids = [s["device_id"] for s in slots
                           if s["kind"] == "gpu"]
+            if ts.exclusive_gpus:
+                # exclusive claims every slot of a device: dedupe for
+                # the launch-side device list
+                seen = set()
+                device_ids = [d for d in device_ids
+                              if not (d in seen or seen.add(d))]
             node_ids = {s.get("node_id", "local") for s in slots}
             self.store.execute(
                 "UPDATE tasks SET state='running', start_time=?, "
@@ -1378,8 +1389,8 @@ class LocalExecutor:
 
     def _try_allocate(self, pool_id: str, gpus: int, cpu_ok: bool,
                       ranks: int = 1, span_ok: bool = False,
-                      chunk: int = 1,
-                      fill: str = "pack") -> Optional[List[dict]]:
+                      chunk: int = 1, fill: str = "pack",
+                      exclusive: bool = False) -> Optional[List[dict]]:
         """Claim slots for a task.  Single tasks stay on one node;
         gangs (span_ok) may span nodes, each node contributing a
         multiple of `chunk` (= gpus_per_rank) devices so every rank's
@@ -1401,6 +1412,38 @@ class LocalExecutor:
                     if r["device_id"] is not None:
                         busy_dev[r["device_id"]] = \
                             busy_dev.get(r["device_id"], 0) + 1
+            if gpus > 0 and exclusive:
+                # exclusive_gpus: the task owns every slot of each
+                # granted device (no max_tasks_per_gpu co-scheduling);
+                # a device qualifies only when ALL its slots are idle
+                per_dev: Dict[Tuple[str, int], List] = {}
+                ok_dev: Dict[Tuple[str, int], bool] = {}
+                for r in conn.execute(
+                        "SELECT slot_id, kind, device_id, node_id, state "
+                        "FROM slots WHERE pool_id=? AND kind='gpu' "
+                        "ORDER BY node_id, device_id, slot_id",
+                        (pool_id,)):
+                    key = (r["node_id"], r["device_id"])
+                    per_dev.setdefault(key, []).append(r)
+                    ok_dev[key] = ok_dev.get(key, True) and \
+                        r["state"] == "idle"
+                by_node: Dict[str, List] = {}
+                for (node, dev), rows_ in per_dev.items():
+                    if ok_dev[(node, dev)]:
+                        by_node.setdefault(node, []).append(rows_)
+                chosen = None  # single-node only (gangs can't be
+                for node, devlists in by_node.items():  # exclusive)
+                    if len(devlists) >= gpus:
+                        chosen = devlists[:gpus]
+                        break
+                if chosen is None:
+                    return None
+                rows = [r for devlist in chosen for r in devlist]
+                for row in rows:
+                    conn.execute(
+                        "UPDATE slots SET state='busy' WHERE pool_id=? "
+                        "AND slot_id=?", (pool_id, row["slot_id"]))
+                return [dict(row) for row in rows]
             if gpus > 0:
                 idle = list(conn.execute(
                     "SELECT slot_id, kind, device_id, node_id FROM slots "
@@ -1471,6 +1514,31 @@ class LocalExecutor:
 
     def _task_wd(self, pool_id: str, jid: str, tid: str) -> Path:
         return self.pool_root(pool_id) / "jobs" / jid / "tasks" / tid / "wd"
+
+    def _process_resource_files(self, ps: cfg.PoolSettings, jid: str,
+                                tid: str, files: List[dict]) -> None:
+        """Stage resource_files into the task wd before launch
+        (reference: SAS resource files on the Batch task; here
+        `source` is a local path or `<account>:<object path>`)."""
+        wd = self._task_wd(ps.id, jid, tid)
+        wd.mkdir(parents=True, exist_ok=True)
+        for rf in files:
+            src = rf["source"]
+            dst = wd / rf["file_path"]
+            dst.parent.mkdir(parents=True, exist_ok=True)
+            if ":" in src and src.split(":", 1)[0] in self.stores:
+                account, remote = src.split(":", 1)
+                data = self.stores[account].download_bytes(remote)
+                dst.write_bytes(data)
+            elif Path(src).is_file():
+                import shutil as _sh
+
+                _sh.copy2(src, dst)
+            else:
+                raise ExecutorError(
+                    f"resource file source not found: {src}")
+            if rf.get("file_mode"):
+                dst.chmod(int(str(rf["file_mode"]), 8))
 
     def _process_input_data(self, ps: cfg.PoolSettings, jid: str, tid: str,
                             specs) -> None:
@@ -1583,7 +1651,15 @@ class LocalExecutor:
             rep = self.replicator(
                 ps.id, concurrency=gs.concurrent_source_downloads,
                 account=gs.storage_account)
-            rep.stage_image(ts.image)
+            try:
+                rep.stage_image(ts.image)
+            except Exception:
+                if not js.allow_run_on_missing_image:
+                    raise
+                logger.warning(
+                    "image %s unavailable; running anyway "
+                    "(allow_run_on_missing_image)", ts.image)
+                return
         env["SHIPYARD_IMAGE_DIR"] = str(cache)
 
     def _launch_task(self, ps: cfg.PoolSettings, js: cfg.JobSettings,
@@ -1605,6 +1681,14 @@ class LocalExecutor:
                                      list(js.input_data) + list(ts.input_data))
             env["SHIPYARD_TASK_INPUT_DIR"] = str(
                 self._task_wd(ps.id, jid, tid))
+        if ts.resource_files:
+            self._process_resource_files(ps, jid, tid, ts.resource_files)
+        working_dir = None
+        if ts.default_working_dir == "shared":
+            working_dir = str(self.pool_root(ps.id) / "jobs" / jid /
+                              "shared")
+        elif str(ts.default_working_dir).startswith("/"):
+            working_dir = ts.default_working_dir
         if ps.per_job_auto_scratch:
             # per-job distributed-scratch analogue (reference
             # shipyard_auto_scratch.sh / BeeOND): one shared fast dir
@@ -1654,8 +1738,31 @@ class LocalExecutor:
             max_wall_time_s=(ts.max_wall_time.total_seconds()
                              if ts.max_wall_time else None),
             wrapper=wrapper,
+            working_dir=working_dir,
         )
         return spec
+
+    def _run_job_release(self, jid: str) -> None:
+        """job_release command: runs once when the job finishes
+        (reference Batch job-release task; here one local run in the
+        job's shared dir)."""
+        if self.store.kv_get(f"job_released:{jid}"):
+            return
+        try:
+            js = self._job_settings(jid)
+        except Exception:
+            return
+        if not js.job_release_command:
+            return
+        self.store.kv_set(f"job_released:{jid}", "1")
+        pool_id = self._job_pool(jid)
+        shared = self.pool_root(pool_id) / "jobs" / jid / "shared"
+        shared.mkdir(parents=True, exist_ok=True)
+        rc, out, err = utils.subprocess_with_output(
+            ["/bin/bash", "-c", js.job_release_command], cwd=str(shared),
+            timeout=300)
+        self.store.add_event(f"job:{jid}", "job-release",
+                             {"rc": rc, "stderr": err.strip()[-500:]})
 
     def _complete_auto_jobs(self) -> None:
         rows = self.store.query(
@@ -1671,6 +1778,7 @@ class LocalExecutor:
                 self.store.execute(
                     "UPDATE jobs SET state='completed', completed_at=? "
                     "WHERE id=?", (time.time(), jid))
+                self._run_job_release(jid)
                 self._reap_auto_pool(jid)
 
     # ----------------------------------------------------------------

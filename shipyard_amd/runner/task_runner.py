@@ -90,6 +90,10 @@ class LaunchSpec:
     rank_start: int = 0
     world_size: Optional[int] = None
     master_addr: str = "127.0.0.1"
+    # cwd override (default_working_dir "shared" / absolute path);
+    # None = the task's wd dir.  Process runtime only — docker mounts
+    # the task wd at /work regardless.
+    working_dir: Optional[str] = None
 
 
 def spec_to_json(spec: LaunchSpec) -> str:
@@ -279,11 +283,12 @@ def _dump_env(env: Dict[str, str], path: Path) -> None:
 
 
 def _spawn(cmd: List[str], env: Dict[str, str], paths: TaskPaths,
-           rank: int) -> RankProc:
+           rank: int, cwd: Optional[str] = None) -> RankProc:
     out = open(paths.stdout, "ab")
     err = open(paths.stderr, "ab")
     proc = subprocess.Popen(
-        cmd, cwd=str(paths.working_dir), env=env, stdout=out, stderr=err,
+        cmd, cwd=cwd or str(paths.working_dir), env=env, stdout=out,
+        stderr=err,
         start_new_session=True)  # own pgid → killable as a group
     return RankProc(rank=rank, proc=proc, paths=paths, _files=(out, err))
 
@@ -310,7 +315,8 @@ def launch(spec: LaunchSpec, pool_root: Path) -> TaskHandle:
         env.update(rt.gpu_env(spec.device_ids))
         _dump_env(env, paths.env_file)
         cmd = _runtime_cmd(spec, command, paths, env)
-        ranks.append(_spawn(cmd, env, paths, 0))
+        ranks.append(_spawn(cmd, env, paths, 0,
+                            cwd=spec.working_dir))
     else:
         port = spec.master_port or _free_port()
         # split LOCALLY granted devices across local ranks
@@ -335,7 +341,8 @@ def launch(spec: LaunchSpec, pool_root: Path) -> TaskHandle:
             })
             _dump_env(env, paths.env_file)
             cmd = _runtime_cmd(spec, command, paths, env)
-            ranks.append(_spawn(cmd, env, paths, rank))
+            ranks.append(_spawn(cmd, env, paths, rank,
+                                cwd=spec.working_dir))
 
     handle = TaskHandle(spec, ranks, start)
     return handle

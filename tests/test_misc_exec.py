@@ -202,3 +202,111 @@ def test_live_tail_streams_incrementally(tmp_path):
     assert len(seen) >= 2, seen
     assert "second" not in seen[0]
     ex.store.close()
+
+
+class TestRound1Fields:
+    """The once parsed-only settings now carry semantics."""
+
+    def _pool(self, ex, **kw):
+        spec = {"pool_specification": {
+            "id": "p", "gpus": {"dedicated": 0}, "cpu_slots": 2,
+            "node_configuration": {"rocm": {"verify": False}}}}
+        spec["pool_specification"].update(kw)
+        ex.pool_add(spec)
+
+    def test_resource_files_staged(self, tmp_path):
+        from shipyard_amd.executor import LocalExecutor
+
+        ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+        self._pool(ex)
+        src = tmp_path / "cfg.txt"
+        src.write_text("from-local\n")
+        ex.stores["default"].upload_bytes("res/remote.txt",
+                                          b"from-store\n")
+        ex.jobs_add({"job_specifications": [{
+            "id": "j", "tasks": [{
+                "id": "t",
+                "command": "cat sub/local.txt remote.txt; "
+                           "test -x runme.sh",
+                "resource_files": [
+                    {"file_path": "sub/local.txt", "source": str(src)},
+                    {"file_path": "remote.txt",
+                     "source": "default:res/remote.txt"},
+                    {"file_path": "runme.sh", "source": str(src),
+                     "file_mode": "755"},
+                ]}]}]}, "p")
+        ex.run_until_idle(timeout=60)
+        t = ex.tasks_list("j")[0]
+        assert t["state"] == "completed", dict(t)
+        out = ex.task_file("p", "j", "t").read_text()
+        assert "from-local" in out and "from-store" in out
+        ex.store.close()
+
+    def test_working_dir_shared(self, tmp_path):
+        from shipyard_amd.executor import LocalExecutor
+
+        ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+        self._pool(ex)
+        ex.jobs_add({"job_specifications": [{
+            "id": "j", "default_working_dir": "shared",
+            "tasks": [
+                {"id": "w", "command": "echo x > handoff.txt"},
+                {"id": "r2", "command": "cat handoff.txt",
+                 "depends_on": ["w"]},
+            ]}]}, "p")
+        ex.run_until_idle(timeout=60)
+        states = {t["id"]: t["state"] for t in ex.tasks_list("j")}
+        assert states == {"w": "completed", "r2": "completed"}
+        assert (ex.pool_root("p") / "jobs" / "j" / "shared" /
+                "handoff.txt").exists()
+        ex.store.close()
+
+    def test_job_release_command_runs_once(self, tmp_path):
+        from shipyard_amd.executor import LocalExecutor
+
+        ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+        self._pool(ex)
+        ex.jobs_add({"job_specifications": [{
+            "id": "j", "auto_complete": True,
+            "job_release": {"command": "echo released >> release.log"},
+            "tasks": [{"id": "t", "command": "true"}]}]}, "p")
+        ex.run_until_idle(timeout=60)
+        ex._complete_auto_jobs()
+        ex._complete_auto_jobs()  # idempotent
+        log = ex.pool_root("p") / "jobs" / "j" / "shared" / "release.log"
+        assert log.read_text().count("released") == 1
+        ex.store.close()
+
+    def test_exclusive_gpus_claims_whole_device(self, tmp_path):
+        from shipyard_amd.executor import LocalExecutor
+
+        ex = LocalExecutor(tmp_path / "r", detect_gpus=False)
+        ex.pool_add({"pool_specification": {
+            "id": "p", "gpus": {"dedicated": 2}, "max_tasks_per_gpu": 2,
+            "node_configuration": {"rocm": {"verify": False}}}})
+        ex.jobs_add({"job_specifications": [{
+            "id": "j", "tasks": [
+                {"id": "ex", "gpus": 1, "exclusive_gpus": True,
+                 "command": "sleep 600"},
+                {"id": "sh1", "gpus": 1, "command": "sleep 600"},
+                {"id": "sh2", "gpus": 1, "command": "sleep 600"},
+                {"id": "sh3", "gpus": 1, "command": "sleep 600"},
+            ]}]}, "p")
+        ex.schedule_once()
+        rows = ex.store.query(
+            "SELECT t.id, t.state, t.slots_json FROM tasks t "
+            "ORDER BY t.seq")
+        import json as _json
+
+        states = {r["id"]: r["state"] for r in rows}
+        # exclusive took BOTH slots of device 0; two sharers fit on
+        # device 1's two slots; the fourth waits
+        assert states["ex"] == "running"
+        ex_slots = _json.loads(
+            [r["slots_json"] for r in rows if r["id"] == "ex"][0])
+        assert len(ex_slots) == 2
+        running = [i for i in ("sh1", "sh2", "sh3")
+                   if states[i] == "running"]
+        assert len(running) == 2
+        ex.job_terminate("j")
+        ex.store.close()
