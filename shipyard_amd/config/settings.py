@@ -269,6 +269,8 @@ class TaskSettings:
     additional_singularity_options: List[str]
     singularity_cmd: str
     default_working_dir: str
+    # consumed pre-expansion by executor._add_tasks_for_job (factories
+    # generate concrete taskspecs before task_settings() runs)
     task_factory: Optional[dict]
     labels: List[str]
     rocprof: bool = False
@@ -301,6 +303,8 @@ class JobSettings:
     default_working_dir: str
     autogen_task_id_prefix: str
     autogen_task_id_zfill: int
+    # accepted no-op: dependencies are always enabled locally (the
+    # reference needs the flag for the Batch service's usesTaskDependencies)
     force_enable_task_dependencies: bool
     recurrence: Optional[RecurrenceSettings]
     federation_constraints: Optional[dict]
