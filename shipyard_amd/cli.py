@@ -873,6 +873,28 @@ def fed_process(ctx, configdir, root, raw):
     ctx.emit({"processed": n})
 
 
+@fed.command("jobs-zap")
+@click.option("--id", "qid", type=int, default=None,
+              help="queued/blocked action id to drop")
+@click.option("--jobid", default=None,
+              help="force-delete a landed federation job")
+@_common
+@pass_ctx
+def fed_jobs_zap(ctx, qid, jobid, configdir, root, raw):
+    """Force-remove a stuck federation action or landed job (reference
+    `fed jobs zap`)."""
+    _apply(ctx, configdir, root, raw)
+    out = {}
+    if qid is not None:
+        out["queue_deleted"] = ctx.executor.store.execute(
+            "DELETE FROM fed_queue WHERE id=?", (qid,)).rowcount
+    if jobid is not None:
+        ctx.executor.job_terminate(jobid)
+        ctx.executor.job_del(jobid)
+        out["job_deleted"] = jobid
+    ctx.emit(out)
+
+
 @fed.command("list")
 @_common
 @pass_ctx

@@ -153,3 +153,33 @@ def test_fed_fixup_gang_to_pool_size(tmp_path):
         "SELECT payload FROM events WHERE category='job-fixup'")
     assert ev is not None and "gpu_slots" in ev["payload"]
     ex.store.close()
+
+
+def test_fed_zap_cli(tmp_path):
+    """fed jobs-zap drops stuck actions and force-deletes landed jobs."""
+    import yaml
+    from click.testing import CliRunner
+
+    from shipyard_amd.cli import cli
+
+    r = CliRunner()
+    cfg = tmp_path / "cfg"
+    cfg.mkdir()
+    (cfg / "pool.yaml").write_text(yaml.safe_dump({"pool_specification": {
+        "id": "zp", "cpu_slots": 1, "gpus": {"dedicated": 0},
+        "node_configuration": {"rocm": {"verify": False}}}}))
+    (cfg / "federation.yaml").write_text(yaml.safe_dump({"federation": {
+        "federations": {"f": {"pools": ["zp"]}}}}))
+    (cfg / "jobs.yaml").write_text(yaml.safe_dump({"job_specifications": [
+        {"id": "zj", "tasks": [{"id": "t", "command": "sleep 600"}]}]}))
+    args = ["--configdir", str(cfg), "--root", str(tmp_path / "root")]
+    assert r.invoke(cli, ["pool", "add", *args]).exit_code == 0
+    res = r.invoke(cli, ["fed", "jobs-add", "--federation-id", "f", *args])
+    assert res.exit_code == 0, res.output
+    # zap the queued action before it is processed
+    res = r.invoke(cli, ["fed", "jobs-list", *args])
+    assert "add_job" in res.output
+    res = r.invoke(cli, ["fed", "jobs-zap", "--id", "1", *args])
+    assert res.exit_code == 0 and '"queue_deleted": 1' in res.output
+    res = r.invoke(cli, ["fed", "jobs-list", *args])
+    assert "add_job" not in res.output
