@@ -34,6 +34,10 @@ import statistics
 import sys
 import time
 
+# multi-process GPU runs need dmabuf IPC (the host driver rejects
+# legacy IPC); harmless for single-rank runs
+os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+
 
 def measure_submit_launch_p50(samples: int = 10):
     """Submit tiny process tasks through the executor; p50 of
