@@ -712,3 +712,28 @@ def test_node_fill_type_spread_balances(tmp_path):
         assert nodes == expect_nodes, (fill, nodes)
         ex.job_terminate("j")
         ex.store.close()
+
+
+def test_agent_idle_exit_flag(tmp_path):
+    """`--idle-exit S` makes an agent exit after S idle seconds
+    (script/CI use, mirroring `daemon --idle-exit`)."""
+    import subprocess
+    import sys
+
+    ex = LocalExecutor(tmp_path / "root", detect_gpus=False)
+    ex.pool_add({"pool_specification": {
+        "id": "p", "nodes": [{"id": "n0", "cpu_slots": 1}],
+        "node_configuration": {"rocm": {"verify": False}}}})
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "shipyard_amd.agent", "--root",
+         str(ex.root), "--pool", "p", "--node", "n0",
+         "--idle-exit", "0.3"])
+    try:
+        rc = proc.wait(timeout=30)
+        assert rc == 0
+    finally:
+        if proc.poll() is None:
+            proc.kill()
+    node = ex.nodes_list("p")[0]
+    assert node["state"] == "offline"  # clean shutdown path ran
+    ex.store.close()
