@@ -560,7 +560,6 @@ def test_membership_churn_under_load(mx, tmp_path):
         # grow: n2 joins mid-flight with its own agent
         mx.node_add("mp", {"id": "n2", "host": "127.0.0.1",
                            "cpu_slots": 1})
-        pkg_root = str((tmp_path / "..").resolve())
         proc = subprocess.Popen(
             [sys.executable, "-m", "shipyard_amd.agent", "--root",
              str(mx.root), "--pool", "mp", "--node", "n2"],
