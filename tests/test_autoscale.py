@@ -1,7 +1,6 @@
 """Autoscale scenario evaluation (reference convoy/autoscale.py parity)."""
 import datetime
 
-import pytest
 
 from shipyard_amd.config.settings import pool_settings
 from shipyard_amd.executor import LocalExecutor

@@ -1,7 +1,6 @@
 """CLI end-to-end tests (the min end-to-end slice of SURVEY.md §7.3:
 pool add -> jobs add -> files stream -> jobs del, all local)."""
 import json
-from pathlib import Path
 
 import pytest
 import yaml

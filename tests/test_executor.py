@@ -4,7 +4,6 @@ Covers the seam the reference leaves to the Azure Batch service
 (SURVEY.md §4): pool/job/task state machines, dependencies, retries,
 exit conditions, merge tasks, gang launch/teardown.
 """
-import json
 import time
 
 import pytest
