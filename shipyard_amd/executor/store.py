@@ -197,8 +197,11 @@ class Store:
                 "DELETE FROM events WHERE ts < ?", (before_ts,)).rowcount
             c2 = self._conn.execute(
                 "DELETE FROM perf WHERE ts < ?", (before_ts,)).rowcount
+            c3 = self._conn.execute(
+                "DELETE FROM fed_queue WHERE state IN ('done','failed') "
+                "AND enqueued_at < ?", (before_ts,)).rowcount
             self._conn.commit()
-        return c1 + c2
+        return c1 + c2 + c3
 
     # -- kv ----------------------------------------------------------
     def kv_set(self, key: str, value: str) -> None:
