@@ -10,7 +10,7 @@ the reference's reliance on dockerd gzip inflate + CPU MD5
 Layout (little-endian):
   8s  magic  b"SYSHARD1"
   u32 flags            (bit0: blocks are LZ4; 0 = all stored)
-  u32 block_raw        (max raw bytes per block; 64 KiB default)
+  u32 block_raw        (max raw bytes per block; 8 KiB default)
   u64 raw_size
   u32 n_blocks
   n_blocks * { u64 comp_off, u32 comp_len, u32 raw_len, u32 crc32c }
